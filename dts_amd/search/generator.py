@@ -1,0 +1,113 @@
+"""Strategy and user-intent generation.
+
+Parity: reference backend/core/dts/components/generator.py:30-180 —
+`generate_strategies` (one structured call → N Strategy), `generate_intents`
+(one structured call → K UserIntent), FIXED_INTENT fallback persona
+(ref :21-27), bounded concurrency, retry on transient errors.
+"""
+
+from __future__ import annotations
+
+import asyncio
+from typing import Any, Callable, Optional
+
+from dts_amd.llm.backend import LLM
+from dts_amd.llm.types import Message
+from dts_amd.search import prompts
+from dts_amd.search.retry import llm_retry
+from dts_amd.search.types import Strategy, UserIntent, format_message_history
+from dts_amd.utils.logging import logger
+
+FIXED_INTENT = UserIntent(
+    id="fixed_engaged_critic",
+    label="Engaged Critic",
+    description="A thoughtful user who engages constructively while maintaining healthy skepticism",
+    emotional_tone="curious but skeptical",
+    cognitive_stance="analytical, asks probing questions",
+)
+
+
+class StrategyGenerator:
+    def __init__(
+        self,
+        llm: LLM,
+        goal: str,
+        model: Optional[str] = None,
+        temperature: float = 0.7,
+        max_concurrency: int = 16,
+        on_usage: Optional[Callable[[Any, str], None]] = None,
+        max_tokens: int = 1024,
+        seed: Optional[int] = None,
+    ) -> None:
+        self.llm = llm
+        self.goal = goal
+        self.model = model
+        self.temperature = temperature
+        self.max_tokens = max_tokens
+        self.seed = seed
+        self._sem = asyncio.Semaphore(max_concurrency)
+        self._on_usage = on_usage
+
+    async def generate_strategies(
+        self,
+        first_message: str,
+        count: int,
+        deep_research_context: Optional[str] = None,
+    ) -> list:
+        system, user = prompts.conversation_tree_generator(
+            num_nodes=count,
+            conversation_goal=self.goal,
+            conversation_context=first_message,
+            deep_research_context=deep_research_context,
+        )
+        result = await self._call_json(system, user, phase="strategy")
+        if not result:
+            raise RuntimeError("Strategy generation failed after retries")
+        strategies = [
+            Strategy(tagline=tagline, description=str(desc))
+            for tagline, desc in result.get("nodes", {}).items()
+        ]
+        return strategies
+
+    async def generate_intents(self, history: list, count: int) -> list:
+        system, user = prompts.user_intent_generator(
+            num_intents=count,
+            conversation_goal=self.goal,
+            conversation_history=format_message_history(history),
+        )
+        result = await self._call_json(system, user, phase="intent")
+        if not result:
+            raise RuntimeError("Intent generation failed after retries")
+        intents = []
+        for data in result.get("intents", []):
+            try:
+                intents.append(
+                    UserIntent(
+                        id=str(data.get("id", "unknown")),
+                        label=str(data.get("label", "Unknown")),
+                        description=str(data.get("description", "")),
+                        emotional_tone=str(data.get("emotional_tone", "neutral")),
+                        cognitive_stance=str(data.get("cognitive_stance", "neutral")),
+                    )
+                )
+            except Exception as e:  # noqa: BLE001
+                logger.warning("Failed to parse intent: %s", e)
+        return intents
+
+    async def _call_json(self, system: str, user: str, phase: str) -> Optional[dict]:
+        async with self._sem:
+            return await self._call_json_inner(system, user, phase)
+
+    @llm_retry(max_attempts=3)
+    async def _call_json_inner(self, system: str, user: str, phase: str) -> Optional[dict]:
+        completion = await self.llm.complete(
+            [Message.system(system), Message.user(user)],
+            model=self.model,
+            temperature=self.temperature,
+            structured_output=True,
+            max_tokens=self.max_tokens,
+            seed=self.seed,
+        )
+        if self._on_usage:
+            self._on_usage(completion, phase)
+        return completion.data

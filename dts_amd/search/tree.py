@@ -1,0 +1,156 @@
+"""Dialogue tree container.
+
+Parity: reference backend/core/dts/tree.py:30-194 — dict-of-nodes tree with
+add_child, active_leaves, path_to_root, backpropagate (visits/value_sum/
+value_mean to root, ref :109-120), prune_subtree, best_leaf_by_score,
+statistics. Node ids are UUID4 strings (ref :20-22).
+
+MI355X note: the node lineage here is the ground truth for paged-KV prefix
+sharing — `add_child` is the fork point at which the serving engine shares
+blocks. The tree itself stays a plain Python
+structure; it is never on the hot path.
+"""
+
+from __future__ import annotations
+
+import uuid
+from typing import Iterator, Optional
+
+from dts_amd.search.types import DialogueNode, NodeStatus
+
+
+def generate_node_id() -> str:
+    return str(uuid.uuid4())
+
+
+class DialogueTree:
+    def __init__(self, root_id: str) -> None:
+        self.root_id = root_id
+        self.nodes: dict[str, DialogueNode] = {}
+
+    @classmethod
+    def create(cls, root: DialogueNode) -> "DialogueTree":
+        tree = cls(root_id=root.id)
+        tree.nodes[root.id] = root
+        return tree
+
+    def get(self, node_id: str) -> DialogueNode:
+        if node_id not in self.nodes:
+            raise KeyError(f"Node {node_id} not found in tree")
+        return self.nodes[node_id]
+
+    def get_root(self) -> DialogueNode:
+        return self.get(self.root_id)
+
+    def add_node(self, node: DialogueNode) -> None:
+        self.nodes[node.id] = node
+
+    def add_child(self, parent_id: str, child: DialogueNode) -> None:
+        parent = self.get(parent_id)
+        child.parent_id = parent_id
+        child.depth = parent.depth + 1
+        self.nodes[child.id] = child
+        parent.children.append(child.id)
+
+    def remove_node(self, node_id: str) -> None:
+        node = self.nodes.get(node_id)
+        if node is None:
+            return
+        if node.parent_id:
+            parent = self.get(node.parent_id)
+            if node_id in parent.children:
+                parent.children.remove(node_id)
+        del self.nodes[node_id]
+
+    def all_nodes(self) -> list:
+        return list(self.nodes.values())
+
+    def active_nodes(self) -> list:
+        return [n for n in self.nodes.values() if n.status == NodeStatus.ACTIVE]
+
+    def active_leaves(self) -> list:
+        return [
+            n
+            for n in self.nodes.values()
+            if n.status == NodeStatus.ACTIVE and not n.children
+        ]
+
+    def leaves_at_depth(self, depth: int) -> list:
+        return [n for n in self.nodes.values() if n.depth == depth and not n.children]
+
+    def path_to_root(self, node_id: str) -> list:
+        path = []
+        current: Optional[str] = node_id
+        while current is not None:
+            node = self.get(current)
+            path.append(node)
+            current = node.parent_id
+        return path
+
+    def path_from_root(self, node_id: str) -> list:
+        return list(reversed(self.path_to_root(node_id)))
+
+    def backpropagate(self, node_id: str, score: float) -> None:
+        current: Optional[str] = node_id
+        while current is not None:
+            node = self.get(current)
+            node.stats.visits += 1
+            node.stats.value_sum += score
+            node.stats.value_mean = node.stats.value_sum / node.stats.visits
+            current = node.parent_id
+
+    def prune_node(self, node_id: str, reason: Optional[str] = None) -> None:
+        node = self.get(node_id)
+        node.status = NodeStatus.PRUNED
+        node.prune_reason = reason
+
+    def prune_subtree(self, node_id: str, reason: Optional[str] = None) -> int:
+        count = 0
+        stack = [node_id]
+        while stack:
+            nid = stack.pop()
+            node = self.get(nid)
+            if node.status != NodeStatus.PRUNED:
+                node.status = NodeStatus.PRUNED
+                node.prune_reason = reason
+                count += 1
+            stack.extend(node.children)
+        return count
+
+    def descendants(self, node_id: str) -> Iterator[DialogueNode]:
+        node = self.get(node_id)
+        for child_id in node.children:
+            child = self.get(child_id)
+            yield child
+            yield from self.descendants(child_id)
+
+    def subtree_size(self, node_id: str) -> int:
+        return 1 + sum(1 for _ in self.descendants(node_id))
+
+    def max_depth(self) -> int:
+        if not self.nodes:
+            return 0
+        return max(n.depth for n in self.nodes.values())
+
+    def best_leaf(self) -> Optional[DialogueNode]:
+        leaves = self.active_leaves()
+        if not leaves:
+            return None
+        return max(leaves, key=lambda n: n.stats.value_mean)
+
+    def best_leaf_by_score(self) -> Optional[DialogueNode]:
+        leaves = self.active_leaves()
+        if not leaves:
+            return None
+        return max(leaves, key=lambda n: n.stats.aggregated_score)
+
+    def statistics(self) -> dict:
+        all_nodes = list(self.nodes.values())
+        return {
+            "total_nodes": len(all_nodes),
+            "active_nodes": sum(1 for n in all_nodes if n.status == NodeStatus.ACTIVE),
+            "pruned_nodes": sum(1 for n in all_nodes if n.status == NodeStatus.PRUNED),
+            "active_leaves": len(self.active_leaves()),
+            "max_depth": self.max_depth(),
+            "total_visits": sum(n.stats.visits for n in all_nodes),
+        }

@@ -1,0 +1,149 @@
+"""Evaluator tests (parity coverage of ref tests/.../test_evaluator.py)."""
+
+import json
+
+import pytest
+
+from dts_amd.llm import LLM, FakeBackend, Message, ScriptedBackend
+from dts_amd.search.evaluator import TrajectoryEvaluator
+from dts_amd.search.types import DialogueNode, Strategy, UserIntent
+
+
+def make_node(nid, parent=None):
+    return DialogueNode(
+        id=nid,
+        parent_id=parent,
+        strategy=Strategy(tagline="t", description="d"),
+        messages=[Message.user("q"), Message.assistant("a")],
+    )
+
+
+def judge_json(total):
+    return json.dumps(
+        {
+            "criteria": {
+                "goal_achieved": {"score": 0.9, "rationale": "good"},
+                "efficient_path": {"score": 0.2, "rationale": "slow"},
+            },
+            "total_score": total,
+            "confidence": "high",
+            "summary": "s",
+            "key_turning_point": "k",
+        }
+    )
+
+
+def make_eval(backend, **kw):
+    kw.setdefault("prune_threshold", 5.0)
+    return TrajectoryEvaluator(LLM(backend, default_model="m"), goal="g", **kw)
+
+
+class TestAbsolute:
+    def test_median_of_three_judges(self, run_async):
+        backend = ScriptedBackend([judge_json(4.0), judge_json(8.0), judge_json(6.0)])
+        ev = make_eval(backend)
+        node = make_node("n1")
+        scores = run_async(ev.evaluate_absolute([node]))
+        assert scores["n1"].aggregated_score == 6.0
+        assert sorted(scores["n1"].individual_scores) == [4.0, 6.0, 8.0]
+        assert node.stats.aggregated_score == 6.0
+        # critique extracted from median judge
+        assert node.stats.critiques["strengths"]
+        assert node.stats.critiques["weaknesses"]
+
+    def test_failed_judge_scores_zero(self, run_async):
+        # one judge returns invalid JSON 3x (its parse retries exhaust) -> 0.0
+        responses = []
+        # gather order: the three judges interleave; use a FakeBackend-free
+        # deterministic script: judge1 ok, judge2 ok, judge3 three bad parses
+        backend = ScriptedBackend(
+            [judge_json(6.0), judge_json(7.0), "bad", "bad", "bad"]
+        )
+        ev = make_eval(backend)
+        node = make_node("n1")
+        scores = run_async(ev.evaluate_absolute([node]))
+        assert 0.0 in scores["n1"].individual_scores
+        assert scores["n1"].aggregated_score == 6.0  # median of {6,7,0}
+
+
+class TestComparative:
+    def _ranking_json(self, ids, scores):
+        return json.dumps(
+            {
+                "critiques": {
+                    i: {"weaknesses": ["w"], "strengths": ["s"], "key_moment": "m"}
+                    for i in ids
+                },
+                "ranking": [
+                    {"rank": r + 1, "trajectory_id": i, "score": s, "reason": "r"}
+                    for r, (i, s) in enumerate(zip(ids, scores))
+                ],
+                "ranking_confidence": "high",
+            }
+        )
+
+    def test_sibling_group_ranked(self, run_async):
+        a, b = make_node("a", parent="p"), make_node("b", parent="p")
+        a.user_intent = UserIntent(
+            id="x", label="X", description="", emotional_tone="e", cognitive_stance="c"
+        )
+        backend = ScriptedBackend([self._ranking_json(["a", "b"], [7.5, 6.0])])
+        ev = make_eval(backend)
+        scores = run_async(ev.evaluate_comparative([a, b]))
+        assert scores["a"].aggregated_score == 7.5
+        assert scores["b"].aggregated_score == 6.0
+        # synthetic [s,s,s] (ref evaluator.py:305-311)
+        assert scores["a"].individual_scores == [7.5, 7.5, 7.5]
+        assert scores["a"].passed
+        assert scores["b"].passed  # 6.0 >= 5.0 threshold
+        assert scores["b"].pass_votes == 3
+
+    def test_missing_node_in_ranking_zeroed(self, run_async):
+        a, b = make_node("a", parent="p"), make_node("b", parent="p")
+        backend = ScriptedBackend([self._ranking_json(["a"], [7.5])])
+        ev = make_eval(backend)
+        scores = run_async(ev.evaluate_comparative([a, b]))
+        assert scores["b"].aggregated_score == 0.0
+
+    def test_invalid_ranking_falls_back_absolute(self, run_async):
+        a, b = make_node("a", parent="p"), make_node("b", parent="p")
+        # first call: no "ranking" key -> fallback: 6 judge calls
+        backend = ScriptedBackend(
+            ['{"nothing": 1}'] + [judge_json(5.0)] * 6
+        )
+        ev = make_eval(backend)
+        scores = run_async(ev.evaluate_comparative([a, b]))
+        assert scores["a"].aggregated_score == 5.0
+        assert scores["b"].aggregated_score == 5.0
+
+    def test_single_node_uses_absolute(self, run_async):
+        node = make_node("solo")
+        backend = ScriptedBackend([judge_json(6.0)] * 3)
+        ev = make_eval(backend)
+        scores = run_async(ev.evaluate_comparative([node]))
+        assert scores["solo"].aggregated_score == 6.0
+
+    def test_mixed_groups(self, run_async):
+        """Singles get 3-judge median, groups get one ranking call
+        (ref evaluator.py:115-138)."""
+        a, b = make_node("a", parent="p1"), make_node("b", parent="p1")
+        solo = make_node("solo", parent="p2")
+        fake = FakeBackend()
+        ev = make_eval(fake)
+        scores = run_async(ev.evaluate_comparative([a, b, solo]))
+        assert set(scores) == {"a", "b", "solo"}
+
+
+def test_comparative_pass_votes():
+    """Score >= threshold gives pass_votes=3 in comparative mode."""
+    # covered through TestComparative, here check boundary semantics directly
+    from dts_amd.search.types import AggregatedScore
+
+    s = AggregatedScore(
+        individual_scores=[6.0] * 3,
+        aggregated_score=6.0,
+        pass_threshold=5.0,
+        pass_votes=3,
+        passed=True,
+    )
+    assert s.passed

@@ -1,0 +1,151 @@
+"""Simulator tests (parity coverage of ref tests/.../test_simulator.py)."""
+
+import json
+
+import pytest
+
+from dts_amd.llm import LLM, FakeBackend, Message, ScriptedBackend
+from dts_amd.search.generator import FIXED_INTENT
+from dts_amd.search.simulator import ConversationSimulator, TERMINATION_SIGNALS
+from dts_amd.search.tree import DialogueTree, generate_node_id
+from dts_amd.search.types import DialogueNode, NodeStatus, Strategy, UserIntent
+
+
+def make_sim(backend, **kw):
+    return ConversationSimulator(LLM(backend, default_model="m"), goal="g", **kw)
+
+
+def make_node(msg="hello, can you help me learn this?"):
+    return DialogueNode(
+        id=generate_node_id(),
+        strategy=Strategy(tagline="t", description="d"),
+        messages=[Message.user(msg)],
+    )
+
+
+INTENT = UserIntent(
+    id="i1", label="L", description="D", emotional_tone="engaged", cognitive_stance="exploring"
+)
+
+
+class TestTermination:
+    @pytest.mark.parametrize("signal", ["goodbye", "stop", "i give up", "never mind"])
+    def test_signals_terminate(self, signal):
+        sim = make_sim(FakeBackend())
+        assert sim._should_terminate(f"ok then, {signal}!")
+
+    def test_short_frustrated(self):
+        sim = make_sim(FakeBackend())
+        assert sim._should_terminate("nope.")
+        assert not sim._should_terminate("This is a wonderful detailed answer thanks a lot")
+
+    def test_terminating_user_marks_terminal(self, run_async):
+        # user reply contains a termination signal -> node TERMINAL, no assistant reply
+        backend = ScriptedBackend(["thanks, bye"])
+        sim = make_sim(backend)
+        node = make_node()
+        result = run_async(sim._expand_linear(node, turns=3))
+        assert result.status == NodeStatus.TERMINAL
+        # one user message appended, no assistant
+        assert [m.role for m in result.messages] == ["user", "user"]
+
+
+class TestLinearExpansion:
+    def test_turn_structure(self, run_async):
+        sim = make_sim(FakeBackend())
+        node = make_node()
+        result = run_async(sim._expand_linear(node, turns=2))
+        roles = [m.role for m in result.messages]
+        assert roles == ["user", "user", "assistant", "user", "assistant"]
+        assert result.status == NodeStatus.ACTIVE
+
+    def test_batch_collects_errors(self, run_async):
+        # all three calls for node 1's first user sim are empty -> error status
+        backend = ScriptedBackend(["", "", ""])
+        sim = make_sim(backend)
+        node = make_node()
+        expanded = run_async(sim._expand_linear_batch([node], turns=1))
+        assert expanded == [node]  # node returned with error status
+        assert node.status == NodeStatus.ERROR
+        assert "empty user response" in node.prune_reason
+
+
+class TestEmptyRetry:
+    def test_retries_then_succeeds(self, run_async):
+        backend = ScriptedBackend(["", "  ", "real answer"])
+        sim = make_sim(backend)
+        out = run_async(
+            sim._call_with_retry([Message.user("x")], phase="user", max_tokens=32)
+        )
+        assert out == "real answer"
+        assert len(backend.calls) == 3
+
+
+class TestForking:
+    def test_fork_creates_children_in_tree(self, run_async):
+        sim = make_sim(FakeBackend())
+        root = DialogueNode(id=generate_node_id(), messages=[Message.user("start")])
+        tree = DialogueTree.create(root)
+        parent = make_node()
+        tree.add_child(root.id, parent)
+
+        async def gen_intents(history, count):
+            return [INTENT, FIXED_INTENT][:count]
+
+        expanded = run_async(
+            sim.expand_nodes(
+                [parent], turns=1, intents_per_node=2, tree=tree, generate_intents=gen_intents
+            )
+        )
+        assert len(expanded) == 2
+        assert all(n.parent_id == parent.id for n in expanded)
+        assert all(n.user_intent is not None for n in expanded)
+        assert len(parent.children) == 2
+        # rephrased first message + assistant reply on turn 0 (user sim skipped)
+        for n in expanded:
+            assert n.messages[0].role == "user"
+            assert n.messages[1].role == "assistant"
+
+    def test_intent_failure_falls_back_linear(self, run_async):
+        sim = make_sim(FakeBackend())
+        parent = make_node()
+
+        async def failing_intents(history, count):
+            raise RuntimeError("no intents")
+
+        expanded = run_async(
+            sim.expand_nodes(
+                [parent], turns=1, intents_per_node=2, tree=None, generate_intents=failing_intents
+            )
+        )
+        # falls back to linear expansion of the parent itself
+        assert expanded == [parent]
+        assert parent.user_intent is None
+
+    def test_intents_per_node_one_short_circuits(self, run_async):
+        """intents_per_node<=1 -> linear, generate_intents never called
+        (ref simulator.py:123-125, the FIXED_INTENT-dead-code fact
+        SURVEY.md §4.1.2)."""
+        sim = make_sim(FakeBackend())
+        parent = make_node()
+        called = []
+
+        async def gen(history, count):
+            called.append(1)
+            return [FIXED_INTENT]
+
+        expanded = run_async(
+            sim.expand_nodes([parent], turns=1, intents_per_node=1, generate_intents=gen)
+        )
+        assert expanded == [parent]
+        assert not called
+
+    def test_rephrase_failure_keeps_original(self, run_async):
+        # rephrase returns empty 3x -> keep original first message, then 1 turn
+        responses = ["", "", ""] + ["assistant reply"]
+        backend = ScriptedBackend(responses)
+        sim = make_sim(backend)
+        node = make_node("original opening")
+        result = run_async(sim._expand_with_intent(node, turns=1, first_intent=INTENT))
+        assert result.messages[0].content == "original opening"
+        assert result.messages[1].content == "assistant reply"
