@@ -1,0 +1,361 @@
+"""The MI355X serving engine + the LocalBackend seam.
+
+ServingEngine owns one model replica, its paged KV pool, the
+continuous-batching scheduler and the sampler, and runs the step loop on a
+dedicated thread (the asyncio search loop stays responsive for events —
+SURVEY.md §7 hard-part 5: request queue + futures across the GIL).
+
+LocalBackend implements the InferenceBackend protocol
+(dts_amd/llm/backend.py) — the in-process replacement for the reference's
+HTTPS client (ref client.py:153): chat messages are rendered to a
+token-exact template, structured phases get a constrained-decoding
+FormGuide (serving/structured.py), and the awaitable future resolves when
+the scheduler finishes the sequence.
+"""
+
+from __future__ import annotations
+
+import asyncio
+import concurrent.futures
+import re
+import threading
+from dataclasses import dataclass
+from typing import Optional
+
+import torch
+
+from dts_amd.llm.errors import ContextLengthError
+from dts_amd.llm.types import Completion, Message, SamplingParams, Usage
+from dts_amd.models.config import ModelSpec, get_model_spec
+from dts_amd.serving import structured
+from dts_amd.serving.kv_cache import BlockManager, KVCachePool
+from dts_amd.serving.sampler import Sampler
+from dts_amd.serving.scheduler import Scheduler
+from dts_amd.serving.sequence import Sequence, SeqStatus
+from dts_amd.serving.tokenizer import ChatTemplate, SyntheticTokenizer
+from dts_amd.utils.logging import logger
+
+
+@dataclass
+class GenerationResult:
+    token_ids: list
+    text: str
+    finish_reason: str
+    prompt_tokens: int
+    completion_tokens: int
+
+
+class ServingEngine:
+    def __init__(
+        self,
+        model_name: str = "llama-3-8b",
+        device: Optional[str] = None,
+        dtype: Optional[torch.dtype] = None,
+        num_blocks: Optional[int] = None,
+        kv_memory_bytes: Optional[int] = None,
+        block_size: int = 16,
+        max_batch_tokens: int = 8192,
+        max_running: int = 256,
+        weight_seed: int = 0,
+        model: Optional[object] = None,
+    ) -> None:
+        self.spec: ModelSpec = get_model_spec(model_name)
+        if device is None:
+            device = "cuda" if torch.cuda.is_available() else "cpu"
+        if dtype is None:
+            dtype = torch.bfloat16 if device != "cpu" else torch.float32
+        self.device = device
+        self.dtype = dtype
+
+        if model is not None:
+            self.model = model
+        else:
+            self.model = build_model(self.spec, dtype=dtype, device=device)
+            self.model.random_init(seed=weight_seed)
+
+        kv_heads = getattr(self.model, "num_kv_heads_local", None)
+        if kv_heads is None:
+            kv_heads = self.spec.num_kv_heads
+        if num_blocks is None:
+            if kv_memory_bytes is None:
+                if device.startswith("cuda"):
+                    free, _total = torch.cuda.mem_get_info()
+                    kv_memory_bytes = int(free * 0.8)
+                else:
+                    kv_memory_bytes = 256 << 20  # CPU tests
+            num_blocks = KVCachePool.blocks_for_memory(
+                kv_memory_bytes,
+                self.spec.num_layers,
+                kv_heads,
+                self.spec.head_dim,
+                block_size,
+                dtype_bytes=2 if dtype == torch.bfloat16 else 4,
+            )
+        self.kv_pool = KVCachePool(
+            self.spec.num_layers,
+            kv_heads,
+            self.spec.head_dim,
+            num_blocks,
+            block_size,
+            dtype=dtype,
+            device=device,
+        )
+        self.block_manager = BlockManager(num_blocks, block_size)
+        self.scheduler = Scheduler(self.block_manager, max_batch_tokens, max_running)
+        self.sampler = Sampler(device)
+        self.tokenizer = SyntheticTokenizer(self.spec.vocab_size)
+        self.template = ChatTemplate(self.tokenizer)
+
+        self._lock = threading.Lock()
+        self._work = threading.Condition(self._lock)
+        self._futures: dict = {}  # seq_id -> Future
+        self._thread: Optional[threading.Thread] = None
+        self._stop = False
+        # stats
+        self.steps = 0
+        self.tokens_sampled = 0
+        self.tokens_prefilled = 0
+
+    # ------------------------------------------------------------------
+    def submit_tokens(
+        self,
+        prompt_ids: list,
+        params: SamplingParams,
+        guide: Optional[object] = None,
+    ) -> concurrent.futures.Future:
+        if len(prompt_ids) + 16 > self.spec.max_position:
+            raise ContextLengthError(
+                f"prompt of {len(prompt_ids)} tokens exceeds max_position "
+                f"{self.spec.max_position}"
+            )
+        seq = Sequence(tokens=list(prompt_ids), params=params)
+        seq.guide = guide
+        if guide is not None:
+            forced = guide.initial_forced()
+            seq.tokens.extend(forced)
+            seq.output_tokens.extend(forced)
+        fut: concurrent.futures.Future = concurrent.futures.Future()
+        with self._work:
+            self._futures[seq.seq_id] = fut
+            self.scheduler.add(seq)
+            self._work.notify_all()
+        return fut
+
+    # ------------------------------------------------------------------
+    @torch.inference_mode()
+    def step(self) -> bool:
+        """One scheduling + forward + sample step. Returns True if it ran."""
+        with self._lock:
+            batch = self.scheduler.schedule()
+            if self.scheduler.stuck:
+                from dts_amd.llm.errors import BackendError
+
+                for seq in self.scheduler.stuck:
+                    fut = self._futures.pop(seq.seq_id, None)
+                    if fut is not None and not fut.done():
+                        fut.set_exception(
+                            BackendError(
+                                f"request of {len(seq.tokens)} tokens exceeds "
+                                f"the KV pool ({self.block_manager.num_blocks}"
+                                f"x{self.block_manager.block_size} tokens)"
+                            )
+                        )
+                self.scheduler.stuck.clear()
+        if batch is None:
+            return False
+        self.steps += 1
+        dev_batch = batch.to(self.device) if self.device != "cpu" else batch
+        logits = self.model.forward(dev_batch, self.kv_pool)
+        sampled_seqs = batch._sampled_seqs  # type: ignore[attr-defined]
+        tokens = self.sampler.sample(logits, sampled_seqs) if sampled_seqs else []
+        with self._lock:
+            self.scheduler.advance_computed(batch)
+            self.tokens_prefilled += batch.num_prefill_tokens
+            self.tokens_sampled += len(tokens)
+            for seq, tok in zip(sampled_seqs, tokens):
+                self._handle_sampled(seq, tok)
+        return True
+
+    def _handle_sampled(self, seq: Sequence, tok: int) -> None:
+        params = seq.params
+        seq.append_token(tok)
+        if seq.guide is not None:
+            forced = seq.guide.on_token(tok)
+            if forced:
+                seq.tokens.extend(forced)
+                seq.output_tokens.extend(forced)
+            if seq.guide.done():
+                self._finish(seq, "stop")
+            elif (
+                len(seq.tokens) + 8 >= self.spec.max_position
+                or seq.num_generated >= params.max_tokens
+            ):
+                self._finish(seq, "length")
+            return
+        if tok in self.template.stop_token_ids:
+            self._finish(seq, "stop")
+            return
+        if (
+            len(seq.tokens) + 8 >= self.spec.max_position
+            or seq.num_generated >= params.max_tokens
+        ):
+            # hard context cap: positions beyond the rope table are invalid
+            self._finish(seq, "length")
+
+    def _finish(self, seq: Sequence, reason: str) -> None:
+        self.scheduler.finish(seq, reason)
+        self.sampler.release(seq)
+        fut = self._futures.pop(seq.seq_id, None)
+        if fut is not None and not fut.done():
+            out_ids = list(seq.output_tokens)
+            # drop a trailing stop token from the text
+            if out_ids and out_ids[-1] in self.template.stop_token_ids:
+                out_ids = out_ids[:-1]
+            fut.set_result(
+                GenerationResult(
+                    token_ids=out_ids,
+                    text=self.tokenizer.decode(out_ids),
+                    finish_reason=reason,
+                    prompt_tokens=seq.num_prompt_tokens,
+                    completion_tokens=len(seq.output_tokens),
+                )
+            )
+
+    # ------------------------------------------------------------------
+    def run_until_idle(self) -> None:
+        """Synchronous drain (tests / bench warm-up)."""
+        while self.scheduler.has_work():
+            if not self.step():
+                break
+
+    def _loop(self) -> None:
+        while True:
+            with self._work:
+                while not self._stop and not self.scheduler.has_work():
+                    self._work.wait(timeout=0.1)
+                if self._stop:
+                    return
+            try:
+                self.step()
+            except Exception as e:  # noqa: BLE001
+                logger.exception("engine step failed: %s", e)
+                with self._lock:
+                    for seq in list(self.scheduler.running):
+                        fut = self._futures.pop(seq.seq_id, None)
+                        if fut and not fut.done():
+                            fut.set_exception(e)
+                        self.scheduler.abort(seq)
+
+    def start(self) -> None:
+        if self._thread is None:
+            self._stop = False
+            self._thread = threading.Thread(target=self._loop, daemon=True)
+            self._thread.start()
+
+    def stop(self) -> None:
+        with self._work:
+            self._stop = True
+            self._work.notify_all()
+        if self._thread is not None:
+            self._thread.join(timeout=5)
+            self._thread = None
+
+    @property
+    def cache_stats(self) -> dict:
+        bm = self.block_manager
+        return {
+            "cache_hit_tokens": bm.cache_hit_tokens,
+            "cache_miss_tokens": bm.cache_miss_tokens,
+            "free_blocks": bm.num_free(),
+            "steps": self.steps,
+            "tokens_sampled": self.tokens_sampled,
+            "tokens_prefilled": self.tokens_prefilled,
+        }
+
+
+def build_model(spec: ModelSpec, dtype, device, tp=None):
+    if spec.arch == "llama":
+        from dts_amd.models.llama import LlamaModel
+
+        return LlamaModel(spec, tp=tp, dtype=dtype, device=device)
+    if spec.arch == "gpt2":
+        from dts_amd.models.gpt2 import GPT2Model
+
+        return GPT2Model(spec, dtype=dtype, device=device)
+    if spec.arch == "mixtral":
+        from dts_amd.models.mixtral import MixtralModel
+
+        return MixtralModel(spec, tp=tp, dtype=dtype, device=device)
+    raise ValueError(f"unknown arch {spec.arch}")
+
+
+# ---------------------------------------------------------------------------
+# Backend seam
+# ---------------------------------------------------------------------------
+
+_N_RE = re.compile(r"exactly (\d+)")
+_TRAJ_RE = re.compile(r"--- Trajectory ([0-9a-fA-F-]+)")
+
+
+class LocalBackend:
+    """InferenceBackend over one or more ServingEngines (per-phase models,
+    ref engine.py:72-76 model resolution)."""
+
+    def __init__(self, engines: dict, default_model: str) -> None:
+        self.engines = engines
+        self.default_model = default_model
+        for e in engines.values():
+            e.start()
+
+    @classmethod
+    def single(cls, engine: ServingEngine, name: str = "local") -> "LocalBackend":
+        return cls({name: engine}, name)
+
+    def shutdown(self) -> None:
+        for e in self.engines.values():
+            e.stop()
+
+    def _engine(self, model: Optional[str]) -> ServingEngine:
+        if model and model in self.engines:
+            return self.engines[model]
+        return self.engines[self.default_model]
+
+    def _build_guide(self, engine: ServingEngine, messages: list):
+        system = messages[0].content if messages and messages[0].role == "system" else ""
+        system = system or ""
+        user = next((m.content for m in reversed(messages) if m.role == "user"), "") or ""
+        tok = engine.tokenizer
+        if "[dts:strategy]" in system:
+            m = _N_RE.search(user)
+            return structured.strategy_form(tok, int(m.group(1)) if m else 4)
+        if "[dts:intent]" in system:
+            m = _N_RE.search(user)
+            return structured.intent_form(tok, int(m.group(1)) if m else 3)
+        if "[dts:judge-absolute]" in system:
+            return structured.absolute_judge_form(tok)
+        if "[dts:judge-comparative]" in system:
+            ids = _TRAJ_RE.findall(user)
+            if ids:
+                return structured.comparative_judge_form(tok, ids)
+        return None
+
+    async def chat(
+        self,
+        messages: list,
+        params: SamplingParams,
+        model: Optional[str] = None,
+    ) -> Completion:
+        engine = self._engine(model)
+        prompt_ids = engine.template.render(messages)
+        guide = self._build_guide(engine, messages) if params.json_mode else None
+        fut = engine.submit_tokens(prompt_ids, params, guide=guide)
+        result: GenerationResult = await asyncio.wrap_future(fut)
+        return Completion(
+            message=Message.assistant(result.text),
+            usage=Usage(
+                prompt_tokens=result.prompt_tokens,
+                completion_tokens=result.completion_tokens,
+            ),
+            model=model or self.default_model,
+            finish_reason=result.finish_reason,
+        )

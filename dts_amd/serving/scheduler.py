@@ -1,0 +1,295 @@
+"""Token-level continuous-batching scheduler.
+
+The MI355X-native replacement for the reference's asyncio-semaphore
+"scheduler" (ref simulator.py:96): every concurrent search
+call — branch rollouts, judge passes, strategy/intent generation — feeds
+one flat token batch per engine step, mixing chunked prefill with decode.
+
+Unified step model: each scheduled sequence processes tokens
+[num_computed : num_computed + chunk]; a chunk of 1 on a fully-prefilled
+sequence IS decode. A sequence samples only on the step whose chunk reaches
+its last token. Prefix-cache hits advance num_computed before any compute
+(kv_cache.py). When the block pool runs dry, the youngest running sequence
+is preempted and recomputed later (its blocks usually still sit in the
+evictable cache, so recompute is mostly cache hits).
+"""
+
+from __future__ import annotations
+
+from collections import deque
+from typing import Optional
+
+import torch
+
+from dts_amd.serving.batch import ForwardBatch
+from dts_amd.serving.kv_cache import BlockManager
+from dts_amd.serving.sequence import Sequence, SeqStatus
+from dts_amd.utils.logging import logger
+
+
+class Scheduler:
+    def __init__(
+        self,
+        block_manager: BlockManager,
+        max_batch_tokens: int = 8192,
+        max_running: int = 256,
+    ) -> None:
+        self.bm = block_manager
+        self.block_size = block_manager.block_size
+        self.max_batch_tokens = max_batch_tokens
+        self.max_running = max_running
+        self.waiting: deque = deque()
+        self.running: list = []
+        self.stuck: list = []  # seqs that can NEVER fit (engine fails them)
+        self._arrival = 0
+
+    # ------------------------------------------------------------------
+    def add(self, seq: Sequence) -> None:
+        """Enqueue a new request; prefix matching happens at admission so a
+        request queued behind an identical in-flight prompt (e.g. the 2nd
+        and 3rd of 3 identical judge calls, ref evaluator.py:171-172) gets
+        the full prefill for free once the first one has run."""
+        seq.arrival_order = self._arrival
+        self._arrival += 1
+        seq.status = SeqStatus.WAITING
+        self.waiting.append(seq)
+
+    def _admit(self, seq: Sequence) -> bool:
+        """Prefix-match + allocate; returns False if blocked (OOM)."""
+        matched_blocks, n_matched = self.bm.match_prefix(seq.tokens)
+        if n_matched >= len(seq.tokens):
+            # whole prompt cached — recompute the last token for its logits
+            # (rewrites identical values into the shared slot)
+            n_matched = len(seq.tokens) - 1
+        seq.block_table = matched_blocks
+        seq.num_computed = n_matched
+        seq.num_hashed_blocks = len(matched_blocks)
+        h = 0
+        for b in matched_blocks:
+            bh = self.bm.blocks[b].hash
+            h = bh if bh is not None else h
+        seq.last_block_hash = h
+        if not self._ensure_blocks(seq, len(seq.tokens)):
+            self._release_blocks(seq)
+            seq.num_computed = 0
+            seq.num_hashed_blocks = 0
+            seq.last_block_hash = 0
+            return False
+        self.bm.cache_hit_tokens += n_matched
+        self.bm.cache_miss_tokens += max(0, len(seq.tokens) - n_matched)
+        return True
+
+    def _prompt_key(self, seq: Sequence) -> int:
+        return hash(tuple(seq.tokens[: seq.num_prompt_tokens]))
+
+    def has_work(self) -> bool:
+        return bool(self.waiting or self.running)
+
+    # ------------------------------------------------------------------
+    def _ensure_blocks(self, seq: Sequence, upto_tokens: int) -> bool:
+        """Grow seq.block_table to cover `upto_tokens` tokens; False if OOM."""
+        need = (upto_tokens + self.block_size - 1) // self.block_size
+        while len(seq.block_table) < need:
+            if self.bm.num_free() == 0:
+                return False
+            seq.block_table.append(self.bm.allocate_fresh())
+        return True
+
+    def _preempt_youngest(self, exclude: Sequence) -> bool:
+        candidates = [s for s in self.running if s is not exclude and not s.in_flight]
+        if not candidates:
+            return False
+        victim = max(candidates, key=lambda s: s.arrival_order)
+        logger.warning("preempting seq %d (KV pressure)", victim.seq_id)
+        self._release_blocks(victim)
+        victim.num_computed = 0
+        victim.num_hashed_blocks = 0
+        victim.last_block_hash = 0
+        victim.status = SeqStatus.WAITING
+        self.running.remove(victim)
+        self.waiting.appendleft(victim)
+        return True
+
+    def _release_blocks(self, seq: Sequence) -> None:
+        for b in seq.block_table:
+            self.bm.free_block(b)
+        seq.block_table = []
+
+    # ------------------------------------------------------------------
+    def schedule(self) -> Optional[ForwardBatch]:
+        # admit — hold back requests whose exact prompt is already being
+        # prefilled by a running sequence (duplicate-prefill dedup)
+        inflight_prefills = {
+            self._prompt_key(s)
+            for s in self.running
+            if s.num_computed < s.num_prompt_tokens
+        }
+        held: list = []
+        while self.waiting and len(self.running) < self.max_running:
+            seq = self.waiting.popleft()
+            key = self._prompt_key(seq)
+            if key in inflight_prefills:
+                held.append(seq)
+                continue
+            if not self._admit(seq):
+                held.append(seq)
+                break
+            seq.status = SeqStatus.RUNNING
+            self.running.append(seq)
+            if seq.num_computed < seq.num_prompt_tokens:
+                inflight_prefills.add(key)
+        for seq in reversed(held):
+            self.waiting.appendleft(seq)
+
+        if not self.running:
+            return None
+
+        budget = self.max_batch_tokens
+        prefills: list = []  # (seq, chunk)
+        decodes: list = []
+        for seq in list(self.running):
+            if seq.in_flight:
+                continue
+            remaining = len(seq.tokens) - seq.num_computed
+            if remaining <= 0:
+                continue
+            if not self._ensure_blocks(seq, len(seq.tokens)):
+                if not self._preempt_youngest(exclude=seq):
+                    if len(self.running) == 1 and not prefills and not decodes:
+                        # alone and still can't fit: the request exceeds the
+                        # whole pool — fail it loudly instead of stalling
+                        self._release_blocks(seq)
+                        self.running.remove(seq)
+                        self.stuck.append(seq)
+                    continue
+                if not self._ensure_blocks(seq, len(seq.tokens)):
+                    continue
+            chunk = min(remaining, budget)
+            if chunk <= 0:
+                continue
+            budget -= chunk
+            if remaining == 1 and seq.output_tokens:
+                decodes.append(seq)
+            elif chunk == 1:
+                decodes.append(seq)
+            else:
+                prefills.append((seq, chunk))
+            if budget <= 0:
+                break
+
+        if not prefills and not decodes:
+            return None
+        return self._build_batch(prefills, decodes)
+
+    def _build_batch(self, prefills: list, decodes: list) -> ForwardBatch:
+        bs = self.block_size
+        token_ids: list = []
+        positions: list = []
+        slots: list = []
+        sample_indices: list = []
+        sampled_seqs: list = []
+
+        cu_q = [0]
+        pf_tables: list = []
+        pf_kv_lens: list = []
+        for seq, chunk in prefills:
+            start = seq.num_computed
+            end = start + chunk
+            for p in range(start, end):
+                token_ids.append(seq.tokens[p])
+                positions.append(p)
+                slots.append(seq.block_table[p // bs] * bs + p % bs)
+            cu_q.append(cu_q[-1] + chunk)
+            pf_tables.append(list(seq.block_table))
+            pf_kv_lens.append(end)
+            if end == len(seq.tokens):
+                sample_indices.append(len(token_ids) - 1)
+                sampled_seqs.append(seq)
+            seq._sched_chunk = chunk  # type: ignore[attr-defined]
+
+        num_prefill_tokens = len(token_ids)
+
+        dc_tables: list = []
+        dc_kv_lens: list = []
+        for seq in decodes:
+            p = seq.num_computed
+            token_ids.append(seq.tokens[p])
+            positions.append(p)
+            slots.append(seq.block_table[p // bs] * bs + p % bs)
+            dc_tables.append(list(seq.block_table))
+            dc_kv_lens.append(p + 1)
+            sample_indices.append(len(token_ids) - 1)
+            sampled_seqs.append(seq)
+            seq._sched_chunk = 1  # type: ignore[attr-defined]
+
+        def pad_tables(tables: list) -> Optional[torch.Tensor]:
+            if not tables:
+                return None
+            m = max(len(t) for t in tables)
+            return torch.tensor(
+                [t + [0] * (m - len(t)) for t in tables], dtype=torch.int32
+            )
+
+        batch = ForwardBatch(
+            token_ids=torch.tensor(token_ids, dtype=torch.long),
+            positions=torch.tensor(positions, dtype=torch.long),
+            slot_mapping=torch.tensor(slots, dtype=torch.long),
+            num_prefill_seqs=len(prefills),
+            num_prefill_tokens=num_prefill_tokens,
+            cu_q=torch.tensor(cu_q, dtype=torch.int32) if prefills else None,
+            prefill_block_tables=pad_tables(pf_tables),
+            prefill_kv_lens=(
+                torch.tensor(pf_kv_lens, dtype=torch.int32) if prefills else None
+            ),
+            num_decode_seqs=len(decodes),
+            decode_block_tables=pad_tables(dc_tables),
+            decode_kv_lens=(
+                torch.tensor(dc_kv_lens, dtype=torch.int32) if decodes else None
+            ),
+            sample_indices=torch.tensor(sample_indices, dtype=torch.long),
+        )
+        scheduled = [s for s, _ in prefills] + decodes
+        for s in scheduled:
+            s.in_flight = True
+        batch._scheduled = scheduled  # type: ignore[attr-defined]
+        batch._sampled_seqs = sampled_seqs  # type: ignore[attr-defined]
+        return batch
+
+    # ------------------------------------------------------------------
+    def advance_computed(self, batch: ForwardBatch) -> None:
+        """After a forward: bump num_computed and register full blocks."""
+        for seq in batch._scheduled:  # type: ignore[attr-defined]
+            chunk = getattr(seq, "_sched_chunk", 0)
+            seq.num_computed += chunk
+            seq.in_flight = False
+            self._register_full_blocks(seq)
+
+    def _register_full_blocks(self, seq: Sequence) -> None:
+        bs = self.block_size
+        while (seq.num_hashed_blocks + 1) * bs <= seq.num_computed:
+            b_idx = seq.num_hashed_blocks
+            chunk = tuple(seq.tokens[b_idx * bs : (b_idx + 1) * bs])
+            block_id = seq.block_table[b_idx]
+            block = self.bm.blocks[block_id]
+            if block.hash is None:
+                seq.last_block_hash = self.bm.register_full_block(
+                    block_id, seq.last_block_hash, chunk
+                )
+            else:
+                seq.last_block_hash = block.hash
+            seq.num_hashed_blocks += 1
+
+    def finish(self, seq: Sequence, reason: str) -> None:
+        seq.status = SeqStatus.FINISHED
+        seq.finish_reason = reason
+        self._release_blocks(seq)
+        if seq in self.running:
+            self.running.remove(seq)
+
+    def abort(self, seq: Sequence) -> None:
+        seq.status = SeqStatus.ABORTED
+        self._release_blocks(seq)
+        if seq in self.running:
+            self.running.remove(seq)
+        if seq in self.waiting:
+            self.waiting.remove(seq)

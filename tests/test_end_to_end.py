@@ -1,0 +1,85 @@
+"""End-to-end: the full DTS search running on the local serving engine.
+
+This is BASELINE.json config 1 ("plumbing, no GPU") at test scale: every
+LLM call — strategy JSON, rollout turns, judges — is served by the
+continuous-batching engine with a random-init tiny model, constrained
+decoding producing the structured outputs.
+"""
+
+import asyncio
+
+import pytest
+import torch
+
+from dts_amd.llm import LLM
+from dts_amd.search import DTSConfig, DTSEngine
+from dts_amd.search.config import GenerationBudget
+from dts_amd.serving import LocalBackend, ServingEngine
+
+
+@pytest.fixture(scope="module")
+def local_llm():
+    engine = ServingEngine(
+        model_name="llama-tiny",
+        device="cpu",
+        dtype=torch.float32,
+        num_blocks=4096,
+        block_size=16,
+        max_batch_tokens=2048,
+        weight_seed=11,
+    )
+    backend = LocalBackend.single(engine, name="llama-tiny")
+    yield LLM(backend, default_model="llama-tiny"), engine
+    backend.shutdown()
+
+
+def small_budget():
+    return GenerationBudget(
+        strategy=2048, intent=2048, rephrase=24, user=24, assistant=24, judge=4096
+    )
+
+
+class TestEndToEnd:
+    def test_absolute_search(self, local_llm):
+        llm, engine = local_llm
+        cfg = DTSConfig(
+            goal="Teach binary search",
+            first_message="How does binary search work?",
+            init_branches=2,
+            turns_per_branch=1,
+            scoring_mode="absolute",
+            prune_threshold=0.0,
+            seed=5,
+            budget=small_budget(),
+        )
+        result = asyncio.run(DTSEngine(llm, cfg).run(rounds=1))
+        assert len(result.all_nodes) == 3
+        branches = [n for n in result.all_nodes if n.strategy is not None]
+        assert all(len(n.stats.judge_scores) == 3 for n in branches)
+        # 1 strategy + 2 branches x (1 user + 1 assistant) + 2 x 3 judges
+        assert result.token_usage["totals"]["total_requests"] >= 1 + 2 * 2 + 2 * 3
+        # prefix cache must have been exercised by shared prompts
+        assert engine.block_manager.cache_hit_tokens > 0
+
+    def test_comparative_search_with_forking(self, local_llm):
+        llm, engine = local_llm
+        cfg = DTSConfig(
+            goal="Sell a fountain pen",
+            first_message="Why would I need a fountain pen?",
+            init_branches=2,
+            turns_per_branch=1,
+            user_intents_per_branch=2,
+            user_variability=True,
+            scoring_mode="comparative",
+            prune_threshold=0.0,
+            seed=9,
+            budget=small_budget(),
+        )
+        result = asyncio.run(DTSEngine(llm, cfg).run(rounds=1))
+        forked = [n for n in result.all_nodes if n.user_intent is not None]
+        assert len(forked) == 4
+        scored = [n for n in forked if n.stats.judge_scores]
+        assert len(scored) == 4
+        # comparative scores follow the forced-ranking schedule shape
+        for n in scored:
+            assert 0.0 <= n.stats.aggregated_score <= 9.9
