@@ -1,0 +1,506 @@
+// Paged attention kernels for MI355X (gfx950, CDNA4).
+//
+// Decode (one query token per sequence): memory-bound streaming of the
+// paged KV; one 4-wave workgroup per (sequence, kv-head), the G q-heads of
+// the GQA group computed together so every K/V byte is read once for G
+// dot products. Waves split the KV pages round-robin and combine their
+// online-softmax partials through LDS (flash-decoding style).
+//
+// Prefill (chunked, causal, attends cached prefix + own chunk): MFMA
+// (v_mfma_f32_16x16x32_bf16) flash-style kernel; one 4-wave workgroup per
+// (sequence, 64-row q-tile, kv-head) where a row is a (position, q-head)
+// pair; K/V pages staged in LDS (padded rows against bank conflicts),
+// online softmax with per-row running max/sum. Correctness-first
+// structure; the optimization ladder (XOR-swizzle, tr_b16 V reads,
+// 8-phase pipeline) lands on top after rocprof baselines.
+//
+// Semantics match dts_amd/ops/torch_ref.py attn_*_paged.
+
+#include "common.h"
+#include <torch/extension.h>
+#include <c10/hip/HIPStream.h>
+
+using bf16x8 = __attribute__((ext_vector_type(8))) short;
+using f32x4 = __attribute__((ext_vector_type(4))) float;
+
+// ---------------------------------------------------------------------------
+// Decode
+// ---------------------------------------------------------------------------
+
+// G = q-heads per kv-head, D = head dim. Block size BS = 16 tokens.
+template <int G, int D>
+__global__ void __launch_bounds__(256)
+attn_decode_kernel(short* __restrict__ out,          // [B, Hq, D]
+                   const short* __restrict__ q,      // [B, Hq, D]
+                   const short* __restrict__ kcache, // [N, Hkv, BS, D]
+                   const short* __restrict__ vcache,
+                   const int* __restrict__ block_tables, // [B, max_blocks]
+                   const int* __restrict__ kv_lens,      // [B]
+                   int max_blocks, int Hkv, float scale) {
+  constexpr int BS = 16;
+  constexpr int NWAVE = 4;
+  constexpr int LPK = 4;            // lanes per key
+  constexpr int DPL = D / LPK;      // dims per lane in K phase
+  const int seq = blockIdx.x;
+  const int kvh = blockIdx.y;
+  const int Hq = Hkv * G;
+  const int wave = threadIdx.x / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int kv_len = kv_lens[seq];
+  const int n_pages = (kv_len + BS - 1) / BS;
+
+  // LDS: per-wave score buffer + combine buffers
+  __shared__ float s_scores[NWAVE][G][BS];
+  __shared__ float s_m[NWAVE][G], s_l[NWAVE][G];
+  __shared__ float s_o[NWAVE][G][D];
+
+  // q fragment: lane covers key-group dims [c*DPL, (c+1)*DPL) with c=lane&3.
+  // Kept PACKED (bf16 pairs in u32) to stay under the VGPR occupancy cliff
+  // (G=4: 64 regs of q instead of 128 fp32).
+  const int c = lane & (LPK - 1);
+  const int key_of_lane = lane / LPK; // 0..15
+  int qp_[G][DPL / 2];
+#pragma unroll
+  for (int g = 0; g < G; ++g) {
+    const short* qp = q + (((long)seq * Hq) + kvh * G + g) * D + c * DPL;
+#pragma unroll
+    for (int i = 0; i < DPL / 2; ++i) qp_[g][i] = ((const int*)qp)[i];
+  }
+
+  float m[G], lsum[G], o[G][2]; // o: lane owns dims {2*lane, 2*lane+1}
+#pragma unroll
+  for (int g = 0; g < G; ++g) {
+    m[g] = -1e30f;
+    lsum[g] = 0.f;
+    o[g][0] = o[g][1] = 0.f;
+  }
+
+  const int* bt = block_tables + (long)seq * max_blocks;
+
+  for (int page = wave; page < n_pages; page += NWAVE) {
+    const long blk = bt[page];
+    const short* kbase =
+        kcache + ((blk * Hkv + kvh) * BS) * (long)D;
+    const short* vbase =
+        vcache + ((blk * Hkv + kvh) * BS) * (long)D;
+    const int valid = min(BS, kv_len - page * BS);
+
+    // ---- K phase: scores for the 16 keys of this page
+    float partial[G];
+#pragma unroll
+    for (int g = 0; g < G; ++g) partial[g] = 0.f;
+    if (key_of_lane < valid) {
+      const short* kp = kbase + key_of_lane * D + c * DPL;
+#pragma unroll
+      for (int i = 0; i < DPL / 2; ++i) {
+        int kw = ((const int*)kp)[i];
+        float k0 = bf2f((short)(kw & 0xffff));
+        float k1 = bf2f((short)((kw >> 16) & 0xffff));
+#pragma unroll
+        for (int g = 0; g < G; ++g) {
+          float q0 = bf2f((short)(qp_[g][i] & 0xffff));
+          float q1 = bf2f((short)((qp_[g][i] >> 16) & 0xffff));
+          partial[g] += q0 * k0 + q1 * k1;
+        }
+      }
+    }
+    // reduce over the 4 lanes of the key group
+#pragma unroll
+    for (int g = 0; g < G; ++g) {
+      partial[g] += __shfl_xor(partial[g], 1, 64);
+      partial[g] += __shfl_xor(partial[g], 2, 64);
+      if (c == 0)
+        s_scores[wave][g][key_of_lane] =
+            (key_of_lane < valid) ? partial[g] * scale : -1e30f;
+    }
+    __builtin_amdgcn_s_waitcnt(0); // lgkm for LDS writes within wave
+    __builtin_amdgcn_wave_barrier();
+
+    // ---- softmax update + V phase; lane owns output dims {2l, 2l+1}
+#pragma unroll
+    for (int g = 0; g < G; ++g) {
+      float tmax = -1e30f;
+#pragma unroll
+      for (int j = 0; j < BS; ++j) tmax = fmaxf(tmax, s_scores[wave][g][j]);
+      float m_new = fmaxf(m[g], tmax);
+      if (m_new > -1e30f) {
+        float alpha = __expf(m[g] - m_new);
+        o[g][0] *= alpha;
+        o[g][1] *= alpha;
+        lsum[g] *= alpha;
+        m[g] = m_new;
+        float p[BS];
+        float psum = 0.f;
+#pragma unroll
+        for (int j = 0; j < BS; ++j) {
+          p[j] = (j < valid) ? __expf(s_scores[wave][g][j] - m_new) : 0.f;
+          psum += p[j];
+        }
+        lsum[g] += psum;
+#pragma unroll
+        for (int j = 0; j < BS; ++j) {
+          if (j < valid) {
+            int vw = *(const int*)(vbase + j * D + 2 * lane);
+            o[g][0] += p[j] * bf2f((short)(vw & 0xffff));
+            o[g][1] += p[j] * bf2f((short)((vw >> 16) & 0xffff));
+          }
+        }
+      }
+    }
+    __builtin_amdgcn_wave_barrier();
+  }
+
+  // ---- cross-wave combine via LDS
+#pragma unroll
+  for (int g = 0; g < G; ++g) {
+    if (lane == 0) {
+      s_m[wave][g] = m[g];
+      s_l[wave][g] = lsum[g];
+    }
+    s_o[wave][g][2 * lane] = o[g][0];
+    s_o[wave][g][2 * lane + 1] = o[g][1];
+  }
+  __syncthreads();
+
+  // threads 0..G*D/2: each handles one (g, dim-pair)
+  const int tid = threadIdx.x;
+  for (int gd = tid; gd < G * (D / 2); gd += blockDim.x) {
+    const int g = gd / (D / 2);
+    const int d0 = (gd % (D / 2)) * 2;
+    float m_star = -1e30f;
+#pragma unroll
+    for (int w = 0; w < NWAVE; ++w) m_star = fmaxf(m_star, s_m[w][g]);
+    float l_star = 0.f, acc0 = 0.f, acc1 = 0.f;
+#pragma unroll
+    for (int w = 0; w < NWAVE; ++w) {
+      float f = (s_m[w][g] > -1e30f) ? __expf(s_m[w][g] - m_star) : 0.f;
+      l_star += s_l[w][g] * f;
+      acc0 += s_o[w][g][d0] * f;
+      acc1 += s_o[w][g][d0 + 1] * f;
+    }
+    float inv = (l_star > 0.f) ? 1.f / l_star : 0.f;
+    short* op = out + (((long)seq * Hq) + kvh * G + g) * D + d0;
+    op[0] = f2bf(acc0 * inv);
+    op[1] = f2bf(acc1 * inv);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Prefill (MFMA)
+// ---------------------------------------------------------------------------
+
+// Rows per workgroup: 64 = (64/G) positions x G heads. Each wave owns 16
+// rows. KV step = 32 keys (2 pages). D = 128 only (Llama/Mixtral).
+template <int G>
+__global__ void __launch_bounds__(256)
+attn_prefill_kernel(short* __restrict__ out,      // [Tq, Hq, D]
+                    const short* __restrict__ q,  // [Tq, Hq, D]
+                    const int* __restrict__ cu_q, // [P+1]
+                    const long* __restrict__ q_pos, // [Tq]
+                    const short* __restrict__ kcache,
+                    const short* __restrict__ vcache,
+                    const int* __restrict__ block_tables,
+                    const int* __restrict__ kv_lens, int max_blocks, int Hkv,
+                    float scale) {
+  constexpr int BS = 16;
+  constexpr int KSTEP = 32;
+  constexpr int D = 128;
+  constexpr int LDS_PAD = 8;          // bf16 elements of row padding
+  constexpr int LDK = D + LDS_PAD;    // LDS row stride (elements)
+  constexpr int ROWS = 64;            // rows per workgroup
+  const int POS_PER_WG = ROWS / G;
+
+  const int seq = blockIdx.y;
+  const int kvh = blockIdx.z;
+  const int Hq = Hkv * G;
+  const int q_start = cu_q[seq];
+  const int q_len = cu_q[seq + 1] - q_start;
+  const int tile = blockIdx.x;
+  if (tile * POS_PER_WG >= q_len) return;
+
+  const int wave = threadIdx.x / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int kv_len = kv_lens[seq];
+  const int* bt = block_tables + (long)seq * max_blocks;
+
+  __shared__ short k_lds[KSTEP][LDK];
+  __shared__ short v_lds[KSTEP][LDK];
+  __shared__ float s_scores[4][16][KSTEP];
+  __shared__ float s_alpha[4][16];
+  __shared__ float s_rowl[4][16];
+  __shared__ short p_lds[4][16][KSTEP];
+
+  // ---- per-wave row bookkeeping: row_global = wave*16 + r
+  // row -> (position, head)
+  long my_pos[4];   // absolute position of rows owned by this lane group
+  // Q A-fragments: 4 k-chunks of 32, 8 bf16 each
+  short a_frag[4][8];
+  {
+    const int r = lane & 15;            // A row = lane & 15
+    const int row_global = wave * 16 + r;
+    const int pos_local = tile * POS_PER_WG + row_global / G;
+    const int head = row_global % G;
+    const bool valid_row = pos_local < q_len;
+    const int tok = q_start + (valid_row ? pos_local : 0);
+    const short* qp = q + ((long)tok * Hq + kvh * G + head) * D;
+#pragma unroll
+    for (int kc = 0; kc < 4; ++kc) {
+      const int kbase = kc * 32 + (lane >> 4) * 8;
+      bf16x8 v8 = *(const bf16x8*)(qp + kbase);
+#pragma unroll
+      for (int i = 0; i < 8; ++i)
+        a_frag[kc][i] = valid_row ? v8[i] : (short)0;
+    }
+  }
+  // rows this lane owns in the softmax phase: row = lane / 4 (4 lanes/row)
+  const int sm_row = lane / 4;
+  const int sm_sub = lane & 3;  // which 8-key span this lane scans
+  const long sm_pos =
+      tile * POS_PER_WG + (wave * 16 + sm_row) / G;  // local position
+  const bool sm_valid = sm_pos < q_len;
+  const long sm_abs_pos = sm_valid ? q_pos[q_start + sm_pos] : -1;
+
+  // O accumulator: 8 column tiles of C-frag f32x4
+  f32x4 o_acc[8];
+#pragma unroll
+  for (int ct = 0; ct < 8; ++ct) o_acc[ct] = {0.f, 0.f, 0.f, 0.f};
+  float run_m = -1e30f, run_l = 0.f;  // per (lane, sm) bookkeeping below
+  // running m/l per row are tracked by the 4-lane row group (same value
+  // in all 4 lanes; reduced via shfl)
+
+  // max absolute position in this workgroup bounds the causal KV range
+  // (rows of this WG cover positions tile*PPW .. tile*PPW+PPW-1)
+  const int last_local_pos = min(q_len, tile * POS_PER_WG + POS_PER_WG) - 1;
+  const long last_abs_pos = q_pos[q_start + last_local_pos];
+  const int kv_hi = min((long)kv_len, last_abs_pos + 1);
+
+  for (int kv_base = 0; kv_base < kv_hi; kv_base += KSTEP) {
+    // ---- cooperative K/V tile load (2 pages), 16B per thread per pass
+    {
+      const int elems = KSTEP * D;          // 32*128 bf16
+      // thread t loads 8 bf16 at flat index t*8 (+ stride)
+      for (int base = threadIdx.x * 8; base < elems; base += 256 * 8) {
+        const int key = base / D;
+        const int d0 = base % D;
+        const int page = kv_base / BS + key / BS;
+        const int koff = key % BS;
+        const int kglob = kv_base + key;
+        if (kglob < kv_len) {
+          const long blk = bt[page];
+          const short* kp =
+              kcache + ((blk * Hkv + kvh) * BS + koff) * (long)D + d0;
+          const short* vp =
+              vcache + ((blk * Hkv + kvh) * BS + koff) * (long)D + d0;
+          *(bf16x8*)(&k_lds[key][d0]) = *(const bf16x8*)kp;
+          *(bf16x8*)(&v_lds[key][d0]) = *(const bf16x8*)vp;
+        } else {
+          bf16x8 z = {0, 0, 0, 0, 0, 0, 0, 0};
+          *(bf16x8*)(&k_lds[key][d0]) = z;
+          *(bf16x8*)(&v_lds[key][d0]) = z;
+        }
+      }
+    }
+    __syncthreads();
+
+    // ---- QK^T: two 16x16 score tiles over 4 k-chunks
+#pragma unroll
+    for (int st = 0; st < 2; ++st) {
+      f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+      for (int kc = 0; kc < 4; ++kc) {
+        // B-frag: col = lane&15 -> key = st*16 + (lane&15),
+        //         k  = kc*32 + (lane>>4)*8 + i  (8 consecutive dims)
+        const int key = st * 16 + (lane & 15);
+        const int d0 = kc * 32 + (lane >> 4) * 8;
+        bf16x8 b8 = *(const bf16x8*)(&k_lds[key][d0]);
+        acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            *(bf16x8*)a_frag[kc], b8, acc, 0, 0, 0);
+      }
+      // scatter C-frag to LDS scores: row=(lane>>4)*4+i, col=lane&15
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        const int row = (lane >> 4) * 4 + i;
+        s_scores[wave][row][st * 16 + (lane & 15)] = acc[i] * scale;
+      }
+    }
+    __builtin_amdgcn_s_waitcnt(0);
+    __builtin_amdgcn_wave_barrier();
+
+    // ---- online softmax: 4 lanes per row, each scans 8 keys
+    {
+      float tmax = -1e30f;
+      float sc[8];
+#pragma unroll
+      for (int jj = 0; jj < 8; ++jj) {
+        const int j = sm_sub * 8 + jj;
+        const long key_abs = kv_base + j;
+        float s = s_scores[wave][sm_row][j];
+        const bool ok = sm_valid && key_abs <= sm_abs_pos &&
+                        key_abs < (long)kv_len;
+        sc[jj] = ok ? s : -1e30f;
+        tmax = fmaxf(tmax, sc[jj]);
+      }
+      tmax = fmaxf(tmax, __shfl_xor(tmax, 1, 64));
+      tmax = fmaxf(tmax, __shfl_xor(tmax, 2, 64));
+      float m_new = fmaxf(run_m, tmax);
+      float alpha, rowsum = 0.f;
+      if (m_new > -1e30f) {
+        alpha = (run_m > -1e30f) ? __expf(run_m - m_new) : 0.f;
+#pragma unroll
+        for (int jj = 0; jj < 8; ++jj) {
+          float p = (sc[jj] > -1e30f) ? __expf(sc[jj] - m_new) : 0.f;
+          p_lds[wave][sm_row][sm_sub * 8 + jj] = f2bf(p);
+          rowsum += p;
+        }
+      } else {
+        alpha = 1.f;
+#pragma unroll
+        for (int jj = 0; jj < 8; ++jj)
+          p_lds[wave][sm_row][sm_sub * 8 + jj] = 0;
+      }
+      rowsum += __shfl_xor(rowsum, 1, 64);
+      rowsum += __shfl_xor(rowsum, 2, 64);
+      run_l = run_l * alpha + rowsum;
+      run_m = m_new;
+      if (sm_sub == 0) {
+        s_alpha[wave][sm_row] = alpha;
+        s_rowl[wave][sm_row] = run_l;
+      }
+    }
+    __builtin_amdgcn_s_waitcnt(0);
+    __builtin_amdgcn_wave_barrier();
+
+    // ---- rescale O and accumulate P*V
+    {
+      // rescale: C row = (lane>>4)*4 + i
+      float al[4];
+#pragma unroll
+      for (int i = 0; i < 4; ++i) al[i] = s_alpha[wave][(lane >> 4) * 4 + i];
+      // P A-frag: row = lane&15, k(key) = (lane>>4)*8 + i
+      bf16x8 pa0, pa1;
+      {
+        const int row = lane & 15;
+        const int k0 = (lane >> 4) * 8;
+#pragma unroll
+        for (int i = 0; i < 8; ++i) pa0[i] = p_lds[wave][row][k0 + i];
+        // second half keys 16..31 wait: KSTEP=32 = one mfma k=32 — single
+        // A-frag covers keys (lane>>4)*8.. only 32 total: k index range is
+        // 0..31 -> (lane>>4)*8+i covers 0..31. pa1 unused.
+        pa1 = pa0;
+      }
+#pragma unroll
+      for (int ct = 0; ct < 8; ++ct) {
+#pragma unroll
+        for (int i = 0; i < 4; ++i) o_acc[ct][i] *= al[i];
+        // V B-frag: col = lane&15 -> dim = ct*16 + (lane&15),
+        //           k(key) = (lane>>4)*8 + i -> strided column read
+        bf16x8 vb;
+        const int dim = ct * 16 + (lane & 15);
+        const int kk0 = (lane >> 4) * 8;
+#pragma unroll
+        for (int i = 0; i < 8; ++i) vb[i] = v_lds[kk0 + i][dim];
+        o_acc[ct] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pa0, vb,
+                                                            o_acc[ct], 0, 0, 0);
+      }
+    }
+    __syncthreads();
+  }
+
+  // ---- epilogue: normalize + store
+  {
+    float invl[4];
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      float l = s_rowl[wave][(lane >> 4) * 4 + i];
+      invl[i] = (l > 0.f) ? 1.f / l : 0.f;
+    }
+#pragma unroll
+    for (int ct = 0; ct < 8; ++ct) {
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        const int row = (lane >> 4) * 4 + i;
+        const int row_global = wave * 16 + row;
+        const int pos_local = tile * POS_PER_WG + row_global / G;
+        if (pos_local >= q_len) continue;
+        const int head = row_global % G;
+        const int tok = q_start + pos_local;
+        const int dim = ct * 16 + (lane & 15);
+        out[((long)tok * Hq + kvh * G + head) * D + dim] =
+            f2bf(o_acc[ct][i] * invl[i]);
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Host wrappers
+// ---------------------------------------------------------------------------
+
+void attn_decode_paged(torch::Tensor out, torch::Tensor q,
+                       torch::Tensor kcache, torch::Tensor vcache,
+                       torch::Tensor block_tables, torch::Tensor kv_lens,
+                       double scale) {
+  const int B = q.size(0), Hq = q.size(1), D = q.size(2);
+  const int Hkv = kcache.size(1);
+  const int G = Hq / Hkv;
+  const int max_blocks = block_tables.size(1);
+  TORCH_CHECK(kcache.size(2) == 16, "block_size must be 16");
+  TORCH_CHECK(block_tables.scalar_type() == torch::kInt32);
+  dim3 grid(B, Hkv);
+  dim3 block(256);
+  auto stream = c10::hip::getCurrentHIPStream();
+#define DECODE_CASE(g, d)                                                    \
+  hipLaunchKernelGGL((attn_decode_kernel<g, d>), grid, block, 0, stream,     \
+                     (short*)out.data_ptr(), (const short*)q.data_ptr(),     \
+                     (const short*)kcache.data_ptr(),                        \
+                     (const short*)vcache.data_ptr(),                        \
+                     (const int*)block_tables.data_ptr(),                    \
+                     (const int*)kv_lens.data_ptr(), max_blocks, Hkv,        \
+                     (float)scale)
+  if (D == 128 && G == 4) DECODE_CASE(4, 128);
+  else if (D == 128 && G == 8) DECODE_CASE(8, 128);
+  else if (D == 128 && G == 1) DECODE_CASE(1, 128);
+  else if (D == 64 && G == 1) DECODE_CASE(1, 64);
+  else TORCH_CHECK(false, "unsupported decode attn shape D=", D, " G=", G);
+#undef DECODE_CASE
+  HIP_CHECK_LAST();
+}
+
+void attn_prefill_paged(torch::Tensor out, torch::Tensor q, torch::Tensor cu_q,
+                        torch::Tensor q_pos, torch::Tensor kcache,
+                        torch::Tensor vcache, torch::Tensor block_tables,
+                        torch::Tensor kv_lens, double scale) {
+  const int Hq = q.size(1), D = q.size(2);
+  const int Hkv = kcache.size(1);
+  const int G = Hq / Hkv;
+  const int P = cu_q.size(0) - 1;
+  const int max_blocks = block_tables.size(1);
+  TORCH_CHECK(D == 128, "prefill kernel supports head_dim 128");
+  TORCH_CHECK(kcache.size(2) == 16, "block_size must be 16");
+  TORCH_CHECK(cu_q.scalar_type() == torch::kInt32);
+  // worst-case tiles per seq: computed on host from q sizes is per-layer
+  // overhead; use Tq (total) as bound and let tiles beyond q_len exit.
+  const long Tq = q.size(0);
+  const int pos_per_wg = 64 / G;
+  const int max_tiles = (int)((Tq + pos_per_wg - 1) / pos_per_wg);
+  dim3 grid(max_tiles, P, Hkv);
+  dim3 block(256);
+  auto stream = c10::hip::getCurrentHIPStream();
+#define PREFILL_CASE(g)                                                       \
+  hipLaunchKernelGGL((attn_prefill_kernel<g>), grid, block, 0, stream,        \
+                     (short*)out.data_ptr(), (const short*)q.data_ptr(),      \
+                     (const int*)cu_q.data_ptr(),                             \
+                     (const long*)q_pos.data_ptr(),                           \
+                     (const short*)kcache.data_ptr(),                         \
+                     (const short*)vcache.data_ptr(),                         \
+                     (const int*)block_tables.data_ptr(),                     \
+                     (const int*)kv_lens.data_ptr(), max_blocks, Hkv,         \
+                     (float)scale)
+  if (G == 4) PREFILL_CASE(4);
+  else if (G == 8) PREFILL_CASE(8);
+  else if (G == 1) PREFILL_CASE(1);
+  else if (G == 2) PREFILL_CASE(2);
+  else TORCH_CHECK(false, "unsupported prefill GQA ratio G=", G);
+#undef PREFILL_CASE
+  HIP_CHECK_LAST();
+}

@@ -1,0 +1,39 @@
+// Python bindings for the dts_amd MI355X kernels.
+#include <torch/extension.h>
+
+void rmsnorm(torch::Tensor out, torch::Tensor x, torch::Tensor w, double eps);
+void fused_add_rmsnorm(torch::Tensor x, torch::Tensor residual,
+                       torch::Tensor w, double eps);
+void silu_mul(torch::Tensor out, torch::Tensor in);
+void rope_kv_append(torch::Tensor q, torch::Tensor k, torch::Tensor v,
+                    torch::Tensor positions, torch::Tensor cos_t,
+                    torch::Tensor sin_t, torch::Tensor k_cache,
+                    torch::Tensor v_cache, torch::Tensor slots);
+void kv_append(torch::Tensor k, torch::Tensor v, torch::Tensor k_cache,
+               torch::Tensor v_cache, torch::Tensor slots);
+void attn_decode_paged(torch::Tensor out, torch::Tensor q,
+                       torch::Tensor kcache, torch::Tensor vcache,
+                       torch::Tensor block_tables, torch::Tensor kv_lens,
+                       double scale);
+void attn_prefill_paged(torch::Tensor out, torch::Tensor q, torch::Tensor cu_q,
+                        torch::Tensor q_pos, torch::Tensor kcache,
+                        torch::Tensor vcache, torch::Tensor block_tables,
+                        torch::Tensor kv_lens, double scale);
+void top_p_sample(torch::Tensor out, torch::Tensor logits, torch::Tensor temps,
+                  torch::Tensor top_ps, torch::Tensor seeds);
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("rmsnorm", &rmsnorm, "RMSNorm (bf16, CDNA4)");
+  m.def("fused_add_rmsnorm", &fused_add_rmsnorm,
+        "fused residual add + RMSNorm (in-place, bf16)");
+  m.def("silu_mul", &silu_mul, "SwiGLU activation (bf16)");
+  m.def("rope_kv_append", &rope_kv_append,
+        "fused rotate-half RoPE + paged KV append (bf16)");
+  m.def("kv_append", &kv_append, "paged KV append without RoPE (bf16)");
+  m.def("attn_decode_paged", &attn_decode_paged,
+        "paged GQA decode attention (bf16, wave-per-kv-head)");
+  m.def("attn_prefill_paged", &attn_prefill_paged,
+        "paged causal prefill attention (bf16, MFMA)");
+  m.def("top_p_sample", &top_p_sample,
+        "fused temperature softmax + top-p sampling (sort-free)");
+}

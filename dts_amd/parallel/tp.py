@@ -53,8 +53,8 @@ class TPContext:
         return torch.cat(parts, dim=dim)
 
 
-def _init_weight(out_f: int, in_f: int, dtype, generator=None) -> torch.Tensor:
-    w = torch.empty(out_f, in_f, dtype=torch.float32)
+def _init_weight(out_f: int, in_f: int, dtype, generator=None, device="cpu") -> torch.Tensor:
+    w = torch.empty(out_f, in_f, dtype=torch.float32, device=device)
     std = 1.0 / math.sqrt(in_f)
     w.normal_(0.0, std, generator=generator)
     return w.to(dtype)
