@@ -56,8 +56,10 @@ MODEL_REGISTRY = {
         num_heads=32,
         num_kv_heads=8,
         head_dim=128,
-        rope_theta=500000.0,
-        max_position=8192,
+        # 16k rope table: the comparative judge re-embeds all sibling
+        # trajectories in one prompt (SURVEY.md §2.3 — largest prompt in
+        # the system) and overflows 8k on the 6x5 config.
+        max_position=16384,
     ),
     # Llama-3-70B — TP=8 config (BASELINE.json config 4)
     "llama-3-70b": ModelSpec(

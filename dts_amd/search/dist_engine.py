@@ -1,0 +1,159 @@
+"""SPMD data-parallel DTS engine (branch sharding across GPUs).
+
+Every rank runs the identical search loop; expansion and judging tasks are
+executed only on their owner rank (dts_amd/parallel/dp.py) against the
+rank-local serving engine, and merged at phase boundaries. All ranks end
+each round with identical trees, so prune/backprop/best-leaf need no
+coordination. See parallel/dp.py for the ordering rules.
+"""
+
+from __future__ import annotations
+
+from typing import Optional
+
+from dts_amd.parallel.dp import (
+    DPContext,
+    NodePayload,
+    ScorePayload,
+    apply_payload,
+    node_to_payload,
+    payload_to_score,
+    score_to_payload,
+)
+from dts_amd.search.engine import DTSEngine
+from dts_amd.search.tree import DialogueTree
+from dts_amd.search.types import DialogueNode, NodeStatus
+from dts_amd.utils.logging import log_phase
+
+
+class _DPSimulator:
+    """Shards parent nodes round-robin; gathers full node payloads."""
+
+    def __init__(self, sim, dp: DPContext, tree_ref) -> None:
+        self.sim = sim
+        self.dp = dp
+        self._tree_ref = tree_ref  # callable returning current tree
+
+    async def expand_nodes(self, nodes, turns, intents_per_node=1, tree=None,
+                           generate_intents=None):
+        if not self.dp.enabled:
+            return await self.sim.expand_nodes(
+                nodes, turns, intents_per_node, tree, generate_intents
+            )
+        ordered = sorted(nodes, key=lambda n: n.id)
+        mine = [n for i, n in enumerate(ordered) if self.dp.owns(i)]
+        expanded_local = await self.sim.expand_nodes(
+            mine, turns, intents_per_node, tree, generate_intents
+        )
+        payloads = [node_to_payload(n) for n in expanded_local]
+        gathered = self.dp.all_gather_obj(payloads)
+        merged: list = []
+        all_payloads = [p for rank_list in gathered for p in rank_list]
+        all_payloads.sort(key=lambda p: (p.parent_id or "", p.node_id))
+        local_ids = {n.id for n in expanded_local}
+        for p in all_payloads:
+            node = apply_payload(tree, p)
+            merged.append(node)
+        return merged
+
+
+class _DPEvaluator:
+    def __init__(self, ev, dp: DPContext) -> None:
+        self.ev = ev
+        self.dp = dp
+
+    async def evaluate_absolute(self, nodes):
+        return await self._evaluate(nodes, "absolute")
+
+    async def evaluate_comparative(self, nodes):
+        return await self._evaluate(nodes, "comparative")
+
+    def set_research_context(self, ctx):
+        self.ev.set_research_context(ctx)
+
+    async def _evaluate(self, nodes, mode):
+        if not self.dp.enabled:
+            fn = (
+                self.ev.evaluate_comparative
+                if mode == "comparative"
+                else self.ev.evaluate_absolute
+            )
+            return await fn(nodes)
+        ordered = sorted(nodes, key=lambda n: n.id)
+        if mode == "comparative":
+            # shard whole sibling groups so forced ranking stays local
+            groups: dict = {}
+            for n in ordered:
+                groups.setdefault(n.parent_id or "root", []).append(n)
+            keys = sorted(groups)
+            mine: list = []
+            for i, k in enumerate(keys):
+                if self.dp.owns(i):
+                    mine.extend(groups[k])
+            local_scores = await self.ev.evaluate_comparative(mine) if mine else {}
+        else:
+            mine = [n for i, n in enumerate(ordered) if self.dp.owns(i)]
+            local_scores = await self.ev.evaluate_absolute(mine) if mine else {}
+
+        by_id = {n.id: n for n in ordered}
+        payloads = [
+            score_to_payload(nid, agg, by_id[nid].stats.critiques)
+            for nid, agg in local_scores.items()
+        ]
+        gathered = self.dp.all_gather_obj(payloads)
+        scores: dict = {}
+        threshold = self.ev.prune_threshold
+        flat = [p for rank_list in gathered for p in rank_list]
+        flat.sort(key=lambda p: p.node_id)
+        for p in flat:
+            agg = payload_to_score(p, threshold)
+            scores[p.node_id] = agg
+            node = by_id.get(p.node_id)
+            if node is not None:
+                node.stats.judge_scores = agg.individual_scores
+                node.stats.aggregated_score = agg.aggregated_score
+                if p.critiques:
+                    node.stats.critiques = p.critiques
+        return scores
+
+
+class DistributedDTSEngine(DTSEngine):
+    """DTSEngine with DP branch sharding over a torch.distributed group."""
+
+    def __init__(self, llm, config, dp: Optional[DPContext] = None,
+                 researcher: Optional[object] = None) -> None:
+        super().__init__(llm, config, researcher=researcher)
+        self.dp = dp or DPContext()
+        self._simulator = _DPSimulator(self._simulator, self.dp, lambda: self.tree)
+        self._evaluator = _DPEvaluator(self._evaluator, self.dp)
+
+    async def _initialize_tree(self) -> DialogueTree:
+        if not self.dp.enabled:
+            return await super()._initialize_tree()
+        if self.dp.rank == 0:
+            tree = await super()._initialize_tree()
+            root = tree.get_root()
+            payloads = [node_to_payload(root)] + [
+                node_to_payload(tree.get(cid)) for cid in root.children
+            ]
+        else:
+            payloads = None
+        payloads = self.dp.broadcast_obj(payloads, src=0)
+        if self.dp.rank != 0:
+            root_p = payloads[0]
+            from dts_amd.llm.types import Message
+
+            root = DialogueNode(
+                id=root_p.node_id,
+                depth=0,
+                messages=[Message(role=r, content=c) for r, c in root_p.messages],
+            )
+            tree = DialogueTree.create(root)
+            for p in payloads[1:]:
+                apply_payload(tree, p)
+            log_phase(
+                "INIT",
+                f"[rank {self.dp.rank}] mirrored tree with "
+                f"{len(payloads) - 1} branches",
+            )
+        return tree
