@@ -1,0 +1,122 @@
+"""Server tests (parity coverage of ref tests/api/test_server.py).
+
+Uses FastAPI TestClient with the deterministic FakeBackend as the
+inference path — the full search runs behind the websocket."""
+
+import json
+
+import pytest
+from fastapi.testclient import TestClient
+
+from dts_amd.llm import LLM, FakeBackend
+from dts_amd.server.app import create_app
+from dts_amd.server.schemas import SearchRequest
+
+
+@pytest.fixture
+def client():
+    app = create_app(lambda: LLM(FakeBackend(), default_model="fake"))
+    with TestClient(app) as c:
+        yield c
+
+
+class TestRest:
+    def test_health(self, client):
+        r = client.get("/health")
+        assert r.status_code == 200
+        assert r.json() == {"status": "ok"}
+
+    def test_config_defaults(self, client):
+        r = client.get("/config")
+        body = r.json()
+        assert body["init_branches"] == 6
+        assert body["turns_per_branch"] == 5
+        assert body["scoring_mode"] == "comparative"
+        assert body["prune_threshold"] == 6.5
+        # additive fields present
+        assert body["user_variability"] is False
+
+    def test_models_list(self, client):
+        r = client.get("/api/models")
+        data = r.json()["data"]
+        ids = {m["id"] for m in data}
+        assert "llama-3-8b" in ids
+        assert all(m["architecture"]["modality"] == "text->text" for m in data)
+
+    def test_index(self, client):
+        r = client.get("/")
+        assert r.status_code == 200
+
+
+class TestSchemas:
+    def test_request_bounds(self):
+        with pytest.raises(Exception):
+            SearchRequest(goal="g", first_message="m", init_branches=0)
+        with pytest.raises(Exception):
+            SearchRequest(goal="g", first_message="m", prune_threshold=11)
+        req = SearchRequest(goal="g", first_message="m")
+        assert req.rounds == 1 and req.scoring_mode == "comparative"
+
+    def test_additive_fields_roundtrip(self):
+        req = SearchRequest(
+            goal="g", first_message="m", user_variability=True, reasoning_enabled=True
+        )
+        assert req.user_variability and req.reasoning_enabled
+
+
+class TestWebSocket:
+    def test_ping_pong(self, client):
+        with client.websocket_connect("/ws") as ws:
+            ws.send_text(json.dumps({"type": "ping"}))
+            msg = json.loads(ws.receive_text())
+            assert msg["type"] == "pong"
+
+    def test_invalid_json(self, client):
+        with client.websocket_connect("/ws") as ws:
+            ws.send_text("not json")
+            msg = json.loads(ws.receive_text())
+            assert msg["type"] == "error"
+
+    def test_invalid_config(self, client):
+        with client.websocket_connect("/ws") as ws:
+            ws.send_text(
+                json.dumps({"type": "start_search", "config": {"goal": "only"}})
+            )
+            msg = json.loads(ws.receive_text())
+            assert msg["type"] == "error"
+            assert "invalid config" in msg["data"]["message"]
+
+    def test_full_search_stream(self, client):
+        with client.websocket_connect("/ws") as ws:
+            ws.send_text(
+                json.dumps(
+                    {
+                        "type": "start_search",
+                        "config": {
+                            "goal": "teach fractions",
+                            "first_message": "can you help me learn fractions?",
+                            "init_branches": 2,
+                            "turns_per_branch": 1,
+                            "scoring_mode": "absolute",
+                            "prune_threshold": 0.0,
+                            "rounds": 1,
+                        },
+                    }
+                )
+            )
+            events = []
+            while True:
+                msg = json.loads(ws.receive_text())
+                events.append(msg["type"])
+                if msg["type"] in ("complete", "error"):
+                    final = msg
+                    break
+            assert final["type"] == "complete"
+            assert "search_started" in events
+            assert "node_added" in events
+            assert "node_updated" in events
+            exploration = final["data"]["exploration"]
+            assert exploration["summary"]["total_branches"] == 2
+            assert {"summary", "best_branch", "branches", "research_report"} <= set(
+                exploration
+            )
