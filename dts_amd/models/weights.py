@@ -1,0 +1,125 @@
+"""Weight I/O: safetensors loading (HF layout) + random-init mode.
+
+SURVEY.md §2.3 inventory row "Tokenizer + safetensors weight loader w/
+random-init mode". Maps HuggingFace Llama/Mixtral checkpoint names onto
+this framework's fused layouts (q/k/v fused into qkv_proj, gate/up fused
+into gate_up_proj) with TP-aware sharding, so a real Llama-3 checkpoint
+drops in when one is available; random-init (models/*.random_init) is the
+synthetic-bench path since this environment has no network for weights.
+"""
+
+from __future__ import annotations
+
+from pathlib import Path
+from typing import Optional
+
+import torch
+
+from dts_amd.utils.logging import logger
+
+
+def _shard(t: torch.Tensor, dim: int, rank: int, size: int) -> torch.Tensor:
+    if size == 1:
+        return t
+    chunk = t.shape[dim] // size
+    return t.narrow(dim, rank * chunk, chunk)
+
+
+def load_llama_safetensors(model, path: str) -> int:
+    """Load an HF-format Llama checkpoint directory into a LlamaModel.
+
+    Returns the number of tensors consumed. TP: column-parallel weights are
+    sharded on dim 0, row-parallel on dim 1, per the model's tp context.
+    """
+    from safetensors import safe_open
+
+    spec = model.spec
+    tp = model.tp
+    files = sorted(Path(path).glob("*.safetensors"))
+    if not files:
+        raise FileNotFoundError(f"no .safetensors under {path}")
+
+    tensors: dict = {}
+    for f in files:
+        with safe_open(str(f), framework="pt") as sf:
+            for name in sf.keys():
+                tensors[name] = sf.get_tensor(name)
+
+    def get(name: str) -> torch.Tensor:
+        if name not in tensors:
+            raise KeyError(f"missing tensor {name}")
+        return tensors[name].to(torch.float32)
+
+    consumed = 0
+    with torch.no_grad():
+        model.embed.copy_(get("model.embed_tokens.weight").to(model.dtype))
+        consumed += 1
+        for i, layer in enumerate(model.layers):
+            p = f"model.layers.{i}."
+            layer.input_norm_w.copy_(get(p + "input_layernorm.weight").to(model.dtype))
+            layer.post_norm_w.copy_(
+                get(p + "post_attention_layernorm.weight").to(model.dtype)
+            )
+            q = _shard(get(p + "self_attn.q_proj.weight"), 0, tp.rank, tp.size)
+            k = _shard(get(p + "self_attn.k_proj.weight"), 0, tp.rank, tp.size)
+            v = _shard(get(p + "self_attn.v_proj.weight"), 0, tp.rank, tp.size)
+            layer.attn.qkv_proj.weight.copy_(
+                torch.cat([q, k, v], dim=0).to(model.dtype)
+            )
+            layer.attn.o_proj.weight.copy_(
+                _shard(get(p + "self_attn.o_proj.weight"), 1, tp.rank, tp.size).to(
+                    model.dtype
+                )
+            )
+            g = _shard(get(p + "mlp.gate_proj.weight"), 0, tp.rank, tp.size)
+            u = _shard(get(p + "mlp.up_proj.weight"), 0, tp.rank, tp.size)
+            layer.mlp.gate_up_proj.weight.copy_(
+                torch.cat([g, u], dim=0).to(model.dtype)
+            )
+            layer.mlp.down_proj.weight.copy_(
+                _shard(get(p + "mlp.down_proj.weight"), 1, tp.rank, tp.size).to(
+                    model.dtype
+                )
+            )
+            consumed += 9
+        model.final_norm_w.copy_(get("model.norm.weight").to(model.dtype))
+        lm = tensors.get("lm_head.weight", tensors.get("model.embed_tokens.weight"))
+        model.lm_head.weight.copy_(
+            _shard(lm.to(torch.float32), 0, tp.rank, tp.size).to(model.dtype)
+        )
+        consumed += 2
+    logger.info("loaded %d tensors from %s", consumed, path)
+    return consumed
+
+
+def save_llama_safetensors(model, path: str) -> None:
+    """Write the model back out in HF Llama naming (single shard).
+
+    Only valid for tp.size == 1; used by tests to round-trip the loader.
+    """
+    from safetensors.torch import save_file
+
+    assert model.tp.size == 1
+    out: dict = {}
+    out["model.embed_tokens.weight"] = model.embed.data.clone()
+    spec = model.spec
+    q_size = spec.num_heads * spec.head_dim
+    kv_size = spec.num_kv_heads * spec.head_dim
+    for i, layer in enumerate(model.layers):
+        p = f"model.layers.{i}."
+        out[p + "input_layernorm.weight"] = layer.input_norm_w.data.clone()
+        out[p + "post_attention_layernorm.weight"] = layer.post_norm_w.data.clone()
+        qkv = layer.attn.qkv_proj.weight.data
+        out[p + "self_attn.q_proj.weight"] = qkv[:q_size].clone()
+        out[p + "self_attn.k_proj.weight"] = qkv[q_size : q_size + kv_size].clone()
+        out[p + "self_attn.v_proj.weight"] = qkv[q_size + kv_size :].clone()
+        out[p + "self_attn.o_proj.weight"] = layer.attn.o_proj.weight.data.clone()
+        gu = layer.mlp.gate_up_proj.weight.data
+        inter = spec.intermediate_size
+        out[p + "mlp.gate_proj.weight"] = gu[:inter].clone()
+        out[p + "mlp.up_proj.weight"] = gu[inter:].clone()
+        out[p + "mlp.down_proj.weight"] = layer.mlp.down_proj.weight.data.clone()
+    out["model.norm.weight"] = model.final_norm_w.data.clone()
+    out["lm_head.weight"] = model.lm_head.weight.data.clone()
+    Path(path).mkdir(parents=True, exist_ok=True)
+    save_file(out, str(Path(path) / "model.safetensors"))

@@ -45,7 +45,16 @@ async def run_dts_session(
     """Run a search, yielding events as they arrive plus a final
     `complete` event (ref dts_service.py:43-98)."""
     config = create_dts_config(request)
-    engine = DTSEngine(llm, config)
+    researcher = None
+    if config.deep_research:
+        from dts_amd.search.researcher import DeepResearcher
+
+        researcher = DeepResearcher(
+            llm,
+            model=config.strategy_model or config.model,
+            cache_dir=config.research_cache_dir,
+        )
+    engine = DTSEngine(llm, config, researcher=researcher)
 
     queue: asyncio.Queue = asyncio.Queue()
 

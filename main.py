@@ -1,0 +1,56 @@
+"""Headless example runner (parity: reference main.py:40-61).
+
+Runs a small search on the local serving engine and saves the
+tree-state JSON checkpoint to dts_output.json. Uses the tiny model on
+CPU and Llama-3-8B on a GPU.
+"""
+
+from __future__ import annotations
+
+import asyncio
+import time
+
+import torch
+
+from dts_amd.llm import LLM
+from dts_amd.search import DTSConfig, DTSEngine
+from dts_amd.search.researcher import DeepResearcher
+from dts_amd.serving import LocalBackend, ServingEngine
+
+
+async def run_dts_example() -> None:
+    use_gpu = torch.cuda.is_available()
+    model = "llama-3-8b" if use_gpu else "llama-tiny"
+    engine_kwargs = {} if use_gpu else {
+        "num_blocks": 4096,
+        "block_size": 16,
+        "dtype": torch.float32,
+    }
+    serving = ServingEngine(model_name=model, **engine_kwargs)
+    backend = LocalBackend.single(serving, name=model)
+    llm = LLM(backend, default_model=model)
+
+    config = DTSConfig(
+        goal="Convince a skeptical team lead to adopt automated testing",
+        first_message="Our team doesn't write tests. Is it really worth the time?",
+        init_branches=3,
+        turns_per_branch=2,
+        user_intents_per_branch=2,
+        user_variability=True,
+        scoring_mode="comparative",
+        prune_threshold=5.0,
+        deep_research=True,
+        seed=0,
+    )
+    researcher = DeepResearcher(llm, cache_dir=config.research_cache_dir)
+    engine = DTSEngine(llm, config, researcher=researcher)
+    result = await engine.run(rounds=2)
+    result.save_json("dts_output.json")
+    print(f"Best score: {result.best_score:.1f} — saved to dts_output.json")
+    backend.shutdown()
+
+
+if __name__ == "__main__":
+    start = time.time()
+    asyncio.run(run_dts_example())
+    print(f"Completed in {time.time() - start:.1f}s")
