@@ -158,7 +158,7 @@ class LLM:
             return completion
 
         last_err: Optional[Exception] = None
-        for attempt in range(self.max_json_retries):
+        for attempt in range(self.max_json_retries):  # ref client.py:148-203
             completion = await self.backend.chat(messages, params, model=model)
             try:
                 completion.data = parse_json_completion(completion.message.content or "")
@@ -175,3 +175,31 @@ class LLM:
                 if params.seed is not None:
                     params.seed += 1
         raise last_err  # type: ignore[misc]
+
+    async def stream(
+        self,
+        messages: list,
+        model: Optional[str] = None,
+        temperature: float = 0.7,
+        max_tokens: Optional[int] = None,
+        top_p: float = 0.95,
+        seed: Optional[int] = None,
+        **_ignored,
+    ):
+        """Async iterator of text deltas (ref client.py:205-272). Backends
+        without native streaming fall back to one whole-message chunk."""
+        params = SamplingParams(
+            temperature=temperature,
+            top_p=top_p,
+            max_tokens=max_tokens or DEFAULT_MAX_TOKENS,
+            seed=seed,
+        )
+        model = model or self._default_model
+        backend_stream = getattr(self.backend, "stream", None)
+        if backend_stream is None:
+            completion = await self.backend.chat(messages, params, model=model)
+            if completion.message.content:
+                yield strip_think_tags(completion.message.content)
+            return
+        async for delta in backend_stream(messages, params, model=model):
+            yield delta

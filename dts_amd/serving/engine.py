@@ -122,6 +122,7 @@ class ServingEngine:
         prompt_ids: list,
         params: SamplingParams,
         guide: Optional[object] = None,
+        stream_cb: Optional[object] = None,
     ) -> concurrent.futures.Future:
         if len(prompt_ids) + 16 > self.spec.max_position:
             raise ContextLengthError(
@@ -130,6 +131,7 @@ class ServingEngine:
             )
         seq = Sequence(tokens=list(prompt_ids), params=params)
         seq.guide = guide
+        seq.stream_cb = stream_cb
         if guide is not None:
             forced = guide.initial_forced()
             seq.tokens.extend(forced)
@@ -179,11 +181,21 @@ class ServingEngine:
     def _handle_sampled(self, seq: Sequence, tok: int) -> None:
         params = seq.params
         seq.append_token(tok)
+        if seq.stream_cb is not None:
+            try:
+                seq.stream_cb([tok])
+            except Exception:  # noqa: BLE001 — streaming must not kill decode
+                seq.stream_cb = None
         if seq.guide is not None:
             forced = seq.guide.on_token(tok)
             if forced:
                 seq.tokens.extend(forced)
                 seq.output_tokens.extend(forced)
+                if seq.stream_cb is not None:
+                    try:
+                        seq.stream_cb(list(forced))
+                    except Exception:  # noqa: BLE001
+                        seq.stream_cb = None
             if seq.guide.done():
                 self._finish(seq, "stop")
             elif (
@@ -359,3 +371,66 @@ class LocalBackend:
             model=model or self.default_model,
             finish_reason=result.finish_reason,
         )
+
+    async def stream(
+        self,
+        messages: list,
+        params: SamplingParams,
+        model: Optional[str] = None,
+    ):
+        """Async iterator of text deltas (parity: ref client.py:205-272
+        `LLM.stream`). Tokens are pushed from the engine thread onto an
+        asyncio queue; UTF-8 multibyte sequences are buffered until whole."""
+        engine = self._engine(model)
+        prompt_ids = engine.template.render(messages)
+        loop = asyncio.get_running_loop()
+        queue: asyncio.Queue = asyncio.Queue()
+
+        def on_tokens(toks: list) -> None:
+            loop.call_soon_threadsafe(queue.put_nowait, list(toks))
+
+        fut = engine.submit_tokens(prompt_ids, params, stream_cb=on_tokens)
+        afut = asyncio.wrap_future(fut)
+        stops = set(engine.template.stop_token_ids)
+        buf = bytearray()
+        tok = engine.tokenizer
+
+        def emit(toks: list) -> str:
+            parts = []
+            for t in toks:
+                if t in stops:
+                    continue
+                if t < 256:
+                    buf.append(t)
+                    try:
+                        parts.append(buf.decode("utf-8"))
+                        buf.clear()
+                    except UnicodeDecodeError:
+                        continue
+                else:
+                    if buf:
+                        parts.append(buf.decode("utf-8", errors="replace"))
+                        buf.clear()
+                    parts.append(tok.decode([t]))
+            return "".join(parts)
+
+        while True:
+            get_task = asyncio.ensure_future(queue.get())
+            done, _ = await asyncio.wait(
+                {get_task, afut}, return_when=asyncio.FIRST_COMPLETED
+            )
+            if get_task in done:
+                delta = emit(get_task.result())
+                if delta:
+                    yield delta
+            else:
+                get_task.cancel()
+                while not queue.empty():
+                    delta = emit(queue.get_nowait())
+                    if delta:
+                        yield delta
+                if buf:  # trailing incomplete UTF-8 sequence
+                    yield buf.decode("utf-8", errors="replace")
+                    buf.clear()
+                await afut  # surface exceptions
+                return

@@ -45,6 +45,8 @@ class Sequence:
     arrival_order: int = 0
     # scheduled in a forward that has not been postprocessed yet
     in_flight: bool = False
+    # optional per-token callback (called from the engine thread)
+    stream_cb: Optional[object] = None
 
     def __post_init__(self) -> None:
         self.num_prompt_tokens = len(self.tokens)
