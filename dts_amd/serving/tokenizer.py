@@ -40,6 +40,12 @@ class SyntheticTokenizer:
         return ([self.bos_id] + ids) if add_bos else ids
 
     def decode(self, ids) -> str:
+        """Ids in the byte range decode to their bytes; synthetic ids
+        (what sampling over a random-init model mostly produces) decode to
+        ONE printable char each, so text that re-enters a prompt re-encodes
+        to ~1 token per originally sampled token — matching how a real
+        BPE round-trips its own output. (A multi-char mapping here would
+        silently inflate every subsequent prompt by that factor.)"""
         out = []
         byte_buf = bytearray()
 
@@ -57,7 +63,8 @@ class SyntheticTokenizer:
                 flush()  # specials render as nothing
             else:
                 flush()
-                out.append(f" w{i % 9973}")  # stable pseudo-word
+                h = i % 95  # printable ASCII 32..126
+                out.append(chr(32 + h))
         flush()
         return "".join(out)
 
