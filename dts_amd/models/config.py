@@ -104,6 +104,22 @@ MODEL_REGISTRY = {
         max_position=1024,
         tie_embeddings=True,
     ),
+    # Llama-3-8B dims at depth 2 — for GPU cross-path consistency tests:
+    # a random-init 32-layer stack amplifies bf16 reduction-order noise
+    # into decorrelated logits, so path-equivalence is asserted at depth 2
+    # where drift stays bounded while the kernels run at real shapes.
+    "llama-3-8b-2l": ModelSpec(
+        name="llama-3-8b-2l",
+        vocab_size=128256,
+        hidden_size=4096,
+        intermediate_size=14336,
+        num_layers=2,
+        num_heads=32,
+        num_kv_heads=8,
+        head_dim=128,
+        rope_theta=500000.0,
+        max_position=16384,
+    ),
     # Tiny models for CPU tests
     "llama-tiny": ModelSpec(
         name="llama-tiny",

@@ -17,6 +17,7 @@ from __future__ import annotations
 
 import asyncio
 import concurrent.futures
+import os
 import re
 import threading
 from dataclasses import dataclass
@@ -100,11 +101,26 @@ class ServingEngine:
             dtype=dtype,
             device=device,
         )
-        self.block_manager = BlockManager(num_blocks, block_size)
+        # last block reserved as the hipGraph pad-row scratch target
+        self.block_manager = BlockManager(max(1, num_blocks - 1), block_size)
+        self._scratch_block = num_blocks - 1
         self.scheduler = Scheduler(self.block_manager, max_batch_tokens, max_running)
         self.sampler = Sampler(device)
         self.tokenizer = SyntheticTokenizer(self.spec.vocab_size)
         self.template = ChatTemplate(self.tokenizer)
+
+        self._graph_runner = None
+        if device.startswith("cuda") and os.environ.get("DTS_NO_HIPGRAPH") != "1":
+            from dts_amd.serving.graph_runner import DecodeGraphRunner
+
+            self._graph_runner = DecodeGraphRunner(
+                self.model,
+                self.kv_pool,
+                device,
+                scratch_block=self._scratch_block,
+                max_blocks_per_seq=self.spec.max_position // block_size,
+                max_bucket=min(256, max_running),
+            )
 
         self._lock = threading.Lock()
         self._work = threading.Condition(self._lock)
@@ -113,6 +129,7 @@ class ServingEngine:
         self._stop = False
         # stats
         self.steps = 0
+        self.graph_steps = 0
         self.tokens_sampled = 0
         self.tokens_prefilled = 0
 
@@ -166,8 +183,12 @@ class ServingEngine:
         if batch is None:
             return False
         self.steps += 1
-        dev_batch = batch.to(self.device) if self.device != "cpu" else batch
-        logits = self.model.forward(dev_batch, self.kv_pool)
+        if self._graph_runner is not None and self._graph_runner.can_run(batch):
+            logits = self._graph_runner.run(batch)
+            self.graph_steps += 1
+        else:
+            dev_batch = batch.to(self.device) if self.device != "cpu" else batch
+            logits = self.model.forward(dev_batch, self.kv_pool)
         sampled_seqs = batch._sampled_seqs  # type: ignore[attr-defined]
         tokens = self.sampler.sample(logits, sampled_seqs) if sampled_seqs else []
         with self._lock:
@@ -280,6 +301,7 @@ class ServingEngine:
             "cache_miss_tokens": bm.cache_miss_tokens,
             "free_blocks": bm.num_free(),
             "steps": self.steps,
+            "graph_steps": self.graph_steps,
             "tokens_sampled": self.tokens_sampled,
             "tokens_prefilled": self.tokens_prefilled,
         }

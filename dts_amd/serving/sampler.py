@@ -57,11 +57,13 @@ class Sampler:
         top_ps = torch.tensor([s.params.top_p for s in seqs], dtype=torch.float32)
 
         if logits.is_cuda:
+            import random as _random
+
             seeds = torch.tensor(
                 [
                     (hash((s.params.seed or 0, s.seq_id, len(s.tokens))) & 0x7FFFFFFF)
                     if s.params.seed is not None
-                    else int(torch.randint(0, 2**31 - 1, (1,)).item())
+                    else _random.getrandbits(31)
                     for s in seqs
                 ],
                 dtype=torch.long,

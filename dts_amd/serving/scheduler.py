@@ -95,11 +95,15 @@ class Scheduler:
             seq.block_table.append(self.bm.allocate_fresh())
         return True
 
-    def _preempt_youngest(self, exclude: Sequence) -> bool:
-        candidates = [s for s in self.running if s is not exclude and not s.in_flight]
+    def _preempt_youngest(self, exclude: Optional[Sequence] = None) -> bool:
+        """Preempt the globally YOUNGEST runnable sequence (the requester
+        itself may be the victim — oldest work survives KV pressure)."""
+        candidates = [s for s in self.running if not s.in_flight]
         if not candidates:
             return False
         victim = max(candidates, key=lambda s: s.arrival_order)
+        if victim is exclude and len(candidates) == 1:
+            return False  # nothing else to evict — caller handles stuck
         logger.warning("preempting seq %d (KV pressure)", victim.seq_id)
         self._release_blocks(victim)
         victim.num_computed = 0
@@ -132,6 +136,10 @@ class Scheduler:
                 held.append(seq)
                 continue
             if not self._admit(seq):
+                if seq.blocks_needed(self.block_size) > self.bm.num_blocks:
+                    # can never fit, even with the whole pool — fail loudly
+                    self.stuck.append(seq)
+                    continue
                 held.append(seq)
                 break
             seq.status = SeqStatus.RUNNING
@@ -162,6 +170,8 @@ class Scheduler:
                         self.running.remove(seq)
                         self.stuck.append(seq)
                     continue
+                if seq.status != SeqStatus.RUNNING:
+                    continue  # seq itself was the preemption victim
                 if not self._ensure_blocks(seq, len(seq.tokens)):
                     continue
             chunk = min(remaining, budget)

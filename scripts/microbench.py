@@ -31,9 +31,12 @@ def timed_steps(engine, n_steps):
         if batch is None:
             break
         t1 = time.perf_counter()
-        dev_batch = batch.to(engine.device)
-        with torch.inference_mode():
-            logits = engine.model.forward(dev_batch, engine.kv_pool)
+        if engine._graph_runner is not None and engine._graph_runner.can_run(batch):
+            logits = engine._graph_runner.run(batch)
+        else:
+            dev_batch = batch.to(engine.device)
+            with torch.inference_mode():
+                logits = engine.model.forward(dev_batch, engine.kv_pool)
         torch.cuda.synchronize()
         t2 = time.perf_counter()
         seqs = batch._sampled_seqs
