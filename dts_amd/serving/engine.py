@@ -104,7 +104,19 @@ class ServingEngine:
         # last block reserved as the hipGraph pad-row scratch target
         self.block_manager = BlockManager(max(1, num_blocks - 1), block_size)
         self._scratch_block = num_blocks - 1
-        self.scheduler = Scheduler(self.block_manager, max_batch_tokens, max_running)
+        use_native = os.environ.get("DTS_NATIVE_CORE", "1") != "0"
+        self.scheduler = None
+        if use_native:
+            try:
+                from dts_amd.serving.native_scheduler import NativeScheduler
+
+                self.scheduler = NativeScheduler(
+                    max(1, num_blocks - 1), block_size, max_batch_tokens, max_running
+                )
+            except Exception as e:  # noqa: BLE001
+                logger.warning("native core unavailable (%s); Python scheduler", e)
+        if self.scheduler is None:
+            self.scheduler = Scheduler(self.block_manager, max_batch_tokens, max_running)
         self.sampler = Sampler(device)
         self.tokenizer = SyntheticTokenizer(self.spec.vocab_size)
         self.template = ChatTemplate(self.tokenizer)
@@ -201,7 +213,7 @@ class ServingEngine:
 
     def _handle_sampled(self, seq: Sequence, tok: int) -> None:
         params = seq.params
-        seq.append_token(tok)
+        self.scheduler.append_token(seq, tok)
         if seq.stream_cb is not None:
             try:
                 seq.stream_cb([tok])
@@ -210,8 +222,7 @@ class ServingEngine:
         if seq.guide is not None:
             forced = seq.guide.on_token(tok)
             if forced:
-                seq.tokens.extend(forced)
-                seq.output_tokens.extend(forced)
+                self.scheduler.extend_tokens(seq, forced)
                 if seq.stream_cb is not None:
                     try:
                         seq.stream_cb(list(forced))
@@ -295,15 +306,20 @@ class ServingEngine:
 
     @property
     def cache_stats(self) -> dict:
-        bm = self.block_manager
+        src = (
+            self.scheduler
+            if hasattr(self.scheduler, "cache_hit_tokens")
+            else self.block_manager
+        )
         return {
-            "cache_hit_tokens": bm.cache_hit_tokens,
-            "cache_miss_tokens": bm.cache_miss_tokens,
-            "free_blocks": bm.num_free(),
+            "cache_hit_tokens": src.cache_hit_tokens,
+            "cache_miss_tokens": src.cache_miss_tokens,
+            "free_blocks": src.num_free(),
             "steps": self.steps,
             "graph_steps": self.graph_steps,
             "tokens_sampled": self.tokens_sampled,
             "tokens_prefilled": self.tokens_prefilled,
+            "native_scheduler": type(self.scheduler).__name__ == "NativeScheduler",
         }
 
 

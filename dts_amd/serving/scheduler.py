@@ -85,6 +85,13 @@ class Scheduler:
     def has_work(self) -> bool:
         return bool(self.waiting or self.running)
 
+    def append_token(self, seq: Sequence, tok: int) -> None:
+        seq.append_token(tok)
+
+    def extend_tokens(self, seq: Sequence, toks: list) -> None:
+        seq.tokens.extend(toks)
+        seq.output_tokens.extend(toks)
+
     # ------------------------------------------------------------------
     def _ensure_blocks(self, seq: Sequence, upto_tokens: int) -> bool:
         """Grow seq.block_table to cover `upto_tokens` tokens; False if OOM."""

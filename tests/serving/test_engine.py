@@ -67,17 +67,17 @@ class TestPrefixCache:
     def test_repeat_prompt_hits_cache(self, engine):
         prompt = list(range(1, 50))  # 49 tokens -> 12 full blocks of 4
         _gen(engine, list(prompt), max_tokens=2)
-        before = engine.block_manager.cache_hit_tokens
+        before = engine.cache_stats["cache_hit_tokens"]
         _gen(engine, list(prompt), max_tokens=2)
-        after = engine.block_manager.cache_hit_tokens
+        after = engine.cache_stats["cache_hit_tokens"]
         assert after - before >= 44  # nearly the whole prompt reused
 
     def test_shared_prefix_divergent_tails(self, engine):
         prefix = list(range(60, 100))  # 40 tokens = 10 blocks
         _gen(engine, prefix + [1, 2, 3], max_tokens=2)
-        before = engine.block_manager.cache_hit_tokens
+        before = engine.cache_stats["cache_hit_tokens"]
         _gen(engine, prefix + [7, 8, 9], max_tokens=2)
-        assert engine.block_manager.cache_hit_tokens - before >= 40
+        assert engine.cache_stats["cache_hit_tokens"] - before >= 40
 
     def test_cache_correctness_vs_cold(self):
         """Same prompt scored cold vs via cache gives identical greedy tokens."""

@@ -63,9 +63,9 @@ class TestGPUEngine:
     def test_prefix_cache_reuses_blocks(self, engine):
         prompt = list(range(7, 700))
         cold = _gen(engine, prompt, max_tokens=12, temperature=0.0)
-        hits0 = engine.block_manager.cache_hit_tokens
+        hits0 = engine.cache_stats["cache_hit_tokens"]
         warm = _gen(engine, prompt, max_tokens=12, temperature=0.0)
-        assert engine.block_manager.cache_hit_tokens - hits0 >= 600
+        assert engine.cache_stats["cache_hit_tokens"] - hits0 >= 600
         assert cold.completion_tokens == warm.completion_tokens == 12
 
     def test_concurrent_mixed_batch(self, engine):

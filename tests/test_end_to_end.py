@@ -59,7 +59,7 @@ class TestEndToEnd:
         # 1 strategy + 2 branches x (1 user + 1 assistant) + 2 x 3 judges
         assert result.token_usage["totals"]["total_requests"] >= 1 + 2 * 2 + 2 * 3
         # prefix cache must have been exercised by shared prompts
-        assert engine.block_manager.cache_hit_tokens > 0
+        assert engine.cache_stats["cache_hit_tokens"] > 0
 
     def test_comparative_search_with_forking(self, local_llm):
         llm, engine = local_llm
