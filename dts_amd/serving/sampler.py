@@ -33,28 +33,55 @@ class Sampler:
     def release(self, seq) -> None:
         self._generators.pop(seq.seq_id, None)
 
+    def _sample_constrained(self, row: torch.Tensor, seq, allowed: list) -> int:
+        """Sample over a SMALL allowed set (constrained decoding): gather
+        the allowed logits instead of masking the full 128k vocab — the
+        set is typically 10-95 bytes, so this is a tiny CPU softmax."""
+        idx = torch.as_tensor(allowed, dtype=torch.long, device=row.device)
+        sub = row[idx].float().cpu()
+        t = seq.params.temperature
+        if t <= 0.0:
+            return int(idx[int(sub.argmax())])
+        probs = torch.softmax(sub / t, dim=-1)
+        p = seq.params.top_p
+        if p < 1.0 and probs.numel() > 1:
+            sp, si = torch.sort(probs, descending=True)
+            cum = torch.cumsum(sp, -1)
+            keep = (cum - sp) < p
+            keep[0] = True
+            kept = sp * keep
+            kept = kept / kept.sum()
+            pick = torch.multinomial(kept, 1, generator=self._generator_for(seq))
+            return int(idx[int(si[pick])])
+        pick = torch.multinomial(probs, 1, generator=self._generator_for(seq))
+        return int(idx[pick])
+
     def sample(self, logits: torch.Tensor, seqs: list) -> list:
         """logits [S, V] fp32 for the sampled rows; returns token ids."""
         S, V = logits.shape
         assert S == len(seqs)
 
-        # constrained-decoding masks
+        out: list = [None] * S
+        free_rows: list = []
         for i, seq in enumerate(seqs):
-            guide = seq.guide
-            if guide is not None:
-                allowed = guide.allowed_tokens()
-                if allowed is not None:
-                    mask = torch.full((V,), float("-inf"), device=logits.device)
-                    idx = torch.as_tensor(
-                        allowed, dtype=torch.long, device=logits.device
-                    )
-                    mask[idx] = 0.0
-                    logits[i] = logits[i] + mask
+            allowed = seq.guide.allowed_tokens() if seq.guide is not None else None
+            if allowed is not None:
+                out[i] = self._sample_constrained(logits[i], seq, allowed)
+            else:
+                free_rows.append(i)
+        if not free_rows:
+            return out
 
-        temps = torch.tensor(
-            [s.params.temperature for s in seqs], dtype=torch.float32
+        free_seqs = [seqs[i] for i in free_rows]
+        free_logits = (
+            logits
+            if len(free_rows) == S
+            else logits[torch.as_tensor(free_rows, device=logits.device)]
         )
-        top_ps = torch.tensor([s.params.top_p for s in seqs], dtype=torch.float32)
+        temps = torch.tensor(
+            [s.params.temperature for s in free_seqs], dtype=torch.float32
+        )
+        top_ps = torch.tensor([s.params.top_p for s in free_seqs], dtype=torch.float32)
 
         if logits.is_cuda:
             import random as _random
@@ -64,16 +91,20 @@ class Sampler:
                     (hash((s.params.seed or 0, s.seq_id, len(s.tokens))) & 0x7FFFFFFF)
                     if s.params.seed is not None
                     else _random.getrandbits(31)
-                    for s in seqs
+                    for s in free_seqs
                 ],
                 dtype=torch.long,
                 device=logits.device,
             )
             toks = ops.top_p_sample(
-                logits, temps.to(logits.device), top_ps.to(logits.device), seeds=seeds
-            )
-            return [int(t) for t in toks.cpu()]
-
-        gens = [self._generator_for(s) for s in seqs]
-        toks = ops.top_p_sample(logits.cpu(), temps, top_ps, generators=gens)
-        return [int(t) for t in toks]
+                free_logits.contiguous(),
+                temps.to(logits.device),
+                top_ps.to(logits.device),
+                seeds=seeds,
+            ).cpu()
+        else:
+            gens = [self._generator_for(s) for s in free_seqs]
+            toks = ops.top_p_sample(free_logits.cpu(), temps, top_ps, generators=gens)
+        for i, t in zip(free_rows, toks):
+            out[i] = int(t)
+        return out
