@@ -56,10 +56,11 @@ MODEL_REGISTRY = {
         num_heads=32,
         num_kv_heads=8,
         head_dim=128,
-        # 16k rope table: the comparative judge re-embeds all sibling
-        # trajectories in one prompt (SURVEY.md §2.3 — largest prompt in
-        # the system) and overflows 8k on the 6x5 config.
-        max_position=16384,
+        # 32k rope table (Llama-3.1-class context): the comparative judge
+        # re-embeds all sibling trajectories in one prompt (SURVEY.md §2.3
+        # — the largest prompt in the system, ~13k tokens on the 6x5
+        # config) and must also fit its ~3k-token structured answer.
+        max_position=32768,
     ),
     # Llama-3-70B — TP=8 config (BASELINE.json config 4)
     "llama-3-70b": ModelSpec(

@@ -57,7 +57,7 @@ class GPT2Layer(nn.Module):
         k_cache, v_cache = kv_layer
         ops.kv_append(k, v, k_cache, v_cache, batch.slot_mapping)
 
-        out = torch.empty_like(q)
+        out = torch.empty(q.shape, dtype=q.dtype, device=q.device)
         np_tok = batch.num_prefill_tokens
         if batch.num_prefill_seqs:
             out[:np_tok] = ops.attn_prefill_paged(

@@ -75,7 +75,8 @@ class LlamaAttention(nn.Module):
             q, k, v, batch.positions, cos, sin, k_cache, v_cache, batch.slot_mapping
         )
 
-        out = torch.empty_like(q)
+        # q may be a strided view of qkv; output is always dense
+        out = torch.empty(q.shape, dtype=q.dtype, device=q.device)
         np_tok = batch.num_prefill_tokens
         if batch.num_prefill_seqs:
             out[:np_tok] = ops.attn_prefill_paged(

@@ -120,7 +120,7 @@ def attn_prefill_paged(q, cu_q, q_positions, k_cache, v_cache, block_tables, kv_
         if ext is not None:
             import math
 
-            out = torch.empty_like(q)
+            out = torch.empty(q.shape, dtype=q.dtype, device=q.device)
             ext.attn_prefill_paged(
                 out, q, cu_q, q_positions, k_cache, v_cache, block_tables, kv_lens,
                 scale if scale is not None else 1.0 / math.sqrt(q.shape[-1]),
@@ -137,7 +137,7 @@ def attn_decode_paged(q, k_cache, v_cache, block_tables, kv_lens, scale=None):
         if ext is not None:
             import math
 
-            out = torch.empty_like(q)
+            out = torch.empty(q.shape, dtype=q.dtype, device=q.device)
             ext.attn_decode_paged(
                 out, q, k_cache, v_cache, block_tables, kv_lens,
                 scale if scale is not None else 1.0 / math.sqrt(q.shape[-1]),

@@ -28,26 +28,47 @@ using f32x4 = __attribute__((ext_vector_type(4))) float;
 // ---------------------------------------------------------------------------
 
 // G = q-heads per kv-head, D = head dim. Block size BS = 16 tokens.
+// Split-KV (flash-decoding): grid.z splits each sequence's pages into S
+// contiguous ranges; each (seq, kvh, split) workgroup computes partial
+// (m, l, o) into the fp32 workspace, combined by attn_decode_combine.
+// S is FIXED per launch so decode steps stay hipGraph-capturable with
+// varying kv_lens; splits past a short sequence's pages exit immediately.
 template <int G, int D>
 __global__ void __launch_bounds__(256)
-attn_decode_kernel(short* __restrict__ out,          // [B, Hq, D]
+attn_decode_kernel(float* __restrict__ ws_m,  // [B, Hkv, S, G]
+                   float* __restrict__ ws_l,  // [B, Hkv, S, G]
+                   float* __restrict__ ws_o,  // [B, Hkv, S, G, D]
                    const short* __restrict__ q,      // [B, Hq, D]
                    const short* __restrict__ kcache, // [N, Hkv, BS, D]
                    const short* __restrict__ vcache,
                    const int* __restrict__ block_tables, // [B, max_blocks]
                    const int* __restrict__ kv_lens,      // [B]
-                   int max_blocks, int Hkv, float scale) {
+                   int max_blocks, int Hkv, float scale, long q_tstride) {
   constexpr int BS = 16;
   constexpr int NWAVE = 4;
   constexpr int LPK = 4;            // lanes per key
   constexpr int DPL = D / LPK;      // dims per lane in K phase
   const int seq = blockIdx.x;
   const int kvh = blockIdx.y;
+  const int split = blockIdx.z;
+  const int S = gridDim.z;
   const int Hq = Hkv * G;
   const int wave = threadIdx.x / WAVE;
   const int lane = threadIdx.x & (WAVE - 1);
   const int kv_len = kv_lens[seq];
   const int n_pages = (kv_len + BS - 1) / BS;
+  const int pages_per_split = (n_pages + S - 1) / S;
+  const int page_lo = split * pages_per_split;
+  const int page_hi = min(n_pages, page_lo + pages_per_split);
+  const long ws_base = (((long)seq * Hkv + kvh) * S + split) * G;
+  if (page_lo >= page_hi) {
+    // empty split: publish -inf partials so the combine skips it
+    for (int g = threadIdx.x; g < G; g += blockDim.x) {
+      ws_m[ws_base + g] = -1e30f;
+      ws_l[ws_base + g] = 0.f;
+    }
+    return;
+  }
 
   // LDS: per-wave score buffer + combine buffers
   __shared__ float s_scores[NWAVE][G][BS];
@@ -62,7 +83,7 @@ attn_decode_kernel(short* __restrict__ out,          // [B, Hq, D]
   int qp_[G][DPL / 2];
 #pragma unroll
   for (int g = 0; g < G; ++g) {
-    const short* qp = q + (((long)seq * Hq) + kvh * G + g) * D + c * DPL;
+    const short* qp = q + (long)seq * q_tstride + (long)(kvh * G + g) * D + c * DPL;
 #pragma unroll
     for (int i = 0; i < DPL / 2; ++i) qp_[g][i] = ((const int*)qp)[i];
   }
@@ -77,7 +98,7 @@ attn_decode_kernel(short* __restrict__ out,          // [B, Hq, D]
 
   const int* bt = block_tables + (long)seq * max_blocks;
 
-  for (int page = wave; page < n_pages; page += NWAVE) {
+  for (int page = page_lo + wave; page < page_hi; page += NWAVE) {
     const long blk = bt[page];
     const short* kbase =
         kcache + ((blk * Hkv + kvh) * BS) * (long)D;
@@ -113,7 +134,7 @@ attn_decode_kernel(short* __restrict__ out,          // [B, Hq, D]
         s_scores[wave][g][key_of_lane] =
             (key_of_lane < valid) ? partial[g] * scale : -1e30f;
     }
-    __builtin_amdgcn_s_waitcnt(0); // lgkm for LDS writes within wave
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
     __builtin_amdgcn_wave_barrier();
 
     // ---- softmax update + V phase; lane owns output dims {2l, 2l+1}
@@ -150,7 +171,7 @@ attn_decode_kernel(short* __restrict__ out,          // [B, Hq, D]
     __builtin_amdgcn_wave_barrier();
   }
 
-  // ---- cross-wave combine via LDS
+  // ---- cross-wave combine via LDS, then publish split partials
 #pragma unroll
   for (int g = 0; g < G; ++g) {
     if (lane == 0) {
@@ -162,7 +183,6 @@ attn_decode_kernel(short* __restrict__ out,          // [B, Hq, D]
   }
   __syncthreads();
 
-  // threads 0..G*D/2: each handles one (g, dim-pair)
   const int tid = threadIdx.x;
   for (int gd = tid; gd < G * (D / 2); gd += blockDim.x) {
     const int g = gd / (D / 2);
@@ -178,10 +198,56 @@ attn_decode_kernel(short* __restrict__ out,          // [B, Hq, D]
       acc0 += s_o[w][g][d0] * f;
       acc1 += s_o[w][g][d0 + 1] * f;
     }
-    float inv = (l_star > 0.f) ? 1.f / l_star : 0.f;
-    short* op = out + (((long)seq * Hq) + kvh * G + g) * D + d0;
-    op[0] = f2bf(acc0 * inv);
-    op[1] = f2bf(acc1 * inv);
+    if (d0 == 0) {
+      ws_m[ws_base + g] = m_star;
+      ws_l[ws_base + g] = l_star;
+    }
+    float* wo = ws_o + (ws_base + g) * D;
+    wo[d0] = acc0;
+    wo[d0 + 1] = acc1;
+  }
+}
+
+// Combine the S split partials into the final bf16 output.
+template <int D>
+__global__ void __launch_bounds__(256)
+attn_decode_combine(short* __restrict__ out,        // [B, Hq, D]
+                    const float* __restrict__ ws_m, // [B, Hkv, S, G]
+                    const float* __restrict__ ws_l,
+                    const float* __restrict__ ws_o, // [B, Hkv, S, G, D]
+                    int S, int G, int Hkv) {
+  const int seq = blockIdx.x;
+  const int hq = blockIdx.y;  // 0..Hq-1
+  const int kvh = hq / G;
+  const int g = hq % G;
+  const int Hq = Hkv * G;
+  const long base0 = (((long)seq * Hkv + kvh) * S) * G + g;
+
+  __shared__ float sh_f[64];  // per-split weight
+  __shared__ float sh_linv[1];
+  // pass 1: global max + weighted l (thread 0 does the tiny reduction)
+  if (threadIdx.x == 0) {
+    float m_star = -1e30f;
+    for (int s2 = 0; s2 < S; ++s2)
+      m_star = fmaxf(m_star, ws_m[base0 + (long)s2 * G]);
+    float l_star = 0.f;
+    for (int s2 = 0; s2 < S; ++s2) {
+      float mm = ws_m[base0 + (long)s2 * G];
+      float f = (mm > -1e30f) ? __expf(mm - m_star) : 0.f;
+      sh_f[s2] = f;
+      l_star += ws_l[base0 + (long)s2 * G] * f;
+    }
+    sh_linv[0] = (l_star > 0.f) ? 1.f / l_star : 0.f;
+  }
+  __syncthreads();
+  const float linv = sh_linv[0];
+  for (int d = threadIdx.x; d < D; d += blockDim.x) {
+    float acc = 0.f;
+    for (int s2 = 0; s2 < S; ++s2) {
+      float f = sh_f[s2];
+      if (f > 0.f) acc += ws_o[(base0 + (long)s2 * G) * D + d] * f;
+    }
+    out[(((long)seq * Hq) + hq) * D + d] = f2bf(acc * linv);
   }
 }
 
@@ -201,7 +267,7 @@ attn_prefill_kernel(short* __restrict__ out,      // [Tq, Hq, D]
                     const short* __restrict__ vcache,
                     const int* __restrict__ block_tables,
                     const int* __restrict__ kv_lens, int max_blocks, int Hkv,
-                    float scale) {
+                    float scale, long q_tstride) {
   constexpr int BS = 16;
   constexpr int KSTEP = 32;
   constexpr int D = 128;
@@ -242,7 +308,7 @@ attn_prefill_kernel(short* __restrict__ out,      // [Tq, Hq, D]
     const int head = row_global % G;
     const bool valid_row = pos_local < q_len;
     const int tok = q_start + (valid_row ? pos_local : 0);
-    const short* qp = q + ((long)tok * Hq + kvh * G + head) * D;
+    const short* qp = q + (long)tok * q_tstride + (long)(kvh * G + head) * D;
 #pragma unroll
     for (int kc = 0; kc < 4; ++kc) {
       const int kbase = kc * 32 + (lane >> 4) * 8;
@@ -446,17 +512,35 @@ void attn_decode_paged(torch::Tensor out, torch::Tensor q,
   const int max_blocks = block_tables.size(1);
   TORCH_CHECK(kcache.size(2) == 16, "block_size must be 16");
   TORCH_CHECK(block_tables.scalar_type() == torch::kInt32);
-  dim3 grid(B, Hkv);
+  TORCH_CHECK(q.stride(2) == 1 && q.stride(1) == D, "q heads must be dense");
+  TORCH_CHECK(out.is_contiguous(), "out must be contiguous");
+  const long q_tstride = q.stride(0);
+  // Fixed split count keeps the launch hipGraph-capturable while filling
+  // the chip at small batch (B=1: 8 WGs -> 8*S WGs).
+  const int S = 16;
+  auto f32 = torch::TensorOptions().dtype(torch::kFloat32).device(q.device());
+  torch::Tensor ws_m = torch::empty({B, Hkv, S, G}, f32);
+  torch::Tensor ws_l = torch::empty({B, Hkv, S, G}, f32);
+  torch::Tensor ws_o = torch::empty({(long)B, Hkv, S, G, D}, f32);
+  dim3 grid(B, Hkv, S);
   dim3 block(256);
   auto stream = c10::hip::getCurrentHIPStream();
 #define DECODE_CASE(g, d)                                                    \
-  hipLaunchKernelGGL((attn_decode_kernel<g, d>), grid, block, 0, stream,     \
-                     (short*)out.data_ptr(), (const short*)q.data_ptr(),     \
-                     (const short*)kcache.data_ptr(),                        \
-                     (const short*)vcache.data_ptr(),                        \
-                     (const int*)block_tables.data_ptr(),                    \
-                     (const int*)kv_lens.data_ptr(), max_blocks, Hkv,        \
-                     (float)scale)
+  do {                                                                       \
+    hipLaunchKernelGGL((attn_decode_kernel<g, d>), grid, block, 0, stream,   \
+                       (float*)ws_m.data_ptr(), (float*)ws_l.data_ptr(),     \
+                       (float*)ws_o.data_ptr(), (const short*)q.data_ptr(),  \
+                       (const short*)kcache.data_ptr(),                      \
+                       (const short*)vcache.data_ptr(),                      \
+                       (const int*)block_tables.data_ptr(),                  \
+                       (const int*)kv_lens.data_ptr(), max_blocks, Hkv,      \
+                       (float)scale, q_tstride);                             \
+    hipLaunchKernelGGL((attn_decode_combine<d>), dim3(B, Hq), dim3(256), 0,  \
+                       stream, (short*)out.data_ptr(),                       \
+                       (const float*)ws_m.data_ptr(),                        \
+                       (const float*)ws_l.data_ptr(),                        \
+                       (const float*)ws_o.data_ptr(), S, G, Hkv);            \
+  } while (0)
   if (D == 128 && G == 4) DECODE_CASE(4, 128);
   else if (D == 128 && G == 8) DECODE_CASE(8, 128);
   else if (D == 128 && G == 1) DECODE_CASE(1, 128);
@@ -475,6 +559,9 @@ void attn_prefill_paged(torch::Tensor out, torch::Tensor q, torch::Tensor cu_q,
   const int G = Hq / Hkv;
   const int P = cu_q.size(0) - 1;
   const int max_blocks = block_tables.size(1);
+  TORCH_CHECK(q.stride(2) == 1 && q.stride(1) == D, "q heads must be dense");
+  TORCH_CHECK(out.is_contiguous(), "out must be contiguous");
+  const long q_tstride = q.stride(0);
   TORCH_CHECK(D == 128, "prefill kernel supports head_dim 128");
   TORCH_CHECK(kcache.size(2) == 16, "block_size must be 16");
   TORCH_CHECK(cu_q.scalar_type() == torch::kInt32);
@@ -495,7 +582,7 @@ void attn_prefill_paged(torch::Tensor out, torch::Tensor q, torch::Tensor cu_q,
                      (const short*)vcache.data_ptr(),                         \
                      (const int*)block_tables.data_ptr(),                     \
                      (const int*)kv_lens.data_ptr(), max_blocks, Hkv,         \
-                     (float)scale)
+                     (float)scale, q_tstride)
   if (G == 4) PREFILL_CASE(4);
   else if (G == 8) PREFILL_CASE(8);
   else if (G == 1) PREFILL_CASE(1);

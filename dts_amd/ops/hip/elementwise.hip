@@ -142,7 +142,9 @@ __global__ void rope_kv_append_kernel(
     const float* __restrict__ sin_t, short* __restrict__ k_cache,
     short* __restrict__ v_cache, const long* __restrict__ slots, int T, int Hq,
     int Hk, int D, int BS, long cache_head_stride, long cache_block_stride,
-    int do_rope) {
+    int do_rope, long q_tstride, long k_tstride, long v_tstride) {
+  // q/k/v may be row-strided VIEWS of a fused qkv projection (token-row
+  // stride != H*D): all row addressing goes through *_tstride.
   // unit = one (token, head); heads 0..Hq-1 are q, Hq..Hq+Hk-1 are k,
   // Hq+Hk..Hq+2Hk-1 are v-copy-only
   int unit = blockIdx.x * (blockDim.x / WAVE) + threadIdx.x / WAVE;
@@ -155,8 +157,8 @@ __global__ void rope_kv_append_kernel(
   int half = D / 2;
 
   if (do_rope && h < Hq + Hk) {
-    short* base = (h < Hq) ? q + ((long)t * Hq + h) * D
-                           : k + ((long)t * Hk + (h - Hq)) * D;
+    short* base = (h < Hq) ? q + (long)t * q_tstride + (long)h * D
+                           : k + (long)t * k_tstride + (long)(h - Hq) * D;
     // lane l handles pair indices l, l+WAVE, ... over half
     for (int i = lane; i < half; i += WAVE) {
       float c = cos_t[pos * half + i];
@@ -172,8 +174,8 @@ __global__ void rope_kv_append_kernel(
     // for k units; v units copy straight through)
     bool is_v = h >= Hq + Hk;
     int kvh = is_v ? (h - Hq - Hk) : (h - Hq);
-    const short* src = is_v ? v + ((long)t * Hk + kvh) * D
-                            : k + ((long)t * Hk + kvh) * D;
+    const short* src = is_v ? v + (long)t * v_tstride + (long)kvh * D
+                            : k + (long)t * k_tstride + (long)kvh * D;
     short* cache = is_v ? v_cache : k_cache;
     long slot = slots[t];
     long block = slot / BS, off = slot % BS;
@@ -226,12 +228,22 @@ void silu_mul(torch::Tensor out, torch::Tensor in) {
   HIP_CHECK_LAST();
 }
 
+static inline void check_qkv_layout(const torch::Tensor& t) {
+  // [T, H, D] with contiguous (head, dim) inner layout; token-row stride
+  // may be larger (a view of a fused qkv projection)
+  TORCH_CHECK(t.dim() == 3 && t.stride(2) == 1 && t.stride(1) == t.size(2),
+              "expected [T,H,D] with dense heads; got strides ", t.strides());
+}
+
 void rope_kv_append(torch::Tensor q, torch::Tensor k, torch::Tensor v,
                     torch::Tensor positions, torch::Tensor cos_t,
                     torch::Tensor sin_t, torch::Tensor k_cache,
                     torch::Tensor v_cache, torch::Tensor slots) {
   TORCH_CHECK(q.scalar_type() == torch::kBFloat16);
   TORCH_CHECK(cos_t.scalar_type() == torch::kFloat32);
+  check_qkv_layout(q);
+  check_qkv_layout(k);
+  check_qkv_layout(v);
   int T = q.size(0), Hq = q.size(1), D = q.size(2), Hk = k.size(1);
   int BS = k_cache.size(2);
   long head_stride = k_cache.stride(1), block_stride = k_cache.stride(0);
@@ -247,7 +259,8 @@ void rope_kv_append(torch::Tensor q, torch::Tensor k, torch::Tensor v,
                      (const float*)sin_t.data_ptr(),
                      (short*)k_cache.data_ptr(), (short*)v_cache.data_ptr(),
                      (const long*)slots.data_ptr(), T, Hq, Hk, D, BS,
-                     head_stride, block_stride, 1);
+                     head_stride, block_stride, 1, q.stride(0), k.stride(0),
+                     v.stride(0));
   HIP_CHECK_LAST();
 }
 
@@ -269,6 +282,7 @@ void kv_append(torch::Tensor k, torch::Tensor v, torch::Tensor k_cache,
                      (const long*)slots.data_ptr(), nullptr, nullptr,
                      (short*)k_cache.data_ptr(), (short*)v_cache.data_ptr(),
                      (const long*)slots.data_ptr(), T, 0, Hk, D, BS,
-                     head_stride, block_stride, 0);
+                     head_stride, block_stride, 0, 0, k.stride(0),
+                     v.stride(0));
   HIP_CHECK_LAST();
 }

@@ -144,6 +144,13 @@ class ServingEngine:
         self.graph_steps = 0
         self.tokens_sampled = 0
         self.tokens_prefilled = 0
+        self.t_schedule = 0.0
+        self.t_forward_graph = 0.0
+        self.t_forward_eager = 0.0
+        self.t_sample = 0.0
+        self.t_post = 0.0
+        self.eager_decode_steps = 0
+        self.prefill_steps = 0
 
     # ------------------------------------------------------------------
     def submit_tokens(
@@ -194,21 +201,35 @@ class ServingEngine:
                 self.scheduler.stuck.clear()
         if batch is None:
             return False
+        import time as _time
+
         self.steps += 1
+        t0 = _time.perf_counter()
         if self._graph_runner is not None and self._graph_runner.can_run(batch):
             logits = self._graph_runner.run(batch)
             self.graph_steps += 1
+            t1 = _time.perf_counter()
+            self.t_forward_graph += t1 - t0
         else:
             dev_batch = batch.to(self.device) if self.device != "cpu" else batch
             logits = self.model.forward(dev_batch, self.kv_pool)
+            t1 = _time.perf_counter()
+            self.t_forward_eager += t1 - t0
+            if batch.num_prefill_seqs:
+                self.prefill_steps += 1
+            else:
+                self.eager_decode_steps += 1
         sampled_seqs = batch._sampled_seqs  # type: ignore[attr-defined]
         tokens = self.sampler.sample(logits, sampled_seqs) if sampled_seqs else []
+        t2 = _time.perf_counter()
+        self.t_sample += t2 - t1
         with self._lock:
             self.scheduler.advance_computed(batch)
             self.tokens_prefilled += batch.num_prefill_tokens
             self.tokens_sampled += len(tokens)
             for seq, tok in zip(sampled_seqs, tokens):
                 self._handle_sampled(seq, tok)
+        self.t_post += _time.perf_counter() - t2
         return True
 
     def _handle_sampled(self, seq: Sequence, tok: int) -> None:
@@ -320,6 +341,12 @@ class ServingEngine:
             "tokens_sampled": self.tokens_sampled,
             "tokens_prefilled": self.tokens_prefilled,
             "native_scheduler": type(self.scheduler).__name__ == "NativeScheduler",
+            "eager_decode_steps": self.eager_decode_steps,
+            "prefill_steps": self.prefill_steps,
+            "t_forward_graph_s": round(self.t_forward_graph, 2),
+            "t_forward_eager_s": round(self.t_forward_eager, 2),
+            "t_sample_s": round(self.t_sample, 2),
+            "t_post_s": round(self.t_post, 2),
         }
 
 

@@ -73,11 +73,11 @@ def main():
     )
     torch.manual_seed(0)
 
-    # ---- decode probes at different batch widths
-    for B in (1, 4, 8, 16, 32, 64):
+    # ---- decode probes at different batch widths (and one long-context)
+    for B, plen in ((1, 512), (4, 512), (8, 512), (16, 512), (32, 512), (64, 512), (1, 12288), (6, 2048)):
         futs = []
         for i in range(B):
-            prompt = [int(x) for x in torch.randint(300, 100000, (512,))]
+            prompt = [int(x) for x in torch.randint(300, 100000, (plen,))]
             futs.append(
                 engine.submit_tokens(
                     prompt, SamplingParams(max_tokens=4096, seed=i, temperature=0.7)
@@ -99,10 +99,11 @@ def main():
                     engine._handle_sampled(s, t)
             if all(s.num_computed >= s.num_prompt_tokens for s in engine.scheduler.running):
                 break
-        # timed decode
+        # warm the graph bucket (capture excluded from timing)
+        timed_steps(engine, 4)
         torch.cuda.synchronize()
         r = timed_steps(engine, 64)
-        r["probe"] = f"decode_B{B}"
+        r["probe"] = f"decode_B{B}_kv{plen}"
         r["tok_per_s"] = r["sampled"] / max(
             1e-9, (r["sched_ms"] + r["fwd_ms"] + r["samp_ms"]) / 1e3 * r["steps"]
         )

@@ -69,6 +69,53 @@ class TestSiluMul:
 
 
 class TestRopeKV:
+    def test_rope_kv_append_strided_views(self):
+        """q/k/v as SPLIT VIEWS of a fused qkv tensor (the model's actual
+        layout — token-row stride != H*D). Regression test for the
+        stride-blind addressing bug that passed contiguous unit tests but
+        corrupted the model path."""
+        T, Hq, Hk, D, BS, NB = 33, 32, 8, 128, 16, 8
+        qkv = bf(torch.randn(T, (Hq + 2 * Hk) * D)).to(DEV)
+        q = qkv[:, : Hq * D].view(T, Hq, D)
+        k = qkv[:, Hq * D : (Hq + Hk) * D].view(T, Hk, D)
+        v = qkv[:, (Hq + Hk) * D :].view(T, Hk, D)
+        assert not q.is_contiguous()
+        positions = torch.randint(0, 500, (T,), dtype=torch.long).to(DEV)
+        cos, sin = torch_ref.build_rope_cache(D, 1024, 500000.0)
+        slots = torch.randperm(NB * BS)[:T].to(torch.long).to(DEV)
+        kc = torch.zeros(NB, Hk, BS, D, dtype=torch.bfloat16, device=DEV)
+        vc = torch.zeros_like(kc)
+
+        q_ref, k_ref = torch_ref.rope_apply(
+            q.cpu().float(), k.cpu().float(), positions.cpu(), cos, sin
+        )
+        kc_ref = torch.zeros(NB, Hk, BS, D)
+        vc_ref = torch.zeros(NB, Hk, BS, D)
+        torch_ref.kv_append(k_ref, v.cpu().float(), kc_ref, vc_ref, slots.cpu())
+
+        ext.rope_kv_append(q, k, v, positions, cos.to(DEV), sin.to(DEV), kc, vc, slots)
+        torch.testing.assert_close(q.cpu().float(), q_ref, atol=2e-2, rtol=2e-2)
+        torch.testing.assert_close(kc.cpu().float(), kc_ref, atol=2e-2, rtol=2e-2)
+        torch.testing.assert_close(vc.cpu().float(), vc_ref, atol=2e-2, rtol=2e-2)
+
+        # strided-q attention over the appended cache
+        kvl = torch.tensor([T], dtype=torch.int32, device=DEV)
+        # build a block table covering the scattered slots? use dense cache
+        # instead: run decode attn with q row 0 against a dense table
+        # (covered fully by TestDecodeAttention) — here we only check the
+        # strided-q prefill read path with a simple dense cache:
+        kc2, vc2, bt = make_paged_kv(1, Hk, D, [T])
+        cu_q = torch.tensor([0, T], dtype=torch.int32, device=DEV)
+        q_pos = torch.arange(T, dtype=torch.long, device=DEV)
+        out = torch.empty(T, Hq, D, dtype=torch.bfloat16, device=DEV)
+        scale = 1.0 / math.sqrt(D)
+        ext.attn_prefill_paged(out, q, cu_q, q_pos, kc2, vc2, bt, kvl, scale)
+        ref = torch_ref.attn_prefill_paged(
+            q.cpu().float(), cu_q.cpu(), q_pos.cpu(), kc2.cpu().float(),
+            vc2.cpu().float(), bt.cpu(), kvl.cpu(), scale,
+        )
+        torch.testing.assert_close(out.cpu().float(), ref, atol=2.5e-2, rtol=2.5e-2)
+
     def test_rope_kv_append(self):
         T, Hq, Hk, D, BS, NB = 50, 32, 8, 128, 16, 8
         q = bf(torch.randn(T, Hq, D)).to(DEV)
