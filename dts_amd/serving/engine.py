@@ -427,6 +427,16 @@ class LocalBackend:
         guide = self._build_guide(engine, messages) if params.json_mode else None
         fut = engine.submit_tokens(prompt_ids, params, guide=guide)
         result: GenerationResult = await asyncio.wrap_future(fut)
+        if guide is not None and result.finish_reason == "length":
+            # a structured form that ran out of context/budget produces
+            # truncated JSON on EVERY retry — fail fast instead of letting
+            # the parse-retry ladder regenerate it (ref client.py:148-203
+            # would loop; locally the failure is deterministic)
+            raise ContextLengthError(
+                f"structured generation truncated at {result.completion_tokens} "
+                f"tokens (prompt {result.prompt_tokens}); raise max_position "
+                "or shrink the form"
+            )
         return Completion(
             message=Message.assistant(result.text),
             usage=Usage(
