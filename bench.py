@@ -197,12 +197,7 @@ def main():
                         "seq_len": "chat-scale (user<=160, assistant<=256 "
                         "tok/turn, judge prompt ~order-10k tok)",
                         "parallelism": f"dp{n_gpus}",
-                        "engine": {
-                            "tokens_sampled": stats["tokens_sampled"],
-                            "tokens_prefilled": stats["tokens_prefilled"],
-                            "cache_hit_tokens": stats["cache_hit_tokens"],
-                            "engine_steps": stats["steps"],
-                        },
+                        "engine": stats,
                     },
                 },
             )

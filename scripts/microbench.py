@@ -108,11 +108,10 @@ def main():
             1e-9, (r["sched_ms"] + r["fwd_ms"] + r["samp_ms"]) / 1e3 * r["steps"]
         )
         print(json.dumps(r))
-        # abort the batch
+        # abort the batch (scheduler-agnostic)
         with engine._lock:
             for s in list(engine.scheduler.running):
                 engine.scheduler.abort(s)
-            engine.scheduler.waiting.clear()
             engine._futures.clear()
 
     # ---- prefill probe: one long prompt
