@@ -58,12 +58,15 @@ class MoEMLP(nn.Module):
         weights, experts = torch.topk(torch.softmax(logits, dim=-1), self.top_k)
         weights = weights / weights.sum(dim=-1, keepdim=True)
 
-        out = torch.zeros(T, self.hidden, dtype=torch.float32, device=x.device)
         flat_expert = experts.reshape(-1)  # [T*k]
         flat_tok = (
             torch.arange(T, device=x.device).unsqueeze(1).expand(T, self.top_k).reshape(-1)
         )
         flat_w = weights.reshape(-1)
+        if self.tp.size > 1:
+            return self._forward_ep(x, flat_tok, flat_expert, flat_w)
+
+        out = torch.zeros(T, self.hidden, dtype=torch.float32, device=x.device)
         # grouped execution: one GEMM per locally-resident expert slice
         for e_local in range(self.experts_local):
             e = self.expert_offset + e_local
@@ -75,8 +78,50 @@ class MoEMLP(nn.Module):
             gu = F.linear(xe, self.gate_up_w[e_local])
             ye = F.linear(ops.silu_mul(gu), self.down_w[e_local])
             out.index_add_(0, toks, ye.float() * flat_w[sel].unsqueeze(1).float())
-        out = out.to(x.dtype)
-        return self.tp.all_reduce(out)
+        return out.to(x.dtype)
+
+    def _forward_ep(self, x, flat_tok, flat_expert, flat_w) -> torch.Tensor:
+        """Expert-parallel dispatch: ship each (token, expert) assignment
+        to the owning rank over all-to-all, compute there, ship back
+        (SURVEY.md §2.3 'RCCL all-to-all (MoE expert dispatch)')."""
+        from dts_amd.parallel.ep import all_to_all_rows, exchange_splits
+
+        T = x.shape[0]
+        world = self.tp.size
+        dest = torch.div(flat_expert, self.experts_local, rounding_mode="floor")
+        order = torch.argsort(dest, stable=True)
+        send_splits = [int((dest == r).sum()) for r in range(world)]
+        sorted_tok = flat_tok[order]
+        sorted_expert = flat_expert[order]
+        send_x = x[sorted_tok]
+
+        recv_splits = exchange_splits(send_splits, self.tp.group)
+        recv_x = all_to_all_rows(send_x, send_splits, recv_splits, self.tp.group)
+        # ship the expert ids alongside (same splits, 1 column)
+        recv_e = all_to_all_rows(
+            sorted_expert.unsqueeze(1).to(x.dtype),
+            send_splits,
+            recv_splits,
+            self.tp.group,
+        ).squeeze(1).long()
+
+        ye = torch.zeros_like(recv_x, dtype=torch.float32)
+        for e_local in range(self.experts_local):
+            e = self.expert_offset + e_local
+            sel = recv_e == e
+            if not bool(sel.any()):
+                continue
+            gu = F.linear(recv_x[sel], self.gate_up_w[e_local])
+            ye[sel] = F.linear(ops.silu_mul(gu), self.down_w[e_local]).float()
+
+        back = all_to_all_rows(
+            ye.to(x.dtype), recv_splits, send_splits, self.tp.group
+        )
+        out = torch.zeros(T, self.hidden, dtype=torch.float32, device=x.device)
+        out.index_add_(
+            0, sorted_tok, back.float() * flat_w[order].unsqueeze(1).float()
+        )
+        return out.to(x.dtype)
 
 
 class MixtralLayer(nn.Module):
