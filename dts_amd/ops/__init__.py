@@ -173,3 +173,26 @@ def top_p_sample(logits, temperatures, top_ps, generators=None, seeds=None):
             ext.top_p_sample(out, logits, temperatures, top_ps, seeds)
             return out
     return torch_ref.top_p_sample(logits, temperatures, top_ps, generators)
+
+
+def linear_bf16(x: torch.Tensor, weight: torch.Tensor) -> torch.Tensor:
+    """y = x @ W^T. Skinny decode batches (M<=8, bf16, K%512==0) stream
+    through the custom GEMV kernel (hipBLASLt runs ~50% of the HBM
+    roofline at these M — profiles/); everything else is a hipBLASLt GEMM
+    via F.linear."""
+    if (
+        x.is_cuda
+        and x.dtype == torch.bfloat16
+        and x.dim() == 2
+        and 1 <= x.shape[0] <= 8
+        and x.shape[1] % 512 == 0
+        and weight.stride(1) == 1
+    ):
+        ext = _require_hip()
+        if ext is not None:
+            out = torch.empty(
+                (x.shape[0], weight.shape[0]), dtype=x.dtype, device=x.device
+            )
+            ext.gemv_bf16(out, x, weight)
+            return out
+    return torch.nn.functional.linear(x, weight)

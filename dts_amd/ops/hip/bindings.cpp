@@ -21,6 +21,7 @@ void attn_prefill_paged(torch::Tensor out, torch::Tensor q, torch::Tensor cu_q,
                         torch::Tensor kv_lens, double scale);
 void top_p_sample(torch::Tensor out, torch::Tensor logits, torch::Tensor temps,
                   torch::Tensor top_ps, torch::Tensor seeds);
+void gemv_bf16(torch::Tensor out, torch::Tensor x, torch::Tensor w);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm", &rmsnorm, "RMSNorm (bf16, CDNA4)");
@@ -36,4 +37,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "paged causal prefill attention (bf16, MFMA)");
   m.def("top_p_sample", &top_p_sample,
         "fused temperature softmax + top-p sampling (sort-free)");
+  m.def("gemv_bf16", &gemv_bf16,
+        "skinny-batch (M<=8) bf16 weight-streaming GEMV");
 }

@@ -80,7 +80,11 @@ class ColumnParallelLinear(nn.Module):
         )
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        y = F.linear(x, self.weight, self.bias)
+        from dts_amd import ops
+
+        y = ops.linear_bf16(x, self.weight)
+        if self.bias is not None:
+            y = y + self.bias
         if self.gather_output:
             y = self.tp.all_gather_dim(y, dim=-1)
         return y
@@ -105,7 +109,9 @@ class RowParallelLinear(nn.Module):
         )
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        y = F.linear(x, self.weight)
+        from dts_amd import ops
+
+        y = ops.linear_bf16(x, self.weight)
         y = self.tp.all_reduce(y)
         if self.bias is not None:
             y = y + self.bias
