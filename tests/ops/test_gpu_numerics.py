@@ -213,6 +213,28 @@ class TestPrefillAttention:
         torch.testing.assert_close(out.cpu().float(), ref, atol=2.5e-2, rtol=2.5e-2)
 
 
+class TestGemv:
+    @pytest.mark.parametrize("M,K,N", [(1, 4096, 4096), (4, 4096, 14336),
+                                       (8, 14336, 4096), (3, 4096, 128256)])
+    def test_gemv_matches_dense(self, M, K, N):
+        x = bf(torch.randn(M, K)).to(DEV)
+        w = bf(torch.randn(N, K) / math.sqrt(K)).to(DEV)
+        out = torch.empty(M, N, dtype=torch.bfloat16, device=DEV)
+        ext.gemv_bf16(out, x, w)
+        ref = (x.cpu().float() @ w.cpu().float().T)
+        torch.testing.assert_close(out.cpu().float(), ref, atol=3e-2, rtol=3e-2)
+
+    def test_gemv_strided_x(self):
+        M, K, N = 4, 4096, 512
+        big = bf(torch.randn(M, 2 * K)).to(DEV)
+        x = big[:, :K]  # strided rows
+        w = bf(torch.randn(N, K) / math.sqrt(K)).to(DEV)
+        out = torch.empty(M, N, dtype=torch.bfloat16, device=DEV)
+        ext.gemv_bf16(out, x, w)
+        ref = x.cpu().float() @ w.cpu().float().T
+        torch.testing.assert_close(out.cpu().float(), ref, atol=3e-2, rtol=3e-2)
+
+
 class TestSampling:
     def test_greedy_matches_argmax(self):
         S, V = 7, 128256
