@@ -43,11 +43,13 @@ def init_distributed(backend: Optional[str] = None, timeout_s: int = 600) -> int
     if world <= 1:
         return 1
     if backend is None:
-        backend = "nccl" if torch.cuda.is_available() else "gloo"
+        backend = os.environ.get("DTS_DIST_BACKEND") or (
+            "nccl" if torch.cuda.is_available() else "gloo"
+        )
     os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
     os.environ.setdefault("MASTER_PORT", "29531")
     dist.init_process_group(backend=backend, timeout=timedelta(seconds=timeout_s))
-    if torch.cuda.is_available():
+    if torch.cuda.is_available() and backend == "nccl":
         torch.cuda.set_device(get_local_rank())
     return get_world_size()
 
