@@ -81,7 +81,7 @@ def _worker(rank, world, port, out_q):
 def test_dp_spmd_two_ranks():
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
-    port = 29612
+    port = 29600 + (os.getpid() * 4 + 0) % 800
     procs = [
         ctx.Process(target=_worker, args=(r, 2, port, q)) for r in range(2)
     ]

@@ -74,7 +74,7 @@ def test_ep_matches_dense(tmp_path):
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
     procs = [
-        ctx.Process(target=_worker, args=(r, 2, 29671, str(f), q)) for r in range(2)
+        ctx.Process(target=_worker, args=(r, 2, 29600 + (os.getpid() * 4 + 3) % 800, str(f), q)) for r in range(2)
     ]
     for p in procs:
         p.start()

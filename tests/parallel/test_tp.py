@@ -65,7 +65,7 @@ def _worker(rank, world, port, out_q):
 def test_tp_pair_matches_dense():
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
-    procs = [ctx.Process(target=_worker, args=(r, 2, 29613, q)) for r in range(2)]
+    procs = [ctx.Process(target=_worker, args=(r, 2, 29600 + (os.getpid() * 4 + 1) % 800, q)) for r in range(2)]
     for p in procs:
         p.start()
     results = [q.get(timeout=90) for _ in range(2)]

@@ -103,7 +103,7 @@ def test_tp_engine_matches_dense(tmp_path):
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
     procs = [
-        ctx.Process(target=_worker, args=(r, 2, 29641, str(tmp_path), q))
+        ctx.Process(target=_worker, args=(r, 2, 29600 + (os.getpid() * 4 + 2) % 800, str(tmp_path), q))
         for r in range(2)
     ]
     for p in procs:
