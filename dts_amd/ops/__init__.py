@@ -180,11 +180,13 @@ def linear_bf16(x: torch.Tensor, weight: torch.Tensor) -> torch.Tensor:
     through the custom GEMV kernel (hipBLASLt runs ~50% of the HBM
     roofline at these M — profiles/); everything else is a hipBLASLt GEMM
     via F.linear."""
+    # measured crossover (profiles/): custom GEMV wins at M<=2 (-18% at
+    # B=1); hipBLASLt takes over from M~4 (its tiling amortizes x reads)
     if (
         x.is_cuda
         and x.dtype == torch.bfloat16
         and x.dim() == 2
-        and 1 <= x.shape[0] <= 8
+        and 1 <= x.shape[0] <= 2
         and x.shape[1] % 512 == 0
         and weight.stride(1) == 1
     ):
