@@ -63,13 +63,17 @@ class Sampler:
 
         out: list = [None] * S
         free_rows: list = []
+        guided: list = []  # (i, seq, allowed)
         for i, seq in enumerate(seqs):
             allowed = seq.guide.allowed_tokens() if seq.guide is not None else None
             if allowed is not None:
-                out[i] = self._sample_constrained(logits[i], seq, allowed)
+                guided.append((i, seq, allowed))
             else:
                 free_rows.append(i)
+
         if not free_rows:
+            for i, seq, allowed in guided:
+                out[i] = self._sample_constrained(logits[i], seq, allowed)
             return out
 
         free_seqs = [seqs[i] for i in free_rows]
@@ -96,13 +100,20 @@ class Sampler:
                 dtype=torch.long,
                 device=logits.device,
             )
-            toks = ops.top_p_sample(
+            # enqueue the free-row kernel FIRST (async), then the guided
+            # gathers — their .cpu() syncs then overlap one GPU drain
+            toks_gpu = ops.top_p_sample(
                 free_logits.contiguous(),
                 temps.to(logits.device),
                 top_ps.to(logits.device),
                 seeds=seeds,
-            ).cpu()
+            )
+            for i, seq, allowed in guided:
+                out[i] = self._sample_constrained(logits[i], seq, allowed)
+            toks = toks_gpu.cpu()
         else:
+            for i, seq, allowed in guided:
+                out[i] = self._sample_constrained(logits[i], seq, allowed)
             gens = [self._generator_for(s) for s in free_seqs]
             toks = ops.top_p_sample(free_logits.cpu(), temps, top_ps, generators=gens)
         for i, t in zip(free_rows, toks):
