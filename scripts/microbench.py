@@ -118,29 +118,35 @@ def main():
     from dts_amd.ops import _hip_ext_loader
 
     ext = _hip_ext_loader.load()
-    for M in (1, 2, 4, 8):
-        K, N = 4096, 14336
-        x = torch.randn(M, K, dtype=torch.bfloat16, device="cuda")
-        w = torch.randn(N, K, dtype=torch.bfloat16, device="cuda")
-        out = torch.empty(M, N, dtype=torch.bfloat16, device="cuda")
-        for _ in range(3):
-            ext.gemv_bf16(out, x, w)
-            torch.nn.functional.linear(x, w)
-        torch.cuda.synchronize()
-        t0 = time.perf_counter()
-        for _ in range(50):
-            ext.gemv_bf16(out, x, w)
-        torch.cuda.synchronize()
-        t_gemv = (time.perf_counter() - t0) / 50
-        t0 = time.perf_counter()
-        for _ in range(50):
-            torch.nn.functional.linear(x, w)
-        torch.cuda.synchronize()
-        t_blas = (time.perf_counter() - t0) / 50
-        bw = N * K * 2 / t_gemv / 1e12
-        print(json.dumps({"probe": f"gemv_M{M}", "gemv_us": round(t_gemv * 1e6, 1),
-                          "hipblaslt_us": round(t_blas * 1e6, 1),
-                          "gemv_TBps": round(bw, 2)}))
+    for M in (1, 2, 4, 8, 16, 24, 32, 64):
+        for K, N in ((4096, 14336), (4096, 6144), (14336, 4096)):
+            x = torch.randn(M, K, dtype=torch.bfloat16, device="cuda")
+            w = torch.randn(N, K, dtype=torch.bfloat16, device="cuda")
+            out = torch.empty(M, N, dtype=torch.bfloat16, device="cuda")
+            t_gemv = None
+            if M <= 8:
+                for _ in range(3):
+                    ext.gemv_bf16(out, x, w)
+                torch.cuda.synchronize()
+                t0 = time.perf_counter()
+                for _ in range(50):
+                    ext.gemv_bf16(out, x, w)
+                torch.cuda.synchronize()
+                t_gemv = (time.perf_counter() - t0) / 50
+            for _ in range(3):
+                torch.nn.functional.linear(x, w)
+            torch.cuda.synchronize()
+            t0 = time.perf_counter()
+            for _ in range(50):
+                torch.nn.functional.linear(x, w)
+            torch.cuda.synchronize()
+            t_blas = (time.perf_counter() - t0) / 50
+            print(json.dumps({
+                "probe": f"gemm_M{M}_K{K}_N{N}",
+                "gemv_us": round(t_gemv * 1e6, 1) if t_gemv else None,
+                "hipblaslt_us": round(t_blas * 1e6, 1),
+                "blas_TBps": round(N * K * 2 / t_blas / 1e12, 2),
+            }))
 
     # ---- prefill probe: one long prompt
     prompt = [int(x) for x in torch.randint(300, 100000, (8192,))]
