@@ -264,6 +264,7 @@ class DTSEngine:
             if best
             else "No surviving branch",
         )
+        logger.info("%s", self._token_tracker.summary_str())
         self._emit(
             "phase",
             {

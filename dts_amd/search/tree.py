@@ -136,13 +136,15 @@ class DialogueTree:
         leaves = self.active_leaves()
         if not leaves:
             return None
-        return max(leaves, key=lambda n: n.stats.value_mean)
+        # id tie-break: deterministic across SPMD ranks whose node
+        # insertion orders differ (dts_amd/search/dist_engine.py)
+        return max(leaves, key=lambda n: (n.stats.value_mean, n.id))
 
     def best_leaf_by_score(self) -> Optional[DialogueNode]:
         leaves = self.active_leaves()
         if not leaves:
             return None
-        return max(leaves, key=lambda n: n.stats.aggregated_score)
+        return max(leaves, key=lambda n: (n.stats.aggregated_score, n.id))
 
     def statistics(self) -> dict:
         all_nodes = list(self.nodes.values())

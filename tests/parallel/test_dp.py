@@ -15,7 +15,7 @@ import pytest
 pytestmark = pytest.mark.dist
 
 
-def _worker(rank, world, port, out_q):
+def _worker(rank, world, port, out_q, forking=False):
     os.environ.update(
         {
             "RANK": str(rank),
@@ -44,7 +44,9 @@ def _worker(rank, world, port, out_q):
             first_message="hello can you help me with this?",
             init_branches=4,
             turns_per_branch=2,
-            scoring_mode="absolute",
+            scoring_mode="comparative" if forking else "absolute",
+            user_intents_per_branch=2 if forking else 1,
+            user_variability=forking,
             prune_threshold=0.0,
             seed=3,
         )
@@ -108,3 +110,29 @@ def test_dp_spmd_two_ranks():
     assert r0["n_llm_calls"] < 29
     # together they cover all the work (strategy on rank0 only)
     assert r0["n_llm_calls"] + r1["n_llm_calls"] >= 1 + 4 * 2 * 2 + 4 * 3
+
+
+@pytest.mark.timeout(180)
+def test_dp_spmd_forking_comparative():
+    """Forked children are created on the owner rank and mirrored via
+    payloads; comparative groups shard whole sibling sets."""
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = 29600 + (os.getpid() * 4 + 2) % 800
+    procs = [
+        ctx.Process(target=_worker, args=(r, 2, port, q, True)) for r in range(2)
+    ]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(2):
+        rank, data = q.get(timeout=150)
+        results[rank] = data
+    for p in procs:
+        p.join(timeout=30)
+        assert p.exitcode == 0
+    r0, r1 = results[0], results[1]
+    assert r0["fingerprint"] == r1["fingerprint"]
+    # root + 4 strategy branches + 4*2 forked children
+    assert r0["n_nodes"] == 1 + 4 + 8
+    assert r0["best"] == r1["best"]
