@@ -10,10 +10,11 @@ from __future__ import annotations
 import json
 import sys
 import time
+from pathlib import Path
 
 import torch
 
-sys.path.insert(0, ".")
+sys.path.insert(0, str(Path(__file__).resolve().parent.parent))
 
 from dts_amd.llm.types import SamplingParams  # noqa: E402
 from dts_amd.serving import ServingEngine  # noqa: E402
