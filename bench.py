@@ -37,6 +37,13 @@ def parse_args():
     p.add_argument("--steps", type=int, default=3)
     p.add_argument("--warmup", type=int, default=1)
     p.add_argument("--model", type=str, default="llama-3-8b")
+    p.add_argument(
+        "--judge-model",
+        type=str,
+        default=None,
+        help="serve judges on a second engine (e.g. mixtral-8x7b — "
+        "BASELINE config 5)",
+    )
     p.add_argument("--branches-per-gpu", type=int, default=6)
     p.add_argument("--turns", type=int, default=5)
     p.add_argument("--device", type=str, default=None)
@@ -69,6 +76,7 @@ def build_config(args, world, seed):
         prune_threshold=6.5,
         min_survivors=1,
         max_concurrency=64,
+        judge_model=args.judge_model,
         seed=seed,
         budget=GenerationBudget(
             strategy=3072,
@@ -121,12 +129,23 @@ def main():
         model_name=args.model,
         device=device,
         dtype=dtype,
-        kv_memory_bytes=kv_bytes,
+        kv_memory_bytes=kv_bytes if not args.judge_model else kv_bytes // 2,
         max_batch_tokens=16384,
         max_running=512,
         weight_seed=0,
     )
-    backend = LocalBackend.single(engine, name=args.model)
+    engines = {args.model: engine}
+    if args.judge_model:
+        engines[args.judge_model] = ServingEngine(
+            model_name=args.judge_model,
+            device=device,
+            dtype=dtype,
+            kv_memory_bytes=kv_bytes // 2,
+            max_batch_tokens=16384,
+            max_running=512,
+            weight_seed=1,
+        )
+    backend = LocalBackend(engines, default_model=args.model)
     llm = LLM(backend, default_model=args.model)
 
     def sync():
