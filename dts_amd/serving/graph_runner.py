@@ -23,7 +23,11 @@ import torch
 from dts_amd.serving.batch import ForwardBatch
 from dts_amd.utils.logging import logger
 
-BUCKETS = (1, 2, 4, 8, 16, 24, 32, 48, 64, 96, 128, 192, 256)
+# measured (profiles/microbench_*): graphed decode at B>=24 replays 3-5x
+# slower than eager (suspected capture-unfriendly hipBLASLt algo choice at
+# those M) — graphs pay off at the small-B launch-bound regime anyway, so
+# buckets stop at 16 and larger decode batches run eager.
+BUCKETS = (1, 2, 4, 8, 16)
 
 
 class DecodeGraphRunner:
