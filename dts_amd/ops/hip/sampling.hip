@@ -51,14 +51,10 @@ top_p_sample_kernel(long* __restrict__ out, const float* __restrict__ logits,
   __shared__ int s_argmax;
   __shared__ float s_bins[NBINS];
 
-  const int per = (V + BLOCK - 1) / BLOCK;
-  const int lo_i = tid * per;
-  const int hi_i = min(V, lo_i + per);
-
   // ---- pass 1: max (and argmax for greedy)
   float mymax = -1e30f;
   int myarg = 0;
-  for (int i = lo_i; i < hi_i; ++i) {
+  for (int i = tid; i < V; i += BLOCK) {
     float v = x[i];
     if (v > mymax) {
       mymax = v;
@@ -94,7 +90,7 @@ top_p_sample_kernel(long* __restrict__ out, const float* __restrict__ logits,
   __syncthreads();
   float zpart = 0.f;
   const float bin_scale = NBINS / (-YMIN);  // bins per unit y
-  for (int i = lo_i; i < hi_i; ++i) {
+  for (int i = tid; i < V; i += BLOCK) {
     float y = (x[i] - m) * invT;
     float e = __expf(y);
     zpart += e;
@@ -134,7 +130,7 @@ top_p_sample_kernel(long* __restrict__ out, const float* __restrict__ logits,
   // ---- kept mass per thread + prefix over threads
   __shared__ float s_pref[BLOCK + 1];
   float mymass = 0.f;
-  for (int i = lo_i; i < hi_i; ++i) {
+  for (int i = tid; i < V; i += BLOCK) {
     float y = (x[i] - m) * invT;
     if (y >= tau) mymass += __expf(y);
   }
@@ -159,10 +155,11 @@ top_p_sample_kernel(long* __restrict__ out, const float* __restrict__ logits,
   __syncthreads();
   if (total_kept > 0.f && u >= s_pref[tid] && u < s_pref[tid + 1]) {
     float acc = s_pref[tid];
-    long pick = -1;
-    for (int i = lo_i; i < hi_i; ++i) {
+    long pick = -1, last_kept = -1;
+    for (int i = tid; i < V; i += BLOCK) {
       float y = (x[i] - m) * invT;
       if (y >= tau) {
+        last_kept = i;
         acc += __expf(y);
         if (acc > u) {
           pick = i;
@@ -170,16 +167,7 @@ top_p_sample_kernel(long* __restrict__ out, const float* __restrict__ logits,
         }
       }
     }
-    if (pick < 0) {
-      for (int i = hi_i - 1; i >= lo_i; --i) {
-        float y = (x[i] - m) * invT;
-        if (y >= tau) {
-          pick = i;
-          break;
-        }
-      }
-    }
-    s_result = pick;
+    s_result = (pick >= 0) ? pick : last_kept;
   }
   __syncthreads();
   if (tid == 0) {
