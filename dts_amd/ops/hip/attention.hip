@@ -137,7 +137,15 @@ attn_decode_kernel(float* __restrict__ ws_m,  // [B, Hkv, S, G]
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
     __builtin_amdgcn_wave_barrier();
 
-    // ---- softmax update + V phase; lane owns output dims {2l, 2l+1}
+    // ---- batch the page's V words up front: 16 independent loads in
+    // flight at once (the per-(g,j) load placement serialized an L2/HBM
+    // round trip into the online-softmax chain — ~5 us/page at long kv)
+    int vw[BS];
+#pragma unroll
+    for (int j = 0; j < BS; ++j)
+      vw[j] = (j < valid) ? *(const int*)(vbase + j * D + 2 * lane) : 0;
+
+    // ---- softmax update + V accumulate; lane owns output dims {2l, 2l+1}
 #pragma unroll
     for (int g = 0; g < G; ++g) {
       float tmax = -1e30f;
@@ -160,11 +168,8 @@ attn_decode_kernel(float* __restrict__ ws_m,  // [B, Hkv, S, G]
         lsum[g] += psum;
 #pragma unroll
         for (int j = 0; j < BS; ++j) {
-          if (j < valid) {
-            int vw = *(const int*)(vbase + j * D + 2 * lane);
-            o[g][0] += p[j] * bf2f((short)(vw & 0xffff));
-            o[g][1] += p[j] * bf2f((short)((vw >> 16) & 0xffff));
-          }
+          o[g][0] += p[j] * bf2f((short)(vw[j] & 0xffff));
+          o[g][1] += p[j] * bf2f((short)((vw[j] >> 16) & 0xffff));
         }
       }
     }
