@@ -81,16 +81,24 @@ class _DPEvaluator:
             return await fn(nodes)
         ordered = sorted(nodes, key=lambda n: n.id)
         if mode == "comparative":
-            # shard whole sibling groups so forced ranking stays local
-            groups: dict = {}
-            for n in ordered:
-                groups.setdefault(n.parent_id or "root", []).append(n)
-            keys = sorted(groups)
-            mine: list = []
-            for i, k in enumerate(keys):
-                if self.dp.owns(i):
-                    mine.extend(groups[k])
-            local_scores = await self.ev.evaluate_comparative(mine) if mine else {}
+            # shard ranking CHUNKS (sibling groups, oversized ones split by
+            # the evaluator) so each forced ranking stays on one rank
+            import asyncio as _asyncio
+
+            chunks = self.ev.comparative_chunks(ordered)
+            tasks = []
+            for i, (parent_id, chunk) in enumerate(chunks):
+                if not self.dp.owns(i):
+                    continue
+                if len(chunk) == 1:
+                    tasks.append(self.ev._judge_single_wrapped(chunk[0]))
+                else:
+                    tasks.append(self.ev._judge_group_comparative(parent_id, chunk))
+            local_scores = {}
+            if tasks:
+                for result in await _asyncio.gather(*tasks, return_exceptions=True):
+                    if isinstance(result, dict):
+                        local_scores.update(result)
         else:
             mine = [n for i, n in enumerate(ordered) if self.dp.owns(i)]
             local_scores = await self.ev.evaluate_absolute(mine) if mine else {}

@@ -55,6 +55,14 @@ class TrajectoryEvaluator:
         self._on_usage = on_usage
         self.deep_research_context = deep_research_context
         self._judge_counter = 0
+        # Sibling groups larger than this are ranked in chunks: a single
+        # forced-ranking prompt embeds EVERY sibling trajectory
+        # (ref prompts.py:349-355 — the largest prompt in the system), so
+        # a 48-branch group (weak-scaling bench at N=8) would need ~100k
+        # tokens of context. The reference caps init_branches at 20 and
+        # would overflow its providers' context the same way; chunked
+        # ranking preserves the forced-ranking semantics within chunks.
+        self.max_comparative_group = 8
 
     def set_research_context(self, context: Optional[str]) -> None:
         self.deep_research_context = context
@@ -75,16 +83,32 @@ class TrajectoryEvaluator:
                 node.update_with_evaluation(agg, critiques)
         return scores
 
+    def comparative_chunks(self, nodes: list) -> list:
+        """Deterministic (parent_id, chunk) list: siblings grouped by
+        parent, oversized groups split into <=max_comparative_group chunks
+        (used by both the local and the DP-sharded paths)."""
+        groups: dict = {}
+        for node in nodes:
+            groups.setdefault(node.parent_id or "root", []).append(node)
+        chunks: list = []
+        for parent_id in sorted(groups):
+            group = sorted(groups[parent_id], key=lambda n: n.id)
+            g = self.max_comparative_group
+            for i in range(0, len(group), g):
+                chunk = group[i : i + g]
+                # avoid a trailing singleton when the group splits unevenly
+                if len(chunk) == 1 and chunks and chunks[-1][0] == parent_id:
+                    chunks[-1][1].append(chunk[0])
+                else:
+                    chunks.append((parent_id, chunk))
+        return chunks
+
     async def evaluate_comparative(self, nodes: list) -> dict:
         if len(nodes) <= 1:
             return await self.evaluate_absolute(nodes)
 
-        groups: dict = {}
-        for node in nodes:
-            groups.setdefault(node.parent_id or "root", []).append(node)
-
         tasks = []
-        for parent_id, group in groups.items():
+        for parent_id, group in self.comparative_chunks(nodes):
             if len(group) == 1:
                 tasks.append(self._judge_single_wrapped(group[0]))
             else:
