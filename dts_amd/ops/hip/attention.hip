@@ -48,6 +48,7 @@ attn_decode_kernel(float* __restrict__ ws_m,  // [B, Hkv, S, G]
   constexpr int NWAVE = 4;
   constexpr int LPK = 4;            // lanes per key
   constexpr int DPL = D / LPK;      // dims per lane in K phase
+  constexpr int DPV = D / WAVE;     // output dims per lane (2 @128, 1 @64)
   const int seq = blockIdx.x;
   const int kvh = blockIdx.y;
   const int split = blockIdx.z;
@@ -88,12 +89,13 @@ attn_decode_kernel(float* __restrict__ ws_m,  // [B, Hkv, S, G]
     for (int i = 0; i < DPL / 2; ++i) qp_[g][i] = ((const int*)qp)[i];
   }
 
-  float m[G], lsum[G], o[G][2]; // o: lane owns dims {2*lane, 2*lane+1}
+  float m[G], lsum[G], o[G][DPV]; // lane owns dims [lane*DPV, lane*DPV+DPV)
 #pragma unroll
   for (int g = 0; g < G; ++g) {
     m[g] = -1e30f;
     lsum[g] = 0.f;
-    o[g][0] = o[g][1] = 0.f;
+#pragma unroll
+    for (int d = 0; d < DPV; ++d) o[g][d] = 0.f;
   }
 
   const int* bt = block_tables + (long)seq * max_blocks;
@@ -142,8 +144,16 @@ attn_decode_kernel(float* __restrict__ ws_m,  // [B, Hkv, S, G]
     // round trip into the online-softmax chain — ~5 us/page at long kv)
     int vw[BS];
 #pragma unroll
-    for (int j = 0; j < BS; ++j)
-      vw[j] = (j < valid) ? *(const int*)(vbase + j * D + 2 * lane) : 0;
+    for (int j = 0; j < BS; ++j) {
+      if (j < valid) {
+        if constexpr (DPV == 2)
+          vw[j] = *(const int*)(vbase + j * D + 2 * lane);
+        else
+          vw[j] = (int)*(const unsigned short*)(vbase + j * D + lane);
+      } else {
+        vw[j] = 0;
+      }
+    }
 
     // ---- softmax update + V accumulate; lane owns output dims {2l, 2l+1}
 #pragma unroll
@@ -169,7 +179,8 @@ attn_decode_kernel(float* __restrict__ ws_m,  // [B, Hkv, S, G]
 #pragma unroll
         for (int j = 0; j < BS; ++j) {
           o[g][0] += p[j] * bf2f((short)(vw[j] & 0xffff));
-          o[g][1] += p[j] * bf2f((short)((vw[j] >> 16) & 0xffff));
+          if constexpr (DPV == 2)
+            o[g][1] += p[j] * bf2f((short)((vw[j] >> 16) & 0xffff));
         }
       }
     }
@@ -183,8 +194,8 @@ attn_decode_kernel(float* __restrict__ ws_m,  // [B, Hkv, S, G]
       s_m[wave][g] = m[g];
       s_l[wave][g] = lsum[g];
     }
-    s_o[wave][g][2 * lane] = o[g][0];
-    s_o[wave][g][2 * lane + 1] = o[g][1];
+#pragma unroll
+    for (int d = 0; d < DPV; ++d) s_o[wave][g][lane * DPV + d] = o[g][d];
   }
   __syncthreads();
 
@@ -211,7 +222,7 @@ attn_decode_kernel(float* __restrict__ ws_m,  // [B, Hkv, S, G]
     wo[d0] = acc0;
     wo[d0 + 1] = acc1;
   }
-}
+}  // (D/2 pairs: valid for D=64 too — s_o is [.][.][D])
 
 // Combine the S split partials into the final bf16 output.
 template <int D>
@@ -261,8 +272,8 @@ attn_decode_combine(short* __restrict__ out,        // [B, Hq, D]
 // ---------------------------------------------------------------------------
 
 // Rows per workgroup: 64 = (64/G) positions x G heads. Each wave owns 16
-// rows. KV step = 32 keys (2 pages). D = 128 only (Llama/Mixtral).
-template <int G>
+// rows. KV step = 32 keys (2 pages). D in {64, 128}.
+template <int G, int D>
 __global__ void __launch_bounds__(256)
 attn_prefill_kernel(short* __restrict__ out,      // [Tq, Hq, D]
                     const short* __restrict__ q,  // [Tq, Hq, D]
@@ -275,7 +286,8 @@ attn_prefill_kernel(short* __restrict__ out,      // [Tq, Hq, D]
                     float scale, long q_tstride) {
   constexpr int BS = 16;
   constexpr int KSTEP = 32;
-  constexpr int D = 128;
+  constexpr int KCHUNKS = D / 32;     // mfma k-chunks per QK^T
+  constexpr int CTILES = D / 16;      // 16-col output tiles
   constexpr int LDS_PAD = 8;          // bf16 elements of row padding
   constexpr int LDK = D + LDS_PAD;    // LDS row stride (elements)
   constexpr int ROWS = 64;            // rows per workgroup
@@ -304,8 +316,8 @@ attn_prefill_kernel(short* __restrict__ out,      // [Tq, Hq, D]
   // ---- per-wave row bookkeeping: row_global = wave*16 + r
   // row -> (position, head)
   long my_pos[4];   // absolute position of rows owned by this lane group
-  // Q A-fragments: 4 k-chunks of 32, 8 bf16 each
-  short a_frag[4][8];
+  // Q A-fragments: KCHUNKS k-chunks of 32, 8 bf16 each
+  short a_frag[KCHUNKS][8];
   {
     const int r = lane & 15;            // A row = lane & 15
     const int row_global = wave * 16 + r;
@@ -315,7 +327,7 @@ attn_prefill_kernel(short* __restrict__ out,      // [Tq, Hq, D]
     const int tok = q_start + (valid_row ? pos_local : 0);
     const short* qp = q + (long)tok * q_tstride + (long)(kvh * G + head) * D;
 #pragma unroll
-    for (int kc = 0; kc < 4; ++kc) {
+    for (int kc = 0; kc < KCHUNKS; ++kc) {
       const int kbase = kc * 32 + (lane >> 4) * 8;
       bf16x8 v8 = *(const bf16x8*)(qp + kbase);
 #pragma unroll
@@ -331,10 +343,10 @@ attn_prefill_kernel(short* __restrict__ out,      // [Tq, Hq, D]
   const bool sm_valid = sm_pos < q_len;
   const long sm_abs_pos = sm_valid ? q_pos[q_start + sm_pos] : -1;
 
-  // O accumulator: 8 column tiles of C-frag f32x4
-  f32x4 o_acc[8];
+  // O accumulator: CTILES column tiles of C-frag f32x4
+  f32x4 o_acc[CTILES];
 #pragma unroll
-  for (int ct = 0; ct < 8; ++ct) o_acc[ct] = {0.f, 0.f, 0.f, 0.f};
+  for (int ct = 0; ct < CTILES; ++ct) o_acc[ct] = {0.f, 0.f, 0.f, 0.f};
   float run_m = -1e30f, run_l = 0.f;  // per (lane, sm) bookkeeping below
   // running m/l per row are tracked by the 4-lane row group (same value
   // in all 4 lanes; reduced via shfl)
@@ -378,7 +390,7 @@ attn_prefill_kernel(short* __restrict__ out,      // [Tq, Hq, D]
     for (int st = 0; st < 2; ++st) {
       f32x4 acc = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
-      for (int kc = 0; kc < 4; ++kc) {
+      for (int kc = 0; kc < KCHUNKS; ++kc) {
         // B-frag: col = lane&15 -> key = st*16 + (lane&15),
         //         k  = kc*32 + (lane>>4)*8 + i  (8 consecutive dims)
         const int key = st * 16 + (lane & 15);
@@ -460,7 +472,7 @@ attn_prefill_kernel(short* __restrict__ out,      // [Tq, Hq, D]
         pa1 = pa0;
       }
 #pragma unroll
-      for (int ct = 0; ct < 8; ++ct) {
+      for (int ct = 0; ct < CTILES; ++ct) {
 #pragma unroll
         for (int i = 0; i < 4; ++i) o_acc[ct][i] *= al[i];
         // V B-frag: col = lane&15 -> dim = ct*16 + (lane&15),
@@ -486,7 +498,7 @@ attn_prefill_kernel(short* __restrict__ out,      // [Tq, Hq, D]
       invl[i] = (l > 0.f) ? 1.f / l : 0.f;
     }
 #pragma unroll
-    for (int ct = 0; ct < 8; ++ct) {
+    for (int ct = 0; ct < CTILES; ++ct) {
 #pragma unroll
       for (int i = 0; i < 4; ++i) {
         const int row = (lane >> 4) * 4 + i;
@@ -567,7 +579,7 @@ void attn_prefill_paged(torch::Tensor out, torch::Tensor q, torch::Tensor cu_q,
   TORCH_CHECK(q.stride(2) == 1 && q.stride(1) == D, "q heads must be dense");
   TORCH_CHECK(out.is_contiguous(), "out must be contiguous");
   const long q_tstride = q.stride(0);
-  TORCH_CHECK(D == 128, "prefill kernel supports head_dim 128");
+  TORCH_CHECK(D == 128 || D == 64, "prefill kernel supports head_dim 64/128");
   TORCH_CHECK(kcache.size(2) == 16, "block_size must be 16");
   TORCH_CHECK(cu_q.scalar_type() == torch::kInt32);
   // worst-case tiles per seq: computed on host from q sizes is per-layer
@@ -578,8 +590,8 @@ void attn_prefill_paged(torch::Tensor out, torch::Tensor q, torch::Tensor cu_q,
   dim3 grid(max_tiles, P, Hkv);
   dim3 block(256);
   auto stream = c10::hip::getCurrentHIPStream();
-#define PREFILL_CASE(g)                                                       \
-  hipLaunchKernelGGL((attn_prefill_kernel<g>), grid, block, 0, stream,        \
+#define PREFILL_CASE(g, d)                                                    \
+  hipLaunchKernelGGL((attn_prefill_kernel<g, d>), grid, block, 0, stream,     \
                      (short*)out.data_ptr(), (const short*)q.data_ptr(),      \
                      (const int*)cu_q.data_ptr(),                             \
                      (const long*)q_pos.data_ptr(),                           \
@@ -588,11 +600,12 @@ void attn_prefill_paged(torch::Tensor out, torch::Tensor q, torch::Tensor cu_q,
                      (const int*)block_tables.data_ptr(),                     \
                      (const int*)kv_lens.data_ptr(), max_blocks, Hkv,         \
                      (float)scale, q_tstride)
-  if (G == 4) PREFILL_CASE(4);
-  else if (G == 8) PREFILL_CASE(8);
-  else if (G == 1) PREFILL_CASE(1);
-  else if (G == 2) PREFILL_CASE(2);
-  else TORCH_CHECK(false, "unsupported prefill GQA ratio G=", G);
+  if (D == 128 && G == 4) PREFILL_CASE(4, 128);
+  else if (D == 128 && G == 8) PREFILL_CASE(8, 128);
+  else if (D == 128 && G == 1) PREFILL_CASE(1, 128);
+  else if (D == 128 && G == 2) PREFILL_CASE(2, 128);
+  else if (D == 64 && G == 1) PREFILL_CASE(1, 64);
+  else TORCH_CHECK(false, "unsupported prefill shape G=", G, " D=", D);
 #undef PREFILL_CASE
   HIP_CHECK_LAST();
 }

@@ -5,6 +5,8 @@ void rmsnorm(torch::Tensor out, torch::Tensor x, torch::Tensor w, double eps);
 void fused_add_rmsnorm(torch::Tensor x, torch::Tensor residual,
                        torch::Tensor w, double eps);
 void silu_mul(torch::Tensor out, torch::Tensor in);
+void layernorm(torch::Tensor out, torch::Tensor x, torch::Tensor w,
+               torch::Tensor b, double eps);
 void rope_kv_append(torch::Tensor q, torch::Tensor k, torch::Tensor v,
                     torch::Tensor positions, torch::Tensor cos_t,
                     torch::Tensor sin_t, torch::Tensor k_cache,
@@ -28,6 +30,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fused_add_rmsnorm", &fused_add_rmsnorm,
         "fused residual add + RMSNorm (in-place, bf16)");
   m.def("silu_mul", &silu_mul, "SwiGLU activation (bf16)");
+  m.def("layernorm", &layernorm, "LayerNorm (bf16, GPT-2 path)");
   m.def("rope_kv_append", &rope_kv_append,
         "fused rotate-half RoPE + paged KV append (bf16)");
   m.def("kv_append", &kv_append, "paged KV append without RoPE (bf16)");

@@ -103,6 +103,64 @@ __global__ void fused_add_rmsnorm_kernel(short* __restrict__ x,
 }
 
 // ---------------------------------------------------------------------------
+// LayerNorm (GPT-2 path): y = (x - mu) / sqrt(var + eps) * w + b
+// ---------------------------------------------------------------------------
+
+template <int BLOCK>
+__global__ void layernorm_kernel(short* __restrict__ out,
+                                 const short* __restrict__ x,
+                                 const short* __restrict__ w,
+                                 const short* __restrict__ b, float eps,
+                                 int T, int H) {
+  __shared__ float red[BLOCK / WAVE];
+  for (int row = blockIdx.x; row < T; row += gridDim.x) {
+    const short* xr = x + (long)row * H;
+    short* yr = out + (long)row * H;
+    float sum = 0.f;
+    for (int i = threadIdx.x * 8; i < H; i += BLOCK * 8) {
+      short8v v = *(const short8v*)(xr + i);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) sum += bf2f(v[j]);
+    }
+    sum = wave_sum(sum);
+    if ((threadIdx.x & (WAVE - 1)) == 0) red[threadIdx.x / WAVE] = sum;
+    __syncthreads();
+    float tot = 0.f;
+#pragma unroll
+    for (int i = 0; i < BLOCK / WAVE; ++i) tot += red[i];
+    const float mu = tot / H;
+    __syncthreads();
+    float ss = 0.f;
+    for (int i = threadIdx.x * 8; i < H; i += BLOCK * 8) {
+      short8v v = *(const short8v*)(xr + i);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float d = bf2f(v[j]) - mu;
+        ss += d * d;
+      }
+    }
+    ss = wave_sum(ss);
+    if ((threadIdx.x & (WAVE - 1)) == 0) red[threadIdx.x / WAVE] = ss;
+    __syncthreads();
+    float vtot = 0.f;
+#pragma unroll
+    for (int i = 0; i < BLOCK / WAVE; ++i) vtot += red[i];
+    const float inv = rsqrtf(vtot / H + eps);
+    for (int i = threadIdx.x * 8; i < H; i += BLOCK * 8) {
+      short8v v = *(const short8v*)(xr + i);
+      short8v wv = *(const short8v*)(w + i);
+      short8v bv = *(const short8v*)(b + i);
+      short8v o;
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        o[j] = f2bf((bf2f(v[j]) - mu) * inv * bf2f(wv[j]) + bf2f(bv[j]));
+      *(short8v*)(yr + i) = o;
+    }
+    __syncthreads();
+  }
+}
+
+// ---------------------------------------------------------------------------
 // SwiGLU: out[t, i] = silu(in[t, i]) * in[t, I + i]     in: [T, 2I]
 // ---------------------------------------------------------------------------
 
@@ -213,6 +271,19 @@ void fused_add_rmsnorm(torch::Tensor x, torch::Tensor residual,
                      dim3(256), 0, c10::hip::getCurrentHIPStream(),
                      (short*)x.data_ptr(), (short*)residual.data_ptr(),
                      (const short*)w.data_ptr(), (float)eps, T, H);
+  HIP_CHECK_LAST();
+}
+
+void layernorm(torch::Tensor out, torch::Tensor x, torch::Tensor w,
+               torch::Tensor b, double eps) {
+  TORCH_CHECK(x.scalar_type() == torch::kBFloat16 && x.is_contiguous());
+  int T = x.numel() / x.size(-1), H = x.size(-1);
+  TORCH_CHECK(H % 8 == 0);
+  hipLaunchKernelGGL((layernorm_kernel<256>), dim3(grid_rows(T)), dim3(256),
+                     0, c10::hip::getCurrentHIPStream(),
+                     (short*)out.data_ptr(), (const short*)x.data_ptr(),
+                     (const short*)w.data_ptr(), (const short*)b.data_ptr(),
+                     (float)eps, T, H);
   HIP_CHECK_LAST();
 }
 

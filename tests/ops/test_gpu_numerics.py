@@ -58,6 +58,20 @@ class TestRMSNorm:
         torch.testing.assert_close(x.cpu().float(), ref_norm, atol=3e-2, rtol=3e-2)
 
 
+class TestLayerNorm:
+    def test_layernorm(self):
+        T, H = 65, 768
+        x = bf(torch.randn(T, H)).to(DEV)
+        w = bf(torch.randn(H).abs() + 0.5).to(DEV)
+        b = bf(torch.randn(H) * 0.1).to(DEV)
+        out = torch.empty_like(x)
+        ext.layernorm(out, x, w, b, 1e-5)
+        ref = torch_ref.layernorm(
+            x.cpu().float(), w.cpu().float(), b.cpu().float(), 1e-5
+        )
+        torch.testing.assert_close(out.cpu().float(), ref, atol=3e-2, rtol=3e-2)
+
+
 class TestSiluMul:
     def test_silu_mul(self):
         T, I = 77, 14336
@@ -158,9 +172,9 @@ def make_paged_kv(B, Hkv, D, kv_lens, BS=16):
 
 
 class TestDecodeAttention:
-    @pytest.mark.parametrize("Hq,Hkv", [(32, 8), (8, 1), (8, 8)])
-    def test_decode(self, Hq, Hkv):
-        B, D = 9, 128
+    @pytest.mark.parametrize("Hq,Hkv,D", [(32, 8, 128), (8, 1, 128), (8, 8, 128), (12, 12, 64)])
+    def test_decode(self, Hq, Hkv, D):
+        B = 9
         kv_lens = [1, 5, 16, 17, 63, 64, 100, 255, 1000][:B]
         q = bf(torch.randn(B, Hq, D)).to(DEV)
         kc, vc, bt = make_paged_kv(B, Hkv, D, kv_lens)
@@ -187,7 +201,13 @@ class TestPrefillAttention:
         ],
     )
     def test_prefill(self, Hq, Hkv, q_lens, ctx_lens):
-        D = 128
+        self._run_prefill(Hq, Hkv, q_lens, ctx_lens, 128)
+
+    def test_prefill_d64(self):
+        # GPT-2 head_dim: D=64 template instantiation
+        self._run_prefill(12, 12, [50], [30], 64)
+
+    def _run_prefill(self, Hq, Hkv, q_lens, ctx_lens, D):
         P = len(q_lens)
         kv_lens = [q + c for q, c in zip(q_lens, ctx_lens)]
         Tq = sum(q_lens)
