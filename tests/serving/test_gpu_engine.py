@@ -195,3 +195,20 @@ class TestKVPathConsistency:
         one = self._last_logits(shallow, [prompt])
         split = self._last_logits(shallow, [prompt[:-1], prompt[-1:]])
         self._assert_logits_close(one, split)
+
+
+class TestGPT2OnGPU:
+    def test_gpt2_small_generates(self):
+        """GPT-2-small end-to-end on the GPU path (layernorm kernel,
+        D=64 attention, no-rope KV append)."""
+        eng = ServingEngine(
+            model_name="gpt2-small",
+            device="cuda:0",
+            dtype=torch.bfloat16,
+            num_blocks=2048,
+            block_size=16,
+            weight_seed=9,
+        )
+        res = _gen(eng, list(range(1, 200)), max_tokens=12)
+        eng.stop()
+        assert res.completion_tokens >= 1
