@@ -19,6 +19,8 @@
 #include "common.h"
 #include <torch/extension.h>
 #include <c10/hip/HIPStream.h>
+#include <unordered_map>
+#include <array>
 
 using bf16x8 = __attribute__((ext_vector_type(8))) short;
 using f32x4 = __attribute__((ext_vector_type(4))) float;
@@ -533,12 +535,31 @@ void attn_decode_paged(torch::Tensor out, torch::Tensor q,
   TORCH_CHECK(out.is_contiguous(), "out must be contiguous");
   const long q_tstride = q.stride(0);
   // Fixed split count keeps the launch hipGraph-capturable while filling
-  // the chip at small batch (B=1: 8 WGs -> 8*S WGs).
-  const int S = 16;
-  auto f32 = torch::TensorOptions().dtype(torch::kFloat32).device(q.device());
-  torch::Tensor ws_m = torch::empty({B, Hkv, S, G}, f32);
-  torch::Tensor ws_l = torch::empty({B, Hkv, S, G}, f32);
-  torch::Tensor ws_o = torch::empty({(long)B, Hkv, S, G, D}, f32);
+  // the chip at small batch (B=1: 8 WGs -> 8*S*Hkv WGs) AND bounding the
+  // serial page chain per wave (the decode critical path at long kv).
+  const int S = 32;
+  // Workspace cached per shape: layers within a step run sequentially on
+  // one stream, so one buffer serves all 32 layer calls (and, being
+  // allocated at warm-up time, lives OUTSIDE the graph pool).
+  static std::unordered_map<std::string, std::array<torch::Tensor, 3>> ws_cache;
+  const std::string key = std::to_string(q.get_device()) + ":" +
+                          std::to_string(B) + ":" + std::to_string(Hkv) + ":" +
+                          std::to_string(G) + ":" + std::to_string(D);
+  auto it = ws_cache.find(key);
+  if (it == ws_cache.end()) {
+    auto f32 =
+        torch::TensorOptions().dtype(torch::kFloat32).device(q.device());
+    it = ws_cache
+             .emplace(key,
+                      std::array<torch::Tensor, 3>{
+                          torch::empty({B, Hkv, S, G}, f32),
+                          torch::empty({B, Hkv, S, G}, f32),
+                          torch::empty({(long)B, Hkv, S, G, D}, f32)})
+             .first;
+  }
+  torch::Tensor ws_m = it->second[0];
+  torch::Tensor ws_l = it->second[1];
+  torch::Tensor ws_o = it->second[2];
   dim3 grid(B, Hkv, S);
   dim3 block(256);
   auto stream = c10::hip::getCurrentHIPStream();
