@@ -8,7 +8,7 @@ GPU engine, not an HTTP client, so no pydantic validation cost per call.
 from __future__ import annotations
 
 from dataclasses import dataclass, field
-from typing import Any, Optional
+from typing import Optional
 
 
 @dataclass

@@ -8,8 +8,7 @@ are random-init or loaded from safetensors (dts_amd/models/weights.py).
 
 from __future__ import annotations
 
-from dataclasses import dataclass, field
-from typing import Optional
+from dataclasses import dataclass
 
 
 @dataclass

@@ -10,7 +10,6 @@ row-parallel all-reduce.
 
 from __future__ import annotations
 
-import math
 from typing import Optional
 
 import torch

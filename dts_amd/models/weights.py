@@ -11,7 +11,6 @@ synthetic-bench path since this environment has no network for weights.
 from __future__ import annotations
 
 from pathlib import Path
-from typing import Optional
 
 import torch
 

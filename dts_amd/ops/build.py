@@ -10,7 +10,6 @@ from __future__ import annotations
 
 import os
 import shutil
-import sys
 from pathlib import Path
 
 OPS_DIR = Path(__file__).resolve().parent

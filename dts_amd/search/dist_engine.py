@@ -13,8 +13,6 @@ from typing import Optional
 
 from dts_amd.parallel.dp import (
     DPContext,
-    NodePayload,
-    ScorePayload,
     apply_payload,
     node_to_payload,
     payload_to_score,
@@ -22,7 +20,7 @@ from dts_amd.parallel.dp import (
 )
 from dts_amd.search.engine import DTSEngine
 from dts_amd.search.tree import DialogueTree
-from dts_amd.search.types import DialogueNode, NodeStatus
+from dts_amd.search.types import DialogueNode
 from dts_amd.utils.logging import log_phase
 
 

@@ -7,7 +7,7 @@ create_event_emitter): events must never block or crash the search loop.
 from __future__ import annotations
 
 import asyncio
-from typing import Any, Callable, Optional
+from typing import Callable, Optional
 
 from dts_amd.utils.logging import logger
 

@@ -21,7 +21,7 @@ from enum import Enum
 from pathlib import Path
 from typing import Optional
 
-from dts_amd.llm.types import Message, Usage
+from dts_amd.llm.types import Usage
 
 
 class NodeStatus(str, Enum):

@@ -18,7 +18,7 @@ from typing import Callable, Optional
 
 from fastapi import FastAPI, WebSocket, WebSocketDisconnect
 from fastapi.middleware.cors import CORSMiddleware
-from fastapi.responses import HTMLResponse, JSONResponse
+from fastapi.responses import HTMLResponse
 
 from dts_amd.llm.backend import LLM
 from dts_amd.models.config import MODEL_REGISTRY

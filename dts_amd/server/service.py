@@ -11,7 +11,7 @@ session.
 from __future__ import annotations
 
 import asyncio
-from typing import AsyncIterator, Optional
+from typing import AsyncIterator
 
 from dts_amd.llm.backend import LLM
 from dts_amd.search import DTSConfig, DTSEngine

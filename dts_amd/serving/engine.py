@@ -32,7 +32,7 @@ from dts_amd.serving import structured
 from dts_amd.serving.kv_cache import BlockManager, KVCachePool
 from dts_amd.serving.sampler import Sampler
 from dts_amd.serving.scheduler import Scheduler
-from dts_amd.serving.sequence import Sequence, SeqStatus
+from dts_amd.serving.sequence import Sequence
 from dts_amd.serving.tokenizer import ChatTemplate, SyntheticTokenizer
 from dts_amd.utils.logging import logger
 

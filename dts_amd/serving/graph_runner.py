@@ -16,8 +16,6 @@ attend 1 key of it — never a live block.
 
 from __future__ import annotations
 
-from typing import Optional
-
 import torch
 
 from dts_amd.serving.batch import ForwardBatch

@@ -25,7 +25,7 @@ decode-attention kernel streams through LDS.
 from __future__ import annotations
 
 from collections import OrderedDict
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Optional
 
 import torch

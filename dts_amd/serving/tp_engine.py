@@ -14,8 +14,6 @@ rank's KV pool holds its own kv-head shard at the same block geometry.
 
 from __future__ import annotations
 
-from typing import Optional
-
 import torch
 import torch.distributed as dist
 
