@@ -33,16 +33,25 @@ class Sampler:
     def release(self, seq) -> None:
         self._generators.pop(seq.seq_id, None)
 
-    def _sample_constrained(self, row: torch.Tensor, seq, allowed: list) -> int:
-        """Sample over a SMALL allowed set (constrained decoding): gather
-        the allowed logits instead of masking the full 128k vocab — the
-        set is typically 10-95 bytes, so this is a tiny CPU softmax."""
-        idx = torch.as_tensor(allowed, dtype=torch.long, device=row.device)
-        sub = row[idx].float().cpu()
+    def _gather_constrained(self, logits: torch.Tensor, guided: list):
+        """One flat gather + ONE device sync for ALL guided rows (a
+        per-row .cpu() costs a stream sync each)."""
+        flat_rows: list = []
+        flat_idx: list = []
+        for i, _seq, allowed in guided:
+            flat_rows.extend([i] * len(allowed))
+            flat_idx.extend(allowed)
+        sub = logits[
+            torch.as_tensor(flat_rows, dtype=torch.long, device=logits.device),
+            torch.as_tensor(flat_idx, dtype=torch.long, device=logits.device),
+        ].float()
+        return sub  # caller moves to CPU (the single sync point)
+
+    def _pick_constrained(self, sub_cpu: torch.Tensor, seq, allowed: list) -> int:
         t = seq.params.temperature
         if t <= 0.0:
-            return int(idx[int(sub.argmax())])
-        probs = torch.softmax(sub / t, dim=-1)
+            return int(allowed[int(sub_cpu.argmax())])
+        probs = torch.softmax(sub_cpu / t, dim=-1)
         p = seq.params.top_p
         if p < 1.0 and probs.numel() > 1:
             sp, si = torch.sort(probs, descending=True)
@@ -52,9 +61,15 @@ class Sampler:
             kept = sp * keep
             kept = kept / kept.sum()
             pick = torch.multinomial(kept, 1, generator=self._generator_for(seq))
-            return int(idx[int(si[pick])])
+            return int(allowed[int(si[pick])])
         pick = torch.multinomial(probs, 1, generator=self._generator_for(seq))
-        return int(idx[pick])
+        return int(allowed[int(pick)])
+
+    def _sample_constrained(self, row: torch.Tensor, seq, allowed: list) -> int:
+        """Single-row convenience path (CPU backend / no free rows)."""
+        idx = torch.as_tensor(allowed, dtype=torch.long, device=row.device)
+        sub = row[idx].float().cpu()
+        return self._pick_constrained(sub, seq, allowed)
 
     def sample(self, logits: torch.Tensor, seqs: list) -> list:
         """logits [S, V] fp32 for the sampled rows; returns token ids."""
@@ -100,16 +115,23 @@ class Sampler:
                 dtype=torch.long,
                 device=logits.device,
             )
-            # enqueue the free-row kernel FIRST (async), then the guided
-            # gathers — their .cpu() syncs then overlap one GPU drain
+            # enqueue the free-row kernel FIRST (async), then ONE flat
+            # gather for all guided rows; a single .cpu() syncs both
             toks_gpu = ops.top_p_sample(
                 free_logits.contiguous(),
                 temps.to(logits.device),
                 top_ps.to(logits.device),
                 seeds=seeds,
             )
-            for i, seq, allowed in guided:
-                out[i] = self._sample_constrained(logits[i], seq, allowed)
+            if guided:
+                sub_cpu = self._gather_constrained(logits, guided).cpu()
+                off = 0
+                for i, seq, allowed in guided:
+                    n_a = len(allowed)
+                    out[i] = self._pick_constrained(
+                        sub_cpu[off : off + n_a], seq, allowed
+                    )
+                    off += n_a
             toks = toks_gpu.cpu()
         else:
             for i, seq, allowed in guided:
