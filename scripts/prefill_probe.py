@@ -1,0 +1,56 @@
+"""Isolated prefill-attention kernel probe (TF + for PMC runs).
+
+One sequence, q_len=N self-attention (causal), Llama-3-8B shapes.
+FLOPs = 2 ops x (QK^T + PV) x Hq x N^2/2 x D.
+"""
+
+import json
+import sys
+import time
+from pathlib import Path
+
+import torch
+
+sys.path.insert(0, str(Path(__file__).resolve().parent.parent))
+
+from dts_amd.ops import _hip_ext_loader  # noqa: E402
+
+ext = _hip_ext_loader.load()
+
+
+def probe(N=4096, Hq=32, Hkv=8, D=128, iters=20):
+    torch.manual_seed(0)
+    BS = 16
+    n_blocks = N // BS + 2
+    q = torch.randn(N, Hq, D, dtype=torch.bfloat16, device="cuda")
+    kc = torch.randn(n_blocks, Hkv, BS, D, dtype=torch.bfloat16, device="cuda")
+    vc = torch.randn_like(kc)
+    bt = torch.arange(1, n_blocks + 1, dtype=torch.int32, device="cuda").unsqueeze(0)
+    kvl = torch.tensor([N], dtype=torch.int32, device="cuda")
+    cu_q = torch.tensor([0, N], dtype=torch.int32, device="cuda")
+    q_pos = torch.arange(N, dtype=torch.long, device="cuda")
+    out = torch.empty_like(q)
+    scale = D ** -0.5
+    for _ in range(3):
+        ext.attn_prefill_paged(out, q, cu_q, q_pos, kc, vc, bt, kvl, scale)
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(iters):
+        ext.attn_prefill_paged(out, q, cu_q, q_pos, kc, vc, bt, kvl, scale)
+    torch.cuda.synchronize()
+    dt = (time.perf_counter() - t0) / iters
+    flops = 2 * 2 * Hq * (N * N / 2) * D
+    print(
+        json.dumps(
+            {
+                "probe": f"prefill_attn_N{N}",
+                "us": round(dt * 1e6, 1),
+                "TF": round(flops / dt / 1e12, 1),
+            }
+        )
+    )
+
+
+if __name__ == "__main__":
+    for n in (2048, 4096, 8192):
+        probe(N=n)
