@@ -309,6 +309,7 @@ class ServingEngine:
                         fut = self._futures.pop(seq.seq_id, None)
                         if fut and not fut.done():
                             fut.set_exception(e)
+                        self.sampler.release(seq)
                         self.scheduler.abort(seq)
 
     def start(self) -> None:
