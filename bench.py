@@ -46,6 +46,9 @@ def parse_args():
     )
     p.add_argument("--branches-per-gpu", type=int, default=6)
     p.add_argument("--turns", type=int, default=5)
+    p.add_argument("--rounds", type=int, default=1)
+    p.add_argument("--scoring", type=str, default="comparative",
+                   choices=["comparative", "absolute"])
     p.add_argument("--device", type=str, default=None)
     p.add_argument("--kv-frac", type=float, default=0.75)
     return p.parse_args()
@@ -72,7 +75,7 @@ def build_config(args, world, seed):
         turns_per_branch=args.turns,
         user_intents_per_branch=1,
         user_variability=False,
-        scoring_mode="comparative",
+        scoring_mode=args.scoring,
         prune_threshold=6.5,
         min_survivors=1,
         max_concurrency=64,
@@ -89,11 +92,11 @@ def build_config(args, world, seed):
     )
 
 
-async def run_search(llm, cfg, dp):
+async def run_search(llm, cfg, dp, rounds=1):
     from dts_amd.search.dist_engine import DistributedDTSEngine
 
     engine = DistributedDTSEngine(llm, cfg, dp=dp)
-    result = await engine.run(rounds=1)
+    result = await engine.run(rounds=rounds)
     scored = sum(
         1 for n in result.all_nodes if n.strategy is not None and n.stats.judge_scores
     )
@@ -159,7 +162,7 @@ def main():
         cfg = type(cfg)(**{**cfg.__dict__,
                            "goal": f"{GOAL} (scenario {step_idx})",
                            "first_message": f"{FIRST_MESSAGE} (case {step_idx})"})
-        scored, _ = asyncio.run(run_search(llm, cfg, dp))
+        scored, _ = asyncio.run(run_search(llm, cfg, dp, rounds=args.rounds))
         return scored
 
     # warmup
@@ -210,8 +213,8 @@ def main():
                         "model": args.model,
                         "init_branches": args.branches_per_gpu * world,
                         "turns_per_branch": args.turns,
-                        "rounds_per_step": 1,
-                        "scoring_mode": "comparative",
+                        "rounds_per_step": args.rounds,
+                        "scoring_mode": args.scoring,
                         "global_batch": args.branches_per_gpu * world,
                         "seq_len": "chat-scale (user<=160, assistant<=256 "
                         "tok/turn, judge prompt ~order-10k tok)",
