@@ -177,13 +177,14 @@ def main():
     sync()
     elapsed = time.perf_counter() - t0
 
-    # max over ranks
+    # max over ranks (nccl collectives need device tensors)
     if world > 1:
         import torch.distributed as dist
 
-        t = torch.tensor([elapsed], dtype=torch.float64)
+        dev = "cuda" if dist.get_backend() == "nccl" else "cpu"
+        t = torch.tensor([elapsed], dtype=torch.float64, device=dev)
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
-        elapsed = float(t[0])
+        elapsed = float(t.cpu()[0])
 
     n_gpus = world if use_gpu else args.gpus
     # total_scored counts the WHOLE tree's scored branches (identical on
