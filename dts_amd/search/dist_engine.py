@@ -48,7 +48,6 @@ class _DPSimulator:
         merged: list = []
         all_payloads = [p for rank_list in gathered for p in rank_list]
         all_payloads.sort(key=lambda p: (p.parent_id or "", p.node_id))
-        local_ids = {n.id for n in expanded_local}
         for p in all_payloads:
             node = apply_payload(tree, p)
             merged.append(node)

@@ -65,9 +65,3 @@ class Sequence:
     def blocks_needed(self, block_size: int, extra_tokens: int = 0) -> int:
         total = len(self.tokens) + extra_tokens
         return (total + block_size - 1) // block_size
-
-    @property
-    def is_prefilling(self) -> bool:
-        return self.num_computed < len(self.tokens) - 1 or (
-            self.num_computed < len(self.tokens) and not self.output_tokens
-        )
