@@ -21,6 +21,7 @@
 #include <c10/hip/HIPStream.h>
 #include <unordered_map>
 #include <array>
+#include <cstdlib>
 
 using bf16x8 = __attribute__((ext_vector_type(8))) short;
 using f32x4 = __attribute__((ext_vector_type(4))) float;
@@ -166,8 +167,8 @@ attn_decode_kernel(float* __restrict__ ws_m,  // [B, Hkv, S, G]
       float m_new = fmaxf(m[g], tmax);
       if (m_new > -1e30f) {
         float alpha = __expf(m[g] - m_new);
-        o[g][0] *= alpha;
-        o[g][1] *= alpha;
+#pragma unroll
+        for (int d = 0; d < DPV; ++d) o[g][d] *= alpha;
         lsum[g] *= alpha;
         m[g] = m_new;
         float p[BS];
@@ -275,7 +276,11 @@ attn_decode_combine(short* __restrict__ out,        // [B, Hq, D]
 
 // Rows per workgroup: 64 = (64/G) positions x G heads. Each wave owns 16
 // rows. KV step = 32 keys (2 pages). D in {64, 128}.
-template <int G, int D>
+// SWZ: XOR-swizzle the K/V LDS images (guide T2): element column block
+// col' = col ^ ((key & 7) << 3), applied identically at the cooperative
+// store and every read — spreads the b128 B-frag lane groups across bank
+// slots. A/B-selectable per call (scripts/prefill_probe.py).
+template <int G, int D, bool SWZ>
 __global__ void __launch_bounds__(256)
 attn_prefill_kernel(short* __restrict__ out,      // [Tq, Hq, D]
                     const short* __restrict__ q,  // [Tq, Hq, D]
@@ -310,6 +315,11 @@ attn_prefill_kernel(short* __restrict__ out,      // [Tq, Hq, D]
 
   __shared__ short k_lds[KSTEP][LDK];
   __shared__ short v_lds[KSTEP][LDK];
+  // element-column swizzle within a row (8-element blocks stay contiguous)
+  auto swz_col = [](int key, int col) -> int {
+    if constexpr (SWZ) return ((col & ~7) ^ ((key & 7) << 3)) | (col & 7);
+    else return col;
+  };
   __shared__ float s_scores[4][16][KSTEP];
   __shared__ float s_alpha[4][16];
   __shared__ float s_rowl[4][16];
@@ -370,18 +380,19 @@ attn_prefill_kernel(short* __restrict__ out,      // [Tq, Hq, D]
         const int page = kv_base / BS + key / BS;
         const int koff = key % BS;
         const int kglob = kv_base + key;
+        const int dst = swz_col(key, d0);
         if (kglob < kv_len) {
           const long blk = bt[page];
           const short* kp =
               kcache + ((blk * Hkv + kvh) * BS + koff) * (long)D + d0;
           const short* vp =
               vcache + ((blk * Hkv + kvh) * BS + koff) * (long)D + d0;
-          *(bf16x8*)(&k_lds[key][d0]) = *(const bf16x8*)kp;
-          *(bf16x8*)(&v_lds[key][d0]) = *(const bf16x8*)vp;
+          *(bf16x8*)(&k_lds[key][dst]) = *(const bf16x8*)kp;
+          *(bf16x8*)(&v_lds[key][dst]) = *(const bf16x8*)vp;
         } else {
           bf16x8 z = {0, 0, 0, 0, 0, 0, 0, 0};
-          *(bf16x8*)(&k_lds[key][d0]) = z;
-          *(bf16x8*)(&v_lds[key][d0]) = z;
+          *(bf16x8*)(&k_lds[key][dst]) = z;
+          *(bf16x8*)(&v_lds[key][dst]) = z;
         }
       }
     }
@@ -397,7 +408,7 @@ attn_prefill_kernel(short* __restrict__ out,      // [Tq, Hq, D]
         //         k  = kc*32 + (lane>>4)*8 + i  (8 consecutive dims)
         const int key = st * 16 + (lane & 15);
         const int d0 = kc * 32 + (lane >> 4) * 8;
-        bf16x8 b8 = *(const bf16x8*)(&k_lds[key][d0]);
+        bf16x8 b8 = *(const bf16x8*)(&k_lds[key][swz_col(key, d0)]);
         acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             *(bf16x8*)a_frag[kc], b8, acc, 0, 0, 0);
       }
@@ -483,7 +494,8 @@ attn_prefill_kernel(short* __restrict__ out,      // [Tq, Hq, D]
         const int dim = ct * 16 + (lane & 15);
         const int kk0 = (lane >> 4) * 8;
 #pragma unroll
-        for (int i = 0; i < 8; ++i) vb[i] = v_lds[kk0 + i][dim];
+        for (int i = 0; i < 8; ++i)
+          vb[i] = v_lds[kk0 + i][swz_col(kk0 + i, dim)];
         o_acc[ct] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pa0, vb,
                                                             o_acc[ct], 0, 0, 0);
       }
@@ -591,7 +603,14 @@ void attn_decode_paged(torch::Tensor out, torch::Tensor q,
 void attn_prefill_paged(torch::Tensor out, torch::Tensor q, torch::Tensor cu_q,
                         torch::Tensor q_pos, torch::Tensor kcache,
                         torch::Tensor vcache, torch::Tensor block_tables,
-                        torch::Tensor kv_lens, double scale) {
+                        torch::Tensor kv_lens, double scale, int64_t swz) {
+  if (swz < 0) {
+    static int env_swz = [] {
+      const char* e = getenv("DTS_PREFILL_SWZ");
+      return (e && e[0] == '0') ? 0 : 1;
+    }();
+    swz = env_swz;
+  }
   const int Hq = q.size(1), D = q.size(2);
   const int Hkv = kcache.size(1);
   const int G = Hq / Hkv;
@@ -612,15 +631,30 @@ void attn_prefill_paged(torch::Tensor out, torch::Tensor q, torch::Tensor cu_q,
   dim3 block(256);
   auto stream = c10::hip::getCurrentHIPStream();
 #define PREFILL_CASE(g, d)                                                    \
-  hipLaunchKernelGGL((attn_prefill_kernel<g, d>), grid, block, 0, stream,     \
-                     (short*)out.data_ptr(), (const short*)q.data_ptr(),      \
-                     (const int*)cu_q.data_ptr(),                             \
-                     (const long*)q_pos.data_ptr(),                           \
-                     (const short*)kcache.data_ptr(),                         \
-                     (const short*)vcache.data_ptr(),                         \
-                     (const int*)block_tables.data_ptr(),                     \
-                     (const int*)kv_lens.data_ptr(), max_blocks, Hkv,         \
-                     (float)scale, q_tstride)
+  do {                                                                        \
+    if (swz)                                                                  \
+      hipLaunchKernelGGL((attn_prefill_kernel<g, d, true>), grid, block, 0,   \
+                         stream, (short*)out.data_ptr(),                      \
+                         (const short*)q.data_ptr(),                          \
+                         (const int*)cu_q.data_ptr(),                         \
+                         (const long*)q_pos.data_ptr(),                       \
+                         (const short*)kcache.data_ptr(),                     \
+                         (const short*)vcache.data_ptr(),                     \
+                         (const int*)block_tables.data_ptr(),                 \
+                         (const int*)kv_lens.data_ptr(), max_blocks, Hkv,     \
+                         (float)scale, q_tstride);                            \
+    else                                                                      \
+      hipLaunchKernelGGL((attn_prefill_kernel<g, d, false>), grid, block, 0,  \
+                         stream, (short*)out.data_ptr(),                      \
+                         (const short*)q.data_ptr(),                          \
+                         (const int*)cu_q.data_ptr(),                         \
+                         (const long*)q_pos.data_ptr(),                       \
+                         (const short*)kcache.data_ptr(),                     \
+                         (const short*)vcache.data_ptr(),                     \
+                         (const int*)block_tables.data_ptr(),                 \
+                         (const int*)kv_lens.data_ptr(), max_blocks, Hkv,     \
+                         (float)scale, q_tstride);                            \
+  } while (0)
   if (D == 128 && G == 4) PREFILL_CASE(4, 128);
   else if (D == 128 && G == 8) PREFILL_CASE(8, 128);
   else if (D == 128 && G == 1) PREFILL_CASE(1, 128);

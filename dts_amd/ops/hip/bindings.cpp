@@ -20,7 +20,7 @@ void attn_decode_paged(torch::Tensor out, torch::Tensor q,
 void attn_prefill_paged(torch::Tensor out, torch::Tensor q, torch::Tensor cu_q,
                         torch::Tensor q_pos, torch::Tensor kcache,
                         torch::Tensor vcache, torch::Tensor block_tables,
-                        torch::Tensor kv_lens, double scale);
+                        torch::Tensor kv_lens, double scale, int64_t swz);
 void top_p_sample(torch::Tensor out, torch::Tensor logits, torch::Tensor temps,
                   torch::Tensor top_ps, torch::Tensor seeds);
 void gemv_bf16(torch::Tensor out, torch::Tensor x, torch::Tensor w);
@@ -37,7 +37,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_decode_paged", &attn_decode_paged,
         "paged GQA decode attention (bf16, wave-per-kv-head)");
   m.def("attn_prefill_paged", &attn_prefill_paged,
-        "paged causal prefill attention (bf16, MFMA)");
+        "paged causal prefill attention (bf16, MFMA)", py::arg("out"),
+        py::arg("q"), py::arg("cu_q"), py::arg("q_pos"), py::arg("kcache"),
+        py::arg("vcache"), py::arg("block_tables"), py::arg("kv_lens"),
+        py::arg("scale"), py::arg("swz") = -1);
   m.def("top_p_sample", &top_p_sample,
         "fused temperature softmax + top-p sampling (sort-free)");
   m.def("gemv_bf16", &gemv_bf16,

@@ -31,21 +31,34 @@ def probe(N=4096, Hq=32, Hkv=8, D=128, iters=20):
     q_pos = torch.arange(N, dtype=torch.long, device="cuda")
     out = torch.empty_like(q)
     scale = D ** -0.5
-    for _ in range(3):
-        ext.attn_prefill_paged(out, q, cu_q, q_pos, kc, vc, bt, kvl, scale)
-    torch.cuda.synchronize()
-    t0 = time.perf_counter()
-    for _ in range(iters):
-        ext.attn_prefill_paged(out, q, cu_q, q_pos, kc, vc, bt, kvl, scale)
-    torch.cuda.synchronize()
-    dt = (time.perf_counter() - t0) / iters
     flops = 2 * 2 * Hq * (N * N / 2) * D
+    results = {}
+    # within-probe interleaved A/B (guide §5.4 rule 24): 6 rounds each
+    for swz in (0, 1, 0, 1, 0, 1):
+        for _ in range(2):
+            ext.attn_prefill_paged(out, q, cu_q, q_pos, kc, vc, bt, kvl,
+                                   scale, swz)
+        torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        for _ in range(iters):
+            ext.attn_prefill_paged(out, q, cu_q, q_pos, kc, vc, bt, kvl,
+                                   scale, swz)
+        torch.cuda.synchronize()
+        dt = (time.perf_counter() - t0) / iters
+        results.setdefault(swz, []).append(round(flops / dt / 1e12, 1))
+    # numerics: both variants must agree bitwise (same math, layout only)
+    out0 = torch.empty_like(q)
+    out1 = torch.empty_like(q)
+    ext.attn_prefill_paged(out0, q, cu_q, q_pos, kc, vc, bt, kvl, scale, 0)
+    ext.attn_prefill_paged(out1, q, cu_q, q_pos, kc, vc, bt, kvl, scale, 1)
+    biteq = bool(torch.equal(out0, out1))
     print(
         json.dumps(
             {
                 "probe": f"prefill_attn_N{N}",
-                "us": round(dt * 1e6, 1),
-                "TF": round(flops / dt / 1e12, 1),
+                "TF_noswz": results[0],
+                "TF_swz": results[1],
+                "swz_bitexact": biteq,
             }
         )
     )
