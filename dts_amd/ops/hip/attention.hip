@@ -9,10 +9,11 @@
 // Prefill (chunked, causal, attends cached prefix + own chunk): MFMA
 // (v_mfma_f32_16x16x32_bf16) flash-style kernel; one 4-wave workgroup per
 // (sequence, 64-row q-tile, kv-head) where a row is a (position, q-head)
-// pair; K/V pages staged in LDS (padded rows against bank conflicts),
-// online softmax with per-row running max/sum. Correctness-first
-// structure; the optimization ladder (XOR-swizzle, tr_b16 V reads,
-// 8-phase pipeline) lands on top after rocprof baselines.
+// pair; K/V pages staged in LDS (+8-element row padding against bank
+// conflicts — the XOR-swizzle alternative measured NEGATIVE here, see the
+// SWZ template note below), online softmax with per-row running max/sum.
+// Remaining ladder (tr_b16 V reads, async-stage split, 8-phase pipeline)
+// is round-2 work; prefill attention is ~8% of the search wall.
 //
 // Semantics match dts_amd/ops/torch_ref.py attn_*_paged.
 
@@ -605,9 +606,14 @@ void attn_prefill_paged(torch::Tensor out, torch::Tensor q, torch::Tensor cu_q,
                         torch::Tensor vcache, torch::Tensor block_tables,
                         torch::Tensor kv_lens, double scale, int64_t swz) {
   if (swz < 0) {
+    // default OFF: within-probe A/B (profiles/prefill_swizzle_ab_r01.log)
+    // measured the XOR swizzle at -24..-30% on this kernel — the +8-element
+    // row padding (272 B stride = 68 dwords, coprime-ish with the 64-bank
+    // row) already spreads the b128 lane groups, so the swizzle only adds
+    // per-access address VALU. Kept behind DTS_PREFILL_SWZ=1 for re-runs.
     static int env_swz = [] {
       const char* e = getenv("DTS_PREFILL_SWZ");
-      return (e && e[0] == '0') ? 0 : 1;
+      return (e && e[0] == '1') ? 1 : 0;
     }();
     swz = env_swz;
   }
