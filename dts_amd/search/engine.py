@@ -60,6 +60,7 @@ class DTSEngine:
             max_concurrency=config.max_concurrency,
             on_usage=self._track_usage,
             max_tokens=config.budget.strategy,
+            intent_max_tokens=config.budget.intent,
             seed=config.seed,
         )
         self._simulator = ConversationSimulator(

@@ -37,6 +37,7 @@ class StrategyGenerator:
         max_concurrency: int = 16,
         on_usage: Optional[Callable[[Any, str], None]] = None,
         max_tokens: int = 1024,
+        intent_max_tokens: Optional[int] = None,
         seed: Optional[int] = None,
     ) -> None:
         self.llm = llm
@@ -44,6 +45,7 @@ class StrategyGenerator:
         self.model = model
         self.temperature = temperature
         self.max_tokens = max_tokens
+        self.intent_max_tokens = intent_max_tokens or max_tokens
         self.seed = seed
         self._sem = asyncio.Semaphore(max_concurrency)
         self._on_usage = on_usage
@@ -75,7 +77,9 @@ class StrategyGenerator:
             conversation_goal=self.goal,
             conversation_history=format_message_history(history),
         )
-        result = await self._call_json(system, user, phase="intent")
+        result = await self._call_json(
+            system, user, phase="intent", max_tokens=self.intent_max_tokens
+        )
         if not result:
             raise RuntimeError("Intent generation failed after retries")
         intents = []
@@ -94,18 +98,22 @@ class StrategyGenerator:
                 logger.warning("Failed to parse intent: %s", e)
         return intents
 
-    async def _call_json(self, system: str, user: str, phase: str) -> Optional[dict]:
+    async def _call_json(
+        self, system: str, user: str, phase: str, max_tokens: Optional[int] = None
+    ) -> Optional[dict]:
         async with self._sem:
-            return await self._call_json_inner(system, user, phase)
+            return await self._call_json_inner(system, user, phase, max_tokens)
 
     @llm_retry(max_attempts=3)
-    async def _call_json_inner(self, system: str, user: str, phase: str) -> Optional[dict]:
+    async def _call_json_inner(
+        self, system: str, user: str, phase: str, max_tokens: Optional[int] = None
+    ) -> Optional[dict]:
         completion = await self.llm.complete(
             [Message.system(system), Message.user(user)],
             model=self.model,
             temperature=self.temperature,
             structured_output=True,
-            max_tokens=self.max_tokens,
+            max_tokens=max_tokens or self.max_tokens,
             seed=self.seed,
         )
         if self._on_usage:
