@@ -83,3 +83,41 @@ class TestEndToEnd:
         # comparative scores follow the forced-ranking schedule shape
         for n in scored:
             assert 0.0 <= n.stats.aggregated_score <= 9.9
+
+
+class TestGPT2Plumbing:
+    """BASELINE.json config 1 shape: a GPT-2-family model serving every
+    phase on CPU (tiny variant for test speed; gpt2-small runs the same
+    code path)."""
+
+    def test_gpt2_absolute_search(self):
+        import torch
+
+        from dts_amd.serving import LocalBackend, ServingEngine
+
+        engine = ServingEngine(
+            model_name="gpt2-tiny",
+            device="cpu",
+            dtype=torch.float32,
+            num_blocks=2048,
+            block_size=16,
+            max_batch_tokens=1024,
+            weight_seed=3,
+        )
+        backend = LocalBackend.single(engine, name="gpt2-tiny")
+        llm = LLM(backend, default_model="gpt2-tiny")
+        cfg = DTSConfig(
+            goal="Explain compound interest",
+            first_message="What is compound interest?",
+            init_branches=2,
+            turns_per_branch=2,
+            scoring_mode="absolute",
+            prune_threshold=0.0,
+            seed=2,
+            budget=small_budget(),
+        )
+        result = asyncio.run(DTSEngine(llm, cfg).run(rounds=1))
+        branches = [n for n in result.all_nodes if n.strategy is not None]
+        assert len(branches) == 2
+        assert all(len(n.stats.judge_scores) == 3 for n in branches)
+        backend.shutdown()
