@@ -180,15 +180,25 @@ def linear_bf16(x: torch.Tensor, weight: torch.Tensor) -> torch.Tensor:
     through the custom GEMV kernel (hipBLASLt runs ~50% of the HBM
     roofline at these M — profiles/); everything else is a hipBLASLt GEMM
     via F.linear."""
-    # measured crossover (profiles/): custom GEMV wins at M<=2 (-18% at
-    # B=1); hipBLASLt takes over from M~4 (its tiling amortizes x reads)
+    # measured crossover (profiles/ GEMM sweep): hipBLASLt has a ~22 us
+    # per-GEMM floor at decode M regardless of weight size, so the custom
+    # GEMV wins (a) at M<=2 on every shape (19.4 us = 6.05 TB/s on the
+    # 117 MB gate_up), and (b) up to M=8 on the smaller qkv/o weights
+    # (<=72 MB streams in ~9-17 us); hipBLASLt keeps the large weights at
+    # M>=3 where its tiling amortizes the x re-reads.
     if (
         x.is_cuda
         and x.dtype == torch.bfloat16
         and x.dim() == 2
-        and 1 <= x.shape[0] <= 2
         and x.shape[1] % 512 == 0
         and weight.stride(1) == 1
+        and (
+            1 <= x.shape[0] <= 2
+            or (
+                x.shape[0] <= 8
+                and weight.shape[0] * weight.shape[1] * 2 <= (72 << 20)
+            )
+        )
     ):
         ext = _require_hip()
         if ext is not None:
