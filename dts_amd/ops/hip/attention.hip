@@ -75,10 +75,8 @@ attn_decode_kernel(float* __restrict__ ws_m,  // [B, Hkv, S, G]
     return;
   }
 
-  // LDS: cross-wave combine buffers only (scores stay in registers and
-  // cross the 4-lane key groups via shuffles — no per-page LDS handoff,
-  // no barrier, so the compiler may overlap the next page's K/V loads
-  // with the current page's softmax chain)
+  // LDS: per-wave score buffer + combine buffers
+  __shared__ float s_scores[NWAVE][G][BS];
   __shared__ float s_m[NWAVE][G], s_l[NWAVE][G];
   __shared__ float s_o[NWAVE][G][D];
 
@@ -133,13 +131,17 @@ attn_decode_kernel(float* __restrict__ ws_m,  // [B, Hkv, S, G]
         }
       }
     }
-    // reduce over the 4 lanes of the key group; lane 4j holds score j
+    // reduce over the 4 lanes of the key group
 #pragma unroll
     for (int g = 0; g < G; ++g) {
       partial[g] += __shfl_xor(partial[g], 1, 64);
       partial[g] += __shfl_xor(partial[g], 2, 64);
-      partial[g] = (key_of_lane < valid) ? partial[g] * scale : -1e30f;
+      if (c == 0)
+        s_scores[wave][g][key_of_lane] =
+            (key_of_lane < valid) ? partial[g] * scale : -1e30f;
     }
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_wave_barrier();
 
     // ---- batch the page's V words up front: 16 independent loads in
     // flight at once (the per-(g,j) load placement serialized an L2/HBM
@@ -157,16 +159,12 @@ attn_decode_kernel(float* __restrict__ ws_m,  // [B, Hkv, S, G]
       }
     }
 
-    // ---- softmax update + V accumulate; lane owns output dims
-    // [lane*DPV, lane*DPV+DPV); page scores broadcast from lanes 4j
+    // ---- softmax update + V accumulate; lane owns output dims {2l, 2l+1}
 #pragma unroll
     for (int g = 0; g < G; ++g) {
-      float sj[BS];
-#pragma unroll
-      for (int j = 0; j < BS; ++j) sj[j] = __shfl(partial[g], j * LPK, 64);
       float tmax = -1e30f;
 #pragma unroll
-      for (int j = 0; j < BS; ++j) tmax = fmaxf(tmax, sj[j]);
+      for (int j = 0; j < BS; ++j) tmax = fmaxf(tmax, s_scores[wave][g][j]);
       float m_new = fmaxf(m[g], tmax);
       if (m_new > -1e30f) {
         float alpha = __expf(m[g] - m_new);
@@ -174,21 +172,23 @@ attn_decode_kernel(float* __restrict__ ws_m,  // [B, Hkv, S, G]
         for (int d = 0; d < DPV; ++d) o[g][d] *= alpha;
         lsum[g] *= alpha;
         m[g] = m_new;
+        float p[BS];
         float psum = 0.f;
 #pragma unroll
         for (int j = 0; j < BS; ++j) {
-          sj[j] = (sj[j] > -1e30f) ? __expf(sj[j] - m_new) : 0.f;
-          psum += sj[j];
+          p[j] = (j < valid) ? __expf(s_scores[wave][g][j] - m_new) : 0.f;
+          psum += p[j];
         }
         lsum[g] += psum;
 #pragma unroll
         for (int j = 0; j < BS; ++j) {
-          o[g][0] += sj[j] * bf2f((short)(vw[j] & 0xffff));
+          o[g][0] += p[j] * bf2f((short)(vw[j] & 0xffff));
           if constexpr (DPV == 2)
-            o[g][1] += sj[j] * bf2f((short)((vw[j] >> 16) & 0xffff));
+            o[g][1] += p[j] * bf2f((short)((vw[j] >> 16) & 0xffff));
         }
       }
     }
+    __builtin_amdgcn_wave_barrier();
   }
 
   // ---- cross-wave combine via LDS, then publish split partials
