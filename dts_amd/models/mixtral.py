@@ -66,12 +66,13 @@ class MoEMLP(nn.Module):
             return self._forward_ep(x, flat_tok, flat_expert, flat_w)
 
         out = torch.zeros(T, self.hidden, dtype=torch.float32, device=x.device)
-        # grouped execution: one GEMM per locally-resident expert slice
+        # grouped execution: one GEMM per locally-resident expert slice.
+        # No host-side `sel.any()` early-outs: a bool() on a device tensor
+        # is a stream sync per expert per layer AND breaks hipGraph
+        # capture; zero-row GEMMs are free.
         for e_local in range(self.experts_local):
             e = self.expert_offset + e_local
             sel = flat_expert == e
-            if not bool(sel.any()):
-                continue
             toks = flat_tok[sel]
             xe = x[toks]
             gu = F.linear(xe, self.gate_up_w[e_local])
@@ -108,8 +109,6 @@ class MoEMLP(nn.Module):
         for e_local in range(self.experts_local):
             e = self.expert_offset + e_local
             sel = recv_e == e
-            if not bool(sel.any()):
-                continue
             gu = F.linear(recv_x[sel], self.gate_up_w[e_local])
             ye[sel] = F.linear(ops.silu_mul(gu), self.down_w[e_local]).float()
 
