@@ -147,6 +147,8 @@ def attn_decode_paged(q, k_cache, v_cache, block_tables, kv_lens, scale=None):
 
 
 def silu_mul(gate_up):
+    if gate_up.shape[0] == 0:  # zero-row expert slices (MoE grouped GEMMs)
+        return gate_up[:, : gate_up.shape[1] // 2]
     if _on_gpu(gate_up):
         ext = _require_hip()
         if ext is not None:
