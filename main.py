@@ -18,7 +18,15 @@ from dts_amd.search.researcher import DeepResearcher
 from dts_amd.serving import LocalBackend, ServingEngine
 
 
-async def run_dts_example() -> None:
+async def run_dts_example(
+    *,
+    init_branches: int = 3,
+    turns_per_branch: int = 2,
+    user_intents_per_branch: int = 2,
+    rounds: int = 2,
+    deep_research: bool = True,
+    output_path: str = "dts_output.json",
+) -> float:
     use_gpu = torch.cuda.is_available()
     model = "llama-3-8b" if use_gpu else "llama-tiny"
     engine_kwargs = {} if use_gpu else {
@@ -33,21 +41,22 @@ async def run_dts_example() -> None:
     config = DTSConfig(
         goal="Convince a skeptical team lead to adopt automated testing",
         first_message="Our team doesn't write tests. Is it really worth the time?",
-        init_branches=3,
-        turns_per_branch=2,
-        user_intents_per_branch=2,
+        init_branches=init_branches,
+        turns_per_branch=turns_per_branch,
+        user_intents_per_branch=user_intents_per_branch,
         user_variability=True,
         scoring_mode="comparative",
         prune_threshold=5.0,
-        deep_research=True,
+        deep_research=deep_research,
         seed=0,
     )
     researcher = DeepResearcher(llm, cache_dir=config.research_cache_dir)
     engine = DTSEngine(llm, config, researcher=researcher)
-    result = await engine.run(rounds=2)
-    result.save_json("dts_output.json")
-    print(f"Best score: {result.best_score:.1f} — saved to dts_output.json")
+    result = await engine.run(rounds=rounds)
+    result.save_json(output_path)
+    print(f"Best score: {result.best_score:.1f} — saved to {output_path}")
     backend.shutdown()
+    return result.best_score
 
 
 if __name__ == "__main__":
