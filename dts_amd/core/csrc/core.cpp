@@ -161,6 +161,11 @@ class CoreScheduler {
       if (chunk <= 0) continue;
       budget -= chunk;
       s.sched_chunk = chunk;
+      // mark immediately: a seq already placed in this step's batch must
+      // not become a preemption victim for a later seq in this loop
+      // (running order != arrival order after re-admission), or
+      // build_batch would index its released block table
+      s.in_flight = true;
       if (chunk == 1)
         decodes.push_back(&s);
       else

@@ -185,6 +185,11 @@ class Scheduler:
             if chunk <= 0:
                 continue
             budget -= chunk
+            # mark immediately: a seq already placed in this step's batch
+            # must not be chosen as a preemption victim by a LATER seq in
+            # this same loop (running order != arrival order after a
+            # re-admission), or _build_batch would index released blocks
+            seq.in_flight = True
             if remaining == 1 and seq.output_tokens:
                 decodes.append(seq)
             elif chunk == 1:

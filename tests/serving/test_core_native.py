@@ -130,3 +130,60 @@ def test_engine_runs_on_native_core(monkeypatch):
     eng.run_until_idle()
     res = fut.result(timeout=10)
     assert res.completion_tokens >= 1
+
+
+@pytest.mark.parametrize("seed", range(6))
+def test_lockstep_under_memory_pressure(seed):
+    """Small pool → preemption, eviction and prefix reuse every few steps.
+
+    Same lockstep protocol as above but sized so the schedulers must
+    constantly preempt (preempt-youngest) and re-admit; any divergence in
+    victim choice, block recycling or cache accounting shows up as a
+    batch mismatch.
+    """
+    rng = random.Random(1000 + seed)
+    py = Scheduler(BlockManager(28, 4), 8)
+    nat = NativeScheduler(28, 4, 8)
+    live = []
+
+    def new_seq(tokens):
+        sp = Sequence(tokens=list(tokens), params=SamplingParams())
+        sn = Sequence(tokens=list(tokens), params=SamplingParams())
+        sn.seq_id = sp.seq_id
+        py.add(sp)
+        nat.add(sn)
+        live.append((sp, sn))
+
+    prefixes = [[7, 7, 7, 7, 7, 7, 7, 7], [9, 9, 9, 9]]
+    for step in range(150):
+        if rng.random() < 0.3 and len(live) < 7:
+            base = rng.choice(prefixes) if rng.random() < 0.5 else []
+            n = rng.randrange(2, 24)
+            new_seq(base + [rng.randrange(50) for _ in range(n)])
+        ba = py.schedule()
+        bb = nat.schedule()
+        assert batches_equal(ba, bb), f"diverged at step {step} (seed {seed})"
+        if ba is None:
+            # both idle-or-stuck; drain one finished/stuck seq if any
+            if py.stuck:
+                assert [s.seq_id for s in py.stuck] == [s.seq_id for s in nat.stuck]
+                sp, sn = next(p for p in live if p[0] in py.stuck)
+                py.abort(sp)
+                nat.abort(sn)
+                live.remove((sp, sn))
+            continue
+        py.advance_computed(ba)
+        nat.advance_computed(bb)
+        assert [s.seq_id for s in ba._sampled_seqs] == [
+            s.seq_id for s in bb._sampled_seqs
+        ]
+        for sp, sn in zip(ba._sampled_seqs, bb._sampled_seqs):
+            tok = rng.randrange(100, 200)
+            py.append_token(sp, tok)
+            nat.append_token(sn, tok)
+            if len(sp.output_tokens) > rng.randrange(3, 16):
+                py.finish(sp, "stop")
+                nat.finish(sn, "stop")
+                live.remove((sp, sn))
+    assert py.bm.cache_hit_tokens == nat.cache_hit_tokens
+    assert py.bm.cache_miss_tokens == nat.cache_miss_tokens
