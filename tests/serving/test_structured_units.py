@@ -97,3 +97,62 @@ class TestForms:
         ranked = [r["trajectory_id"] for r in obj["ranking"]]
         assert sorted(ranked) == sorted(ids)
         assert [r["rank"] for r in obj["ranking"]] == [1, 2, 3, 4]
+
+
+class TestRandomDrivesProduceValidJson:
+    """Property: ANY token stream the guide permits decodes to JSON that
+    parses and matches the phase schema — the guarantee the search layer
+    relies on when sampling from random-init weights."""
+
+    @pytest.mark.parametrize("seed", range(5))
+    def test_strategy_form(self, tok, seed):
+        import random
+
+        rng = random.Random(seed)
+        text = drive(tok, strategy_form(tok, 4), pick=lambda a, s: rng.choice(a))
+        d = json.loads(text)
+        assert set(d) == {"goal", "nodes", "coverage_rationale"}
+        assert len(d["nodes"]) == 4
+        for i, key in enumerate(d["nodes"], start=1):
+            assert key.startswith(f"Strategy {i}: ")
+
+    @pytest.mark.parametrize("seed", range(5))
+    def test_intent_form(self, tok, seed):
+        import random
+
+        rng = random.Random(seed)
+        text = drive(tok, intent_form(tok, 3), pick=lambda a, s: rng.choice(a))
+        d = json.loads(text)
+        assert len(d["intents"]) == 3
+        for n, i in enumerate(d["intents"], start=1):
+            assert i["id"] == f"intent_{n}"
+            assert set(i) == {
+                "id", "label", "description", "emotional_tone", "cognitive_stance",
+            }
+
+    @pytest.mark.parametrize("seed", range(5))
+    def test_absolute_judge_form(self, tok, seed):
+        import random
+
+        from dts_amd.serving.structured import absolute_judge_form
+
+        rng = random.Random(seed)
+        text = drive(tok, absolute_judge_form(tok), pick=lambda a, s: rng.choice(a))
+        d = json.loads(text)
+        assert len(d["criteria"]) == 10
+        for c in d["criteria"].values():
+            assert 0.0 <= float(c["score"]) <= 1.0
+        assert "total_score" in d
+
+    @pytest.mark.parametrize("seed", range(8))
+    def test_ranking_guide_permutation(self, tok, seed):
+        import random
+
+        ids = [f"00000000-0000-0000-0000-{i:012d}" for i in range(5)]
+        rng = random.Random(seed)
+        text = drive(
+            tok, comparative_judge_form(tok, ids), pick=lambda a, s: rng.choice(a)
+        )
+        d = json.loads(text)
+        ranked = [r["trajectory_id"] for r in d["ranking"]]
+        assert sorted(ranked) == sorted(ids)  # a true permutation, no repeats
