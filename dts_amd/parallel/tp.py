@@ -18,7 +18,6 @@ from typing import Optional
 import torch
 import torch.distributed as dist
 import torch.nn as nn
-import torch.nn.functional as F
 
 
 class TPContext:
