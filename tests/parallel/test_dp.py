@@ -136,3 +136,33 @@ def test_dp_spmd_forking_comparative():
     # root + 4 strategy branches + 4*2 forked children
     assert r0["n_nodes"] == 1 + 4 + 8
     assert r0["best"] == r1["best"]
+
+
+@pytest.mark.timeout(240)
+def test_dp_spmd_four_ranks_forking():
+    """world_size=4 with forking: branches (4) < some rank counts and
+    forked children (8) spread unevenly — exercises the owner-mapping
+    and payload-merge math the round-end N=4/8 scaling bench relies on."""
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = 29600 + (os.getpid() * 4 + 3) % 800
+    procs = [
+        ctx.Process(target=_worker, args=(r, 4, port, q, True)) for r in range(4)
+    ]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(4):
+        rank, data = q.get(timeout=200)
+        results[rank] = data
+    for p in procs:
+        p.join(timeout=30)
+        assert p.exitcode == 0
+    fps = [results[r]["fingerprint"] for r in range(4)]
+    assert all(f == fps[0] for f in fps[1:])
+    assert len({results[r]["best"] for r in range(4)}) == 1
+    assert results[0]["n_nodes"] == 1 + 4 + 8
+    # every rank did strictly less than the whole job
+    single_rank_calls = 1 + 4 + 4 * 2 * (1 + 2) + 4 * 2  # rough upper shape
+    for r in range(4):
+        assert results[r]["n_llm_calls"] < single_rank_calls
