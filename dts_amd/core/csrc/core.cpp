@@ -72,6 +72,7 @@ class CoreScheduler {
   // ---- stats ------------------------------------------------------------
   int64_t cache_hit_tokens = 0;
   int64_t cache_miss_tokens = 0;
+  int64_t preemptions = 0;
 
   int num_free() const { return (int)(free_ids_.size() + evictable_.size()); }
 
@@ -364,6 +365,7 @@ class CoreScheduler {
     victim->status = Status::WAITING;
     erase_running(victim->id);
     waiting_.push_front(victim->id);
+    preemptions++;
     return true;
   }
 
@@ -511,5 +513,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
       .def("take_stuck", &CoreScheduler::take_stuck)
       .def("num_free", &CoreScheduler::num_free)
       .def_readonly("cache_hit_tokens", &CoreScheduler::cache_hit_tokens)
-      .def_readonly("cache_miss_tokens", &CoreScheduler::cache_miss_tokens);
+      .def_readonly("cache_miss_tokens", &CoreScheduler::cache_miss_tokens)
+      .def_readonly("preemptions", &CoreScheduler::preemptions);
 }

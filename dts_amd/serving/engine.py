@@ -342,6 +342,7 @@ class ServingEngine:
             "tokens_sampled": self.tokens_sampled,
             "tokens_prefilled": self.tokens_prefilled,
             "native_scheduler": type(self.scheduler).__name__ == "NativeScheduler",
+            "preemptions": getattr(self.scheduler, "preemptions", 0),
             "eager_decode_steps": self.eager_decode_steps,
             "prefill_steps": self.prefill_steps,
             "t_forward_graph_s": round(self.t_forward_graph, 2),

@@ -41,6 +41,10 @@ class NativeScheduler:
         return self.core.cache_hit_tokens
 
     @property
+    def preemptions(self) -> int:
+        return self.core.preemptions
+
+    @property
     def cache_miss_tokens(self) -> int:
         return self.core.cache_miss_tokens
 

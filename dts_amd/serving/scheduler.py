@@ -41,6 +41,7 @@ class Scheduler:
         self.waiting: deque = deque()
         self.running: list = []
         self.stuck: list = []  # seqs that can NEVER fit (engine fails them)
+        self.preemptions = 0
         self._arrival = 0
 
     # ------------------------------------------------------------------
@@ -119,6 +120,7 @@ class Scheduler:
         victim.status = SeqStatus.WAITING
         self.running.remove(victim)
         self.waiting.appendleft(victim)
+        self.preemptions += 1
         return True
 
     def _release_blocks(self, seq: Sequence) -> None:
