@@ -12,20 +12,31 @@ TARGET = CORE_DIR / "_dts_core.so"
 
 
 def build(verbose: bool = True) -> Path:
+    """Set DTS_CORE_ASAN=1 to produce an AddressSanitizer build (run the
+    test suite with LD_PRELOAD=libasan.so ASAN_OPTIONS=detect_leaks=0);
+    profiles/asan_core.md records the clean pass."""
     os.environ.setdefault("MAX_JOBS", str(os.cpu_count() or 4))
-    BUILD_DIR.mkdir(parents=True, exist_ok=True)
+    asan = os.environ.get("DTS_CORE_ASAN") == "1"
+    build_dir = BUILD_DIR.parent / "build_asan" if asan else BUILD_DIR
+    build_dir.mkdir(parents=True, exist_ok=True)
+    cflags = ["-O3", "-std=c++17"]
+    ldflags = []
+    if asan:
+        cflags += ["-fsanitize=address", "-fno-omit-frame-pointer", "-g"]
+        ldflags += ["-fsanitize=address"]
     from torch.utils.cpp_extension import load
 
     load(
         name="_dts_core",
         sources=[str(CORE_DIR / "csrc" / "core.cpp")],
-        build_directory=str(BUILD_DIR),
-        extra_cflags=["-O3", "-std=c++17"],
+        build_directory=str(build_dir),
+        extra_cflags=cflags,
+        extra_ldflags=ldflags,
         verbose=verbose,
         is_python_module=False,
         with_cuda=False,
     )
-    built = BUILD_DIR / "_dts_core.so"
+    built = build_dir / "_dts_core.so"
     if not built.exists():
         raise RuntimeError(f"build produced no {built}")
     shutil.copy2(built, TARGET)
