@@ -28,7 +28,9 @@ from dts_amd.search.types import (
     DialogueNode,
     DTSRunResult,
     NodeStatus,
+    Strategy,
     TokenTracker,
+    UserIntent,
 )
 from dts_amd.utils.logging import log_phase, logger
 
@@ -106,7 +108,12 @@ class DTSEngine:
         return self._token_tracker
 
     # ------------------------------------------------------------------
-    async def run(self, rounds: int = 1) -> DTSRunResult:
+    async def run(self, rounds: int = 1, resume_from=None) -> DTSRunResult:
+        """resume_from: a saved exploration dict (or a path to its JSON) —
+        rebuilds the tree from the checkpoint instead of generating
+        strategies, then continues `rounds` more rounds. The reference has
+        no mid-search resume (SURVEY.md §5 Checkpoint/resume); this uses
+        the same tree-state JSON it persists."""
         cfg = self.config
         log_phase("INIT", f"Goal: {cfg.goal[:60]}")
         log_phase(
@@ -131,7 +138,10 @@ class DTSEngine:
         )
 
         self._emit("phase", {"phase": "initializing", "message": "Creating tree structure..."})
-        tree = await self._initialize_tree()
+        if resume_from is not None:
+            tree = self._load_tree(resume_from)
+        else:
+            tree = await self._initialize_tree()
         self._tree = tree
 
         total_pruned = 0
@@ -344,6 +354,71 @@ class DTSEngine:
             )
             tree.add_child(root.id, child)
             self._emit("node_added", self._node_event(child))
+        return tree
+
+    def _load_tree(self, checkpoint) -> DialogueTree:
+        """Rebuild a DialogueTree from a saved exploration dict
+        (to_exploration_dict output, possibly loaded from disk)."""
+        import json as _json
+
+        cfg = self.config
+        if isinstance(checkpoint, str):
+            with open(checkpoint) as f:
+                checkpoint = _json.load(f)
+        root = DialogueNode(
+            id=generate_node_id(),
+            depth=0,
+            messages=[Message.user(cfg.first_message)],
+        )
+        tree = DialogueTree.create(root)
+        self._emit("node_added", self._node_event(root))
+        self._research_report = checkpoint.get("research_report")
+        if self._research_report:
+            self._evaluator.set_research_context(self._research_report)
+        branches = sorted(
+            checkpoint.get("branches", []),
+            key=lambda b: (b.get("depth", 1), b["id"]),
+        )
+        for b in branches:
+            strat = b.get("strategy") or {}
+            ui = b.get("user_intent")
+            node = DialogueNode(
+                id=b["id"],
+                strategy=Strategy(
+                    tagline=strat.get("tagline", ""),
+                    description=strat.get("description", ""),
+                ),
+                user_intent=(
+                    UserIntent(
+                        id=ui.get("id", "intent_1"),
+                        label=ui.get("label", ""),
+                        description=ui.get("description", ""),
+                        emotional_tone=ui.get("emotional_tone", ""),
+                        cognitive_stance=ui.get("cognitive_stance", ""),
+                    )
+                    if ui
+                    else None
+                ),
+                messages=[
+                    Message(role=m["role"], content=m["content"])
+                    for m in b.get("trajectory", [])
+                ],
+                status=NodeStatus(b.get("status", "active")),
+                prune_reason=b.get("prune_reason"),
+            )
+            sc = b.get("scores") or {}
+            node.stats.judge_scores = list(sc.get("individual") or [])
+            node.stats.aggregated_score = sc.get("aggregated", 0.0) or 0.0
+            node.stats.visits = sc.get("visits", 0) or 0
+            node.stats.value_mean = sc.get("value_mean", 0.0) or 0.0
+            node.stats.value_sum = node.stats.value_mean * node.stats.visits
+            node.stats.critiques = dict(sc.get("critiques") or {})
+            parent = b.get("parent_id")
+            if not parent or parent not in tree.nodes:
+                parent = root.id  # pre-extension checkpoints: flat under root
+            tree.add_child(parent, node)
+            self._emit("node_added", self._node_event(node))
+        log_phase("INIT", f"Resumed {len(branches)} branches from checkpoint")
         return tree
 
     def _prune(self, nodes: list, scores: dict) -> list:

@@ -254,7 +254,9 @@ class DTSRunResult:
         """The tree-state JSON checkpoint — schema kept byte-compatible with
         ref types.py:457-554 (summary / research_report / best_branch /
         branches[...] with strategy, user_intent, status, depth, scores,
-        trajectory, prune_reason)."""
+        trajectory, prune_reason). Additive extensions (ignored by the
+        reference frontend, required for DTSEngine.run(resume_from=...)):
+        branch-level parent_id and user_intent id/description."""
         branches = []
         for node in self.all_nodes:
             if node.strategy is None:
@@ -262,13 +264,16 @@ class DTSRunResult:
             branches.append(
                 {
                     "id": node.id,
+                    "parent_id": node.parent_id,
                     "strategy": {
                         "tagline": node.strategy.tagline,
                         "description": node.strategy.description,
                     },
                     "user_intent": (
                         {
+                            "id": node.user_intent.id,
                             "label": node.user_intent.label,
+                            "description": node.user_intent.description,
                             "emotional_tone": node.user_intent.emotional_tone,
                             "cognitive_stance": node.user_intent.cognitive_stance,
                         }

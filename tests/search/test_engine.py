@@ -91,6 +91,7 @@ class TestFullRun:
         b = d["branches"][0]
         assert set(b) == {
             "id",
+            "parent_id",  # additive extension: enables run(resume_from=...)
             "strategy",
             "user_intent",
             "status",

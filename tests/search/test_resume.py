@@ -1,0 +1,106 @@
+"""Checkpoint → resume tests.
+
+The reference has no mid-search resume (SURVEY.md §5 Checkpoint/resume:
+the exploration-dict JSON is persisted but only the frontend re-loads
+it). Here DTSEngine.run(resume_from=...) rebuilds the tree from that same
+JSON and continues searching.
+"""
+
+import asyncio
+import json
+
+import pytest
+
+from dts_amd.llm import LLM, FakeBackend
+from dts_amd.search import DTSConfig, DTSEngine
+from dts_amd.search.types import NodeStatus
+
+
+def make_engine(**over):
+    kw = dict(
+        goal="g",
+        first_message="hello can you help me choose a database?",
+        init_branches=3,
+        turns_per_branch=1,
+        user_intents_per_branch=1,
+        scoring_mode="absolute",
+        prune_threshold=0.0,
+        seed=7,
+    )
+    kw.update(over)
+    cfg = DTSConfig(**kw)
+    llm = LLM(FakeBackend(), default_model="fake")
+    return DTSEngine(llm, cfg), cfg
+
+
+class TestResume:
+    def test_round_trip_preserves_tree(self, tmp_path):
+        engine, _ = make_engine()
+        result = asyncio.run(engine.run(rounds=1))
+        path = tmp_path / "ckpt.json"
+        result.save_json(str(path))
+
+        # resume into a FRESH engine; 0 further rounds of work is not
+        # allowed by the API (rounds>=1), so run 1 round and compare the
+        # loaded state before expansion via the engine's tree
+        engine2, _ = make_engine()
+        d = json.loads(path.read_text())
+        tree = engine2._load_tree(d)
+        by_id = {n.id: n for n in tree.all_nodes()}
+        for b in d["branches"]:
+            n = by_id[b["id"]]
+            assert n.status.value == b["status"]
+            assert [m.content for m in n.messages] == [
+                m["content"] for m in b["trajectory"]
+            ]
+            assert n.stats.aggregated_score == b["scores"]["aggregated"]
+            assert n.stats.judge_scores == b["scores"]["individual"]
+            assert n.strategy.tagline == b["strategy"]["tagline"]
+
+    def test_resume_continues_search(self, tmp_path):
+        engine, _ = make_engine()
+        r1 = asyncio.run(engine.run(rounds=1))
+        path = tmp_path / "ckpt.json"
+        r1.save_json(str(path))
+        msgs_before = {
+            n.id: len(n.messages) for n in r1.all_nodes if n.strategy is not None
+        }
+
+        engine2, _ = make_engine()
+        r2 = asyncio.run(engine2.run(rounds=1, resume_from=str(path)))
+        assert r2.best_node_id is not None
+        # active branches got expanded further (linear rounds extend in place)
+        extended = [
+            n
+            for n in r2.all_nodes
+            if n.strategy is not None
+            and n.status == NodeStatus.ACTIVE
+            and len(n.messages) > msgs_before.get(n.id, 0)
+        ]
+        assert extended, "resumed round must extend active branches"
+        # and they were re-scored in the resumed round
+        for n in extended:
+            assert n.stats.visits >= 1
+
+    def test_pruned_branches_stay_pruned(self, tmp_path):
+        engine, _ = make_engine(prune_threshold=11.0)  # prune everything
+        r1 = asyncio.run(engine.run(rounds=1))
+        path = tmp_path / "ckpt.json"
+        r1.save_json(str(path))
+        pruned_ids = {
+            n.id for n in r1.all_nodes if n.status == NodeStatus.PRUNED
+        }
+        assert pruned_ids
+
+        engine2, _ = make_engine(prune_threshold=11.0)
+        r2 = asyncio.run(engine2.run(rounds=1, resume_from=str(path)))
+        for n in r2.all_nodes:
+            if n.id in pruned_ids:
+                assert n.status == NodeStatus.PRUNED
+                assert len(n.messages) == len(
+                    next(
+                        m
+                        for m in r1.all_nodes
+                        if m.id == n.id
+                    ).messages
+                ), "pruned branches must not be expanded on resume"
