@@ -33,6 +33,10 @@ class SearchRequest(BaseModel):
     # additive fields (closed schema gap)
     user_variability: bool = Field(default=False)
     reasoning_enabled: bool = Field(default=False)
+    # additive: a saved exploration dict to resume from (run continues
+    # `rounds` more rounds on the restored tree instead of generating
+    # fresh strategies)
+    resume_from: Optional[dict] = Field(default=None)
 
 
 class EventMessage(BaseModel):

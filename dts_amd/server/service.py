@@ -62,7 +62,9 @@ async def run_dts_session(
         await queue.put({"type": event_type, "data": data})
 
     engine.set_event_callback(on_event)
-    task = asyncio.create_task(engine.run(rounds=request.rounds))
+    task = asyncio.create_task(
+        engine.run(rounds=request.rounds, resume_from=request.resume_from)
+    )
 
     try:
         while True:

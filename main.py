@@ -26,6 +26,7 @@ async def run_dts_example(
     rounds: int = 2,
     deep_research: bool = True,
     output_path: str = "dts_output.json",
+    resume_from: str | None = None,
 ) -> float:
     use_gpu = torch.cuda.is_available()
     model = "llama-3-8b" if use_gpu else "llama-tiny"
@@ -52,7 +53,7 @@ async def run_dts_example(
     )
     researcher = DeepResearcher(llm, cache_dir=config.research_cache_dir)
     engine = DTSEngine(llm, config, researcher=researcher)
-    result = await engine.run(rounds=rounds)
+    result = await engine.run(rounds=rounds, resume_from=resume_from)
     result.save_json(output_path)
     print(f"Best score: {result.best_score:.1f} — saved to {output_path}")
     backend.shutdown()
@@ -60,6 +61,12 @@ async def run_dts_example(
 
 
 if __name__ == "__main__":
+    import argparse
+
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--rounds", type=int, default=2)
+    ap.add_argument("--resume", default=None, help="exploration JSON to resume")
+    a = ap.parse_args()
     start = time.time()
-    asyncio.run(run_dts_example())
+    asyncio.run(run_dts_example(rounds=a.rounds, resume_from=a.resume))
     print(f"Completed in {time.time() - start:.1f}s")
