@@ -365,8 +365,13 @@ class DTSEngine:
         if isinstance(checkpoint, str):
             with open(checkpoint) as f:
                 checkpoint = _json.load(f)
+        # deterministic root id: every DP rank loads the same checkpoint
+        # independently and their trees must match node-for-node
+        import uuid as _uuid
+
+        root_seed = f"{cfg.goal}::{cfg.first_message}::{len(checkpoint.get('branches', []))}"
         root = DialogueNode(
-            id=generate_node_id(),
+            id=str(_uuid.uuid5(_uuid.NAMESPACE_URL, root_seed)),
             depth=0,
             messages=[Message.user(cfg.first_message)],
         )

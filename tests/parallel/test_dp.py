@@ -166,3 +166,75 @@ def test_dp_spmd_four_ranks_forking():
     single_rank_calls = 1 + 4 + 4 * 2 * (1 + 2) + 4 * 2  # rough upper shape
     for r in range(4):
         assert results[r]["n_llm_calls"] < single_rank_calls
+
+
+def _resume_worker(rank, world, port, out_q):
+    os.environ.update(
+        {
+            "RANK": str(rank),
+            "WORLD_SIZE": str(world),
+            "LOCAL_RANK": str(rank),
+            "MASTER_ADDR": "127.0.0.1",
+            "MASTER_PORT": str(port),
+        }
+    )
+    import asyncio
+
+    import torch.distributed as dist
+
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from dts_amd.llm import LLM, FakeBackend
+        from dts_amd.parallel.dp import DPContext
+        from dts_amd.search import DTSConfig
+        from dts_amd.search.dist_engine import DistributedDTSEngine
+
+        def mk():
+            llm = LLM(FakeBackend(score_salt=f"r{rank}"), default_model="fake")
+            cfg = DTSConfig(
+                goal="g",
+                first_message="hello can you help me with this?",
+                init_branches=4,
+                turns_per_branch=1,
+                user_intents_per_branch=1,
+                scoring_mode="absolute",
+                prune_threshold=0.0,
+                seed=3,
+            )
+            return DistributedDTSEngine(llm, cfg, dp=DPContext())
+
+        r1 = asyncio.run(mk().run(rounds=1))
+        ckpt = r1.to_exploration_dict()
+        r2 = asyncio.run(mk().run(rounds=1, resume_from=ckpt))
+        fp = sorted(
+            (n.id, n.status.value, len(n.messages),
+             round(n.stats.aggregated_score, 4))
+            for n in r2.all_nodes
+        )
+        out_q.put((rank, {"fingerprint": fp, "best": r2.best_node_id}))
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(240)
+def test_dp_resume_identical_across_ranks():
+    """Every rank loads the same checkpoint independently; the resumed
+    trees (including the synthesized root) must match node-for-node so
+    DP sharding stays consistent."""
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = 29600 + (os.getpid() * 4 + 1) % 800
+    procs = [
+        ctx.Process(target=_resume_worker, args=(r, 2, port, q)) for r in range(2)
+    ]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(2):
+        rank, data = q.get(timeout=200)
+        results[rank] = data
+    for p in procs:
+        p.join(timeout=30)
+        assert p.exitcode == 0
+    assert results[0]["fingerprint"] == results[1]["fingerprint"]
+    assert results[0]["best"] == results[1]["best"]
