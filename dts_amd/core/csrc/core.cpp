@@ -140,9 +140,16 @@ class CoreScheduler {
     admit();
     std::vector<Seq*> prefills, decodes;
     int64_t budget = max_batch_tokens_;
-    for (int64_t sid : running_) {
-      Seq& s = seqs_.at(sid);
-      if (s.in_flight) continue;
+    // iterate a snapshot: preempt_youngest erases victims from running_
+    // mid-loop (mutating a range-for'd container is UB), and a victim
+    // visited later must be skipped — growing its released block table
+    // would leak blocks permanently (refcount 1, no owner)
+    const std::vector<int64_t> snapshot(running_.begin(), running_.end());
+    for (int64_t sid : snapshot) {
+      auto sit = seqs_.find(sid);
+      if (sit == seqs_.end()) continue;
+      Seq& s = sit->second;
+      if (s.status != Status::RUNNING || s.in_flight) continue;
       int remaining = (int)s.tokens.size() - s.num_computed;
       if (remaining <= 0) continue;
       if (!ensure_blocks(s, (int)s.tokens.size())) {

@@ -289,9 +289,19 @@ class ServingEngine:
     # ------------------------------------------------------------------
     def run_until_idle(self) -> None:
         """Synchronous drain (tests / bench warm-up)."""
+        fruitless = 0
         while self.scheduler.has_work():
-            if not self.step():
-                break
+            if self.step():
+                fruitless = 0
+            else:
+                # a no-batch step is not idle: it may have failed stuck
+                # requests or preempted; re-schedule while work remains
+                fruitless += 1
+                if fruitless > 10000:
+                    raise RuntimeError(
+                        "engine live-lock: requests pending but the "
+                        "scheduler produced no batch for 10000 steps"
+                    )
 
     def _loop(self) -> None:
         while True:

@@ -165,7 +165,14 @@ class Scheduler:
         prefills: list = []  # (seq, chunk)
         decodes: list = []
         for seq in list(self.running):
-            if seq.in_flight:
+            # snapshot iteration: a preemption triggered by an EARLIER seq
+            # in this loop removes its victim from self.running — the stale
+            # snapshot still visits it. Growing a WAITING victim's (already
+            # released) block table here leaked blocks permanently: the
+            # victim's next _admit rebinds block_table and the orphaned
+            # blocks keep refcount 1 forever, shrinking the pool until
+            # healthy requests are falsely failed as stuck.
+            if seq.status != SeqStatus.RUNNING or seq.in_flight:
                 continue
             remaining = len(seq.tokens) - seq.num_computed
             if remaining <= 0:

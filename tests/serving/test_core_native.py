@@ -187,3 +187,18 @@ def test_lockstep_under_memory_pressure(seed):
                 live.remove((sp, sn))
     assert py.bm.cache_hit_tokens == nat.cache_hit_tokens
     assert py.bm.cache_miss_tokens == nat.cache_miss_tokens
+    # no-leak invariant (identical-bug blind spot: both sides once leaked
+    # blocks the same way, so batch comparison alone missed it): after
+    # draining every live sequence the whole pool must be free/evictable
+    for sp, sn in list(live):
+        py.finish(sp, "drain")
+        nat.finish(sn, "drain")
+    for w in list(py.waiting):
+        py.abort(w)
+    while True:
+        ids = [s.seq_id for s in nat._seqs.values()]
+        if not ids:
+            break
+        nat.abort(nat._seqs[ids[0]])
+    assert py.bm.num_free() == py.bm.num_blocks, "python scheduler leaked blocks"
+    assert nat.num_free() == py.bm.num_free(), "native scheduler leaked blocks"
