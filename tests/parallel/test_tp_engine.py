@@ -40,11 +40,13 @@ def _worker(rank, world, port, tmpdir, out_q):
 
         prompt = list(range(1, 40))
         if rank == 0:
+            from dts_amd.serving.structured import strategy_form
+
             engine = ServingEngine(
                 model_name="llama-tiny",
                 device="cpu",
                 dtype=torch.float32,
-                num_blocks=128,
+                num_blocks=256,
                 block_size=8,
                 model=model,
             )
@@ -52,10 +54,16 @@ def _worker(rank, world, port, tmpdir, out_q):
             fut = engine.submit_tokens(
                 prompt, SamplingParams(max_tokens=8, temperature=0.0, seed=0)
             )
+            gfut = engine.submit_tokens(
+                list(range(40, 80)),
+                SamplingParams(max_tokens=4096, temperature=0.0, seed=0),
+                guide=strategy_form(engine.tokenizer, 2),
+            )
             engine.run_until_idle()
             res = fut.result(timeout=30)
+            gtext = gfut.result(timeout=30).text
             TPDriverMixin.shutdown()
-            out_q.put(("tp", res.token_ids))
+            out_q.put(("tp", res.token_ids, gtext))
         else:
             pool = KVCachePool(
                 spec.num_layers,
@@ -86,19 +94,27 @@ def test_tp_engine_matches_dense(tmp_path):
     dense = LlamaModel(spec, dtype=torch.float32, device="cpu")
     dense.random_init(seed=21)
     save_llama_safetensors(dense, str(tmp_path))
+    from dts_amd.serving.structured import strategy_form
+
     engine = ServingEngine(
         model_name="llama-tiny",
         device="cpu",
         dtype=torch.float32,
-        num_blocks=128,
+        num_blocks=256,
         block_size=8,
         model=dense,
     )
     fut = engine.submit_tokens(
         list(range(1, 40)), SamplingParams(max_tokens=8, temperature=0.0, seed=0)
     )
+    gfut = engine.submit_tokens(
+        list(range(40, 80)),
+        SamplingParams(max_tokens=4096, temperature=0.0, seed=0),
+        guide=strategy_form(engine.tokenizer, 2),
+    )
     engine.run_until_idle()
     dense_tokens = fut.result(timeout=30).token_ids
+    dense_guided = gfut.result(timeout=30).text
 
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
@@ -108,9 +124,13 @@ def test_tp_engine_matches_dense(tmp_path):
     ]
     for p in procs:
         p.start()
-    tag, tp_tokens = q.get(timeout=150)
+    tag, tp_tokens, tp_guided = q.get(timeout=150)
     for p in procs:
         p.join(timeout=60)
         assert p.exitcode == 0
     assert tag == "tp"
     assert tp_tokens == dense_tokens
+    # constrained JSON decoding over the TP broadcast path: same text
+    import json
+
+    assert json.loads(tp_guided) == json.loads(dense_guided)
