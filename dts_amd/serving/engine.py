@@ -165,6 +165,14 @@ class ServingEngine:
                 f"prompt of {len(prompt_ids)} tokens exceeds max_position "
                 f"{self.spec.max_position}"
             )
+        if guide is not None and hasattr(guide, "token_budget"):
+            # a finite form defines its own output size; never let a
+            # free-text phase budget truncate it mid-form
+            need = guide.token_budget()
+            if params.max_tokens < need:
+                import dataclasses
+
+                params = dataclasses.replace(params, max_tokens=need)
         seq = Sequence(tokens=list(prompt_ids), params=params)
         seq.guide = guide
         seq.stream_cb = stream_cb

@@ -121,6 +121,29 @@ class FormGuide:
     def initial_forced(self) -> list:
         return self._collect_forced()
 
+    def token_budget(self) -> int:
+        """Worst-case total output tokens (forced + sampled) for the whole
+        form. Guided requests size their max_tokens from THIS, not the
+        textual phase budget: a finite form must never be truncated by a
+        budget sized for free-text (e.g. a 48-entry strategy form at
+        DP world 8 far exceeds the 1024-token strategy budget)."""
+        total = 0
+        for seg in self.segments:
+            if isinstance(seg, Fixed):
+                total += len(self._encode(seg.text))
+            elif isinstance(seg, Free):
+                total += seg.max_tokens + (1 if seg.stop is not None else 0)
+            elif isinstance(seg, Choice):
+                total += max(len(self._encode(c)) for c in seg.choices)
+            else:  # lazily-materialized marker (ranking id choice)
+                ids = getattr(self, "ids", None)
+                total += (
+                    max((len(self._encode(i)) for i in ids), default=8)
+                    if ids
+                    else 8
+                )
+        return total + 8
+
     def done(self) -> bool:
         return self._done
 

@@ -113,3 +113,28 @@ class TestEngineStress:
         assert p1 > 0, "pool sized to force preemption"
         for r, g_ in zip(ref, got):
             assert json.loads(r) == json.loads(g_)
+
+
+class TestGuideBudgetOverride:
+    """A finite form must complete even when the caller's textual
+    max_tokens budget is smaller than the form (regression: a 48-entry
+    strategy form at DP world 8 exceeded the 1024-token strategy budget
+    and died as truncated)."""
+
+    def test_large_form_overrides_small_budget(self):
+        from dts_amd.serving.structured import strategy_form
+
+        eng = make_engine("0", num_blocks=4096)
+        g = strategy_form(eng.tokenizer, 20)
+        assert g.token_budget() > 1024
+        fut = eng.submit_tokens(
+            list(range(1, 30)),
+            SamplingParams(max_tokens=64, temperature=0.0),
+            guide=g,
+        )
+        eng.run_until_idle()
+        r = fut.result(timeout=10)
+        assert r.finish_reason == "stop"
+        d = json.loads(r.text)
+        assert len(d["nodes"]) == 20
+        eng.stop()
