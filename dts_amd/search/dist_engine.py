@@ -19,9 +19,25 @@ from dts_amd.parallel.dp import (
     score_to_payload,
 )
 from dts_amd.search.engine import DTSEngine
-from dts_amd.search.tree import DialogueTree
-from dts_amd.search.types import DialogueNode
+from dts_amd.search.tree import DialogueTree, generate_node_id
+from dts_amd.search.types import DialogueNode, Strategy
 from dts_amd.utils.logging import log_phase
+
+# per-rank lenses for sharded strategy generation: each rank generates its
+# slice of the initial branches through a distinct angle, so the union
+# stays diverse without the single serial mega-call (one 6*world-entry
+# form decoded on rank 0 while the other GPUs idle was the Amdahl
+# bottleneck of the weak-scaling bench)
+_DIVERSITY_ANGLES = [
+    "empathy-first emotional connection",
+    "pragmatic step-by-step problem solving",
+    "data-driven evidence and benchmarks",
+    "narrative framing and storytelling",
+    "expert-authority framing with credible sourcing",
+    "collaborative co-design with the user",
+    "contrarian assumption-challenging",
+    "incremental trust-building and small commitments",
+]
 
 
 class _DPSimulator:
@@ -135,6 +151,84 @@ class DistributedDTSEngine(DTSEngine):
     async def _initialize_tree(self) -> DialogueTree:
         if not self.dp.enabled:
             return await super()._initialize_tree()
+        cfg = self.config
+        if (
+            self.dp.world >= 2
+            and cfg.init_branches >= self.dp.world
+            and cfg.init_branches % self.dp.world == 0
+        ):
+            return await self._initialize_tree_sharded()
+        return await self._initialize_tree_rank0()
+
+    async def _initialize_tree_sharded(self) -> DialogueTree:
+        """Every rank generates init_branches/world strategies on its own
+        GPU through a distinct diversity lens; an object all-gather plus
+        deterministic child ids (uuid5 over root:index) rebuilds the same
+        tree everywhere. Replaces the serial rank-0 mega-form."""
+        import uuid as _uuid
+
+        from dts_amd.llm.types import Message
+
+        cfg = self.config
+        world, rank = self.dp.world, self.dp.rank
+        if rank == 0:
+            root_id = generate_node_id()
+            deep = None
+            if cfg.deep_research and self._researcher is not None:
+                try:
+                    deep = await self._researcher.research(
+                        goal=cfg.goal, first_message=cfg.first_message
+                    )
+                except Exception:  # noqa: BLE001
+                    deep = None
+            meta = (root_id, deep)
+        else:
+            meta = None
+        root_id, deep = self.dp.broadcast_obj(meta, src=0)
+        self._research_report = deep
+        if deep:
+            self._evaluator.set_research_context(deep)
+        root = DialogueNode(
+            id=root_id, depth=0, messages=[Message.user(cfg.first_message)]
+        )
+        tree = DialogueTree.create(root)
+        self._emit("node_added", self._node_event(root))
+
+        local_n = cfg.init_branches // world
+        angle = _DIVERSITY_ANGLES[rank % len(_DIVERSITY_ANGLES)]
+        ctx = (f"{deep}\n\n" if deep else "") + (
+            "Diversity constraint: approach ALL of your strategies through "
+            f"the lens of {angle}; other strategy sets cover other lenses."
+        )
+        strategies = await self._generator.generate_strategies(
+            cfg.first_message, local_n, ctx
+        )
+        gathered = self.dp.all_gather_obj(
+            [(st.tagline, st.description) for st in strategies]
+        )
+        idx = 0
+        for plist in gathered:
+            for tagline, desc in plist:
+                child = DialogueNode(
+                    id=str(
+                        _uuid.uuid5(
+                            _uuid.NAMESPACE_URL, f"{root_id}:strategy:{idx}"
+                        )
+                    ),
+                    strategy=Strategy(tagline=tagline, description=str(desc)),
+                    messages=[Message.user(cfg.first_message)],
+                )
+                tree.add_child(root.id, child)
+                self._emit("node_added", self._node_event(child))
+                idx += 1
+        log_phase(
+            "INIT",
+            f"[rank {rank}] sharded strategy init: {idx} branches "
+            f"({local_n}/rank, lens: {angle})",
+        )
+        return tree
+
+    async def _initialize_tree_rank0(self) -> DialogueTree:
         if self.dp.rank == 0:
             tree = await super()._initialize_tree()
             root = tree.get_root()
