@@ -238,3 +238,89 @@ def test_dp_resume_identical_across_ranks():
         assert p.exitcode == 0
     assert results[0]["fingerprint"] == results[1]["fingerprint"]
     assert results[0]["best"] == results[1]["best"]
+
+
+def _judge_fail_worker(rank, world, port, out_q):
+    os.environ.update(
+        {
+            "RANK": str(rank),
+            "WORLD_SIZE": str(world),
+            "LOCAL_RANK": str(rank),
+            "MASTER_ADDR": "127.0.0.1",
+            "MASTER_PORT": str(port),
+        }
+    )
+    import asyncio
+
+    import torch.distributed as dist
+
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from dts_amd.llm import LLM, FakeBackend
+        from dts_amd.parallel.dp import DPContext
+        from dts_amd.search import DTSConfig
+        from dts_amd.search.dist_engine import DistributedDTSEngine
+
+        backend = FakeBackend(score_salt=f"r{rank}")
+        if rank == 1:
+            # rank 1's judges always blow up AFTER retries would too
+            orig = backend._respond
+
+            def failing(system, user, messages):
+                if "[dts:judge" in system:
+                    raise RuntimeError("injected judge failure on rank 1")
+                return orig(system, user, messages)
+
+            backend._respond = failing
+        llm = LLM(backend, default_model="fake")
+        cfg = DTSConfig(
+            goal="g",
+            first_message="hello can you help me with this?",
+            init_branches=4,
+            turns_per_branch=1,
+            user_intents_per_branch=1,
+            scoring_mode="absolute",
+            prune_threshold=0.0,
+            min_survivors=1,
+            seed=3,
+        )
+        engine = DistributedDTSEngine(llm, cfg, dp=DPContext())
+        result = asyncio.run(engine.run(rounds=1))
+        fp = sorted(
+            (n.id, n.status.value, len(n.stats.judge_scores),
+             round(n.stats.aggregated_score, 4))
+            for n in result.all_nodes
+        )
+        out_q.put((rank, {"fingerprint": fp, "best": result.best_node_id}))
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(240)
+def test_dp_asymmetric_judge_failure_converges():
+    """Rank 1's judge calls all fail; rank 0's succeed. Owner-computes
+    means the failure pattern is identical on every rank, so trees must
+    still converge (failed-judge branches carry zero scores everywhere,
+    not diverging ones)."""
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = 29600 + (os.getpid() * 4 + 2) % 800
+    procs = [
+        ctx.Process(target=_judge_fail_worker, args=(r, 2, port, q))
+        for r in range(2)
+    ]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(2):
+        rank, data = q.get(timeout=200)
+        results[rank] = data
+    for p in procs:
+        p.join(timeout=30)
+        assert p.exitcode == 0
+    assert results[0]["fingerprint"] == results[1]["fingerprint"]
+    assert results[0]["best"] == results[1]["best"]
+    # rank 0's judged branches carry real 3-judge scores; rank 1's carry
+    # the zero-score fallback — both visible identically on every rank
+    n_scored = sum(1 for *_, nsc, _s in results[0]["fingerprint"] if nsc == 3)
+    assert n_scored >= 1
