@@ -324,11 +324,16 @@ class ServingEngine:
                 logger.exception("engine step failed: %s", e)
                 with self._lock:
                     for seq in list(self.scheduler.running):
-                        fut = self._futures.pop(seq.seq_id, None)
-                        if fut and not fut.done():
-                            fut.set_exception(e)
                         self.sampler.release(seq)
                         self.scheduler.abort(seq)
+                    # fail EVERY pending future (waiting/held requests
+                    # included) — a broken step means no one behind it
+                    # will ever be served, and a silent hang is worse
+                    # than a loud error
+                    for sid, fut in list(self._futures.items()):
+                        if not fut.done():
+                            fut.set_exception(e)
+                        self._futures.pop(sid, None)
 
     def start(self) -> None:
         if self._thread is None:
