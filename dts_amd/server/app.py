@@ -159,6 +159,41 @@ def create_app(llm_factory: Optional[Callable[[], LLM]] = None) -> FastAPI:
             )
         return {"data": out}
 
+    @app.get("/metrics")
+    async def metrics():
+        """Prometheus metrics (framework addition — the reference has no
+        metrics endpoint, SURVEY.md §5 Tracing: observability is logs
+        only). Exposes the live serving engines' counters."""
+        from fastapi.responses import PlainTextResponse
+        from prometheus_client import (
+            CollectorRegistry,
+            Gauge,
+            generate_latest,
+        )
+
+        registry = CollectorRegistry()
+        g = Gauge(
+            "dts_engine_stat",
+            "dts_amd serving-engine counter",
+            ["model", "stat"],
+            registry=registry,
+        )
+        llm = app.state.llm
+        backend = getattr(llm, "backend", None) if llm else None
+        engines = getattr(backend, "engines", None) or {}
+        for name, eng in engines.items():
+            try:
+                for stat, value in eng.cache_stats.items():
+                    if isinstance(value, bool):
+                        value = int(value)
+                    if isinstance(value, (int, float)):
+                        g.labels(model=name, stat=stat).set(value)
+            except Exception:  # noqa: BLE001 — metrics must never 500
+                continue
+        return PlainTextResponse(
+            generate_latest(registry), media_type="text/plain; version=0.0.4"
+        )
+
     @app.get("/", response_class=HTMLResponse)
     async def index():
         static_index = Path(__file__).parent / "static" / "index.html"

@@ -120,3 +120,35 @@ class TestWebSocket:
             assert {"summary", "best_branch", "branches", "research_report"} <= set(
                 exploration
             )
+
+
+def test_metrics_endpoint():
+    """Prometheus /metrics exposes the live engine counters."""
+    import torch
+    from fastapi.testclient import TestClient
+
+    from dts_amd.llm import LLM
+    from dts_amd.server.app import create_app
+    from dts_amd.serving import LocalBackend, ServingEngine
+
+    eng = ServingEngine(
+        model_name="llama-tiny",
+        device="cpu",
+        dtype=torch.float32,
+        num_blocks=64,
+        block_size=8,
+        weight_seed=1,
+    )
+    backend = LocalBackend.single(eng, name="llama-tiny")
+    app = create_app(llm_factory=lambda: LLM(backend, default_model="llama-tiny"))
+    client = TestClient(app)
+    # force backend creation (metrics reads app.state.llm lazily)
+    with client.websocket_connect("/ws") as ws:
+        ws.send_text('{"type": "ping"}')
+        assert ws.receive_json()["type"] == "pong"
+    app.state.llm = LLM(backend, default_model="llama-tiny")
+    r = client.get("/metrics")
+    assert r.status_code == 200
+    assert 'dts_engine_stat{model="llama-tiny",stat="free_blocks"}' in r.text
+    assert 'stat="preemptions"' in r.text
+    backend.shutdown()
