@@ -138,3 +138,36 @@ class TestGuideBudgetOverride:
         d = json.loads(r.text)
         assert len(d["nodes"]) == 20
         eng.stop()
+
+
+class TestSoakNoStateGrowth:
+    """Thousands of short requests: every per-request datum must be
+    reclaimed (futures, sampler generators, scheduler seqs, KV blocks)."""
+
+    def test_soak(self):
+        eng = make_engine("1", num_blocks=256)
+        rng = random.Random(0)
+        done = 0
+        for wave in range(40):
+            futs = [
+                eng.submit_tokens(
+                    [rng.randrange(1, 500) for _ in range(rng.randrange(4, 60))],
+                    SamplingParams(
+                        max_tokens=rng.randrange(1, 12),
+                        temperature=0.7,
+                        seed=rng.randrange(10_000),
+                    ),
+                )
+                for _ in range(25)
+            ]
+            eng.run_until_idle()
+            for f in futs:
+                f.result(timeout=5)
+                done += 1
+        assert done == 1000
+        assert not eng._futures
+        assert not eng.sampler._generators
+        assert not eng.scheduler.has_work()
+        # native core retains no sequences
+        assert not eng.scheduler._seqs
+        eng.stop()
