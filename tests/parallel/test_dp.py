@@ -52,7 +52,7 @@ def _worker(rank, world, port, out_q, forking=False):
         )
         dp = DPContext()
         engine = DistributedDTSEngine(llm, cfg, dp=dp)
-        result = asyncio.run(engine.run(rounds=1))
+        result = asyncio.run(engine.run(rounds=2 if not forking else 1))
         tree_fingerprint = sorted(
             (
                 n.id,
@@ -104,12 +104,14 @@ def test_dp_spmd_two_ranks():
     assert r0["best_score"] == r1["best_score"]
     # root + 4 branches
     assert r0["n_nodes"] == 5
-    # expansion/judging actually sharded: rank1 (non-strategy rank) makes
-    # strictly fewer calls than a single-rank run would (1 + 4*2*2 + 4*3 = 29)
-    assert r1["n_llm_calls"] < 29
-    assert r0["n_llm_calls"] < 29
-    # together they cover all the work (strategy on rank0 only)
-    assert r0["n_llm_calls"] + r1["n_llm_calls"] >= 1 + 4 * 2 * 2 + 4 * 3
+    # expansion/judging actually sharded over 2 linear rounds: a
+    # single-rank run would make 2 strategy-shard calls + 2 rounds x
+    # (4 branches x 2 calls + 4 x 3 judges) = 42; each rank does less
+    single = 2 + 2 * (4 * 2 * 1 + 4 * 3)
+    assert r1["n_llm_calls"] < single
+    assert r0["n_llm_calls"] < single
+    # together they cover all the work
+    assert r0["n_llm_calls"] + r1["n_llm_calls"] >= single
 
 
 @pytest.mark.timeout(180)
