@@ -206,9 +206,19 @@ class DistributedDTSEngine(DTSEngine):
         gathered = self.dp.all_gather_obj(
             [(st.tagline, st.description) for st in strategies]
         )
+        total = sum(len(p) for p in gathered)
         idx = 0
         for plist in gathered:
             for tagline, desc in plist:
+                self._emit(
+                    "strategy_generated",
+                    {
+                        "index": idx + 1,
+                        "total": total,
+                        "tagline": tagline,
+                        "description": desc,
+                    },
+                )
                 child = DialogueNode(
                     id=str(
                         _uuid.uuid5(
