@@ -221,3 +221,37 @@ class TestPrefixSharingCorrectnessFuzz:
             cold.run_until_idle()
             assert f.result(timeout=10).token_ids == expect
             cold.stop()
+
+
+class TestBatchCompositionInvariance:
+    """A request's output must not depend on what else is in the batch —
+    per-sequence RNG streams and row-indexed sampling, not batch-shared
+    state."""
+
+    @pytest.mark.parametrize("temperature", [0.0, 0.8])
+    def test_same_output_solo_vs_cobatched(self, temperature):
+        prompt = list(range(7, 47))
+        params = dict(max_tokens=12, temperature=temperature, seed=123)
+
+        eng1 = make_engine("1", num_blocks=512)
+        f = eng1.submit_tokens(list(prompt), SamplingParams(**params))
+        eng1.run_until_idle()
+        solo = f.result(timeout=10).token_ids
+        eng1.stop()
+
+        eng2 = make_engine("1", num_blocks=512)
+        noise = [
+            eng2.submit_tokens(
+                list(range(60 + 13 * i, 90 + 13 * i)),
+                SamplingParams(max_tokens=15, temperature=0.9, seed=i),
+            )
+            for i in range(5)
+        ]
+        f = eng2.submit_tokens(list(prompt), SamplingParams(**params))
+        eng2.run_until_idle()
+        cobatched = f.result(timeout=10).token_ids
+        for n in noise:
+            n.result(timeout=10)
+        eng2.stop()
+
+        assert solo == cobatched
