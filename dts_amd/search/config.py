@@ -31,6 +31,11 @@ class DTSConfig:
     init_branches: int = 6
     deep_research: bool = False
     research_cache_dir: str = ".cache/research"
+    # mid-run crash recovery (framework addition — the reference persists
+    # the tree-state JSON only at the end): after every round the engine
+    # atomically writes the exploration dict here; resume with
+    # DTSEngine.run(resume_from=checkpoint_path). DP: rank 0 writes.
+    checkpoint_path: Optional[str] = None
     turns_per_branch: int = 5
     user_intents_per_branch: int = 3
     user_variability: bool = False

@@ -267,6 +267,7 @@ class DTSEngine:
                     }
                 },
             )
+            self._save_checkpoint(tree, round_num + 1)
 
         best = tree.best_leaf_by_score()
         log_phase(
@@ -295,6 +296,42 @@ class DTSEngine:
             total_rounds=rounds,
             research_report=self._research_report,
         )
+
+    def _interim_result(self, tree: DialogueTree, rounds_done: int) -> DTSRunResult:
+        best = tree.best_leaf_by_score()
+        return DTSRunResult(
+            best_node_id=best.id if best else None,
+            best_score=best.stats.aggregated_score if best else 0.0,
+            best_messages=list(best.messages) if best else [],
+            all_nodes=tree.all_nodes(),
+            pruned_count=sum(
+                1 for n in tree.all_nodes() if n.status == NodeStatus.PRUNED
+            ),
+            token_usage=self._token_tracker.to_dict(),
+            total_rounds=rounds_done,
+            research_report=self._research_report,
+        )
+
+    def _save_checkpoint(self, tree: DialogueTree, rounds_done: int) -> None:
+        """Atomic per-round tree-state write for crash recovery (no
+        reference analogue — SURVEY.md §5 Checkpoint: final JSON only)."""
+        path = self.config.checkpoint_path
+        if not path:
+            return
+        if getattr(self, "dp", None) is not None and getattr(self.dp, "rank", 0) != 0:
+            return  # identical trees on every rank; rank 0 writes
+        import json as _json
+        import os as _os
+
+        try:
+            d = self._interim_result(tree, rounds_done).to_exploration_dict()
+            tmp = f"{path}.tmp"
+            with open(tmp, "w") as f:
+                _json.dump(d, f)
+            _os.replace(tmp, path)
+            logger.info("checkpoint after round %d -> %s", rounds_done, path)
+        except Exception as e:  # noqa: BLE001 — never kill the search
+            logger.warning("checkpoint write failed: %s", e)
 
     # ------------------------------------------------------------------
     async def _initialize_tree(self) -> DialogueTree:

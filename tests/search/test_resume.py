@@ -104,3 +104,25 @@ class TestResume:
                         if m.id == n.id
                     ).messages
                 ), "pruned branches must not be expanded on resume"
+
+
+class TestPeriodicCheckpoint:
+    def test_written_each_round_and_resumable(self, tmp_path):
+        path = tmp_path / "auto.json"
+        engine, _ = make_engine(checkpoint_path=str(path))
+        r = asyncio.run(engine.run(rounds=2))
+        assert path.exists()
+        d = json.loads(path.read_text())
+        # checkpoint reflects the completed search state
+        assert d["summary"]["total_rounds"] == 2
+        assert len(d["branches"]) == 3
+        # and a crash at this point is resumable
+        engine2, _ = make_engine()
+        r2 = asyncio.run(engine2.run(rounds=1, resume_from=str(path)))
+        assert r2.best_node_id is not None
+
+    def test_no_checkpoint_by_default(self, tmp_path, monkeypatch):
+        monkeypatch.chdir(tmp_path)
+        engine, _ = make_engine()
+        asyncio.run(engine.run(rounds=1))
+        assert not list(tmp_path.glob("*.json"))
