@@ -37,6 +37,9 @@ class SearchRequest(BaseModel):
     # `rounds` more rounds on the restored tree instead of generating
     # fresh strategies)
     resume_from: Optional[dict] = Field(default=None)
+    # additive: per-round crash-recovery checkpoint file (see
+    # DTSConfig.checkpoint_path)
+    checkpoint_path: Optional[str] = Field(default=None)
 
 
 class EventMessage(BaseModel):

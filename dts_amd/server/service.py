@@ -36,6 +36,7 @@ def create_dts_config(request: SearchRequest) -> DTSConfig:
         judge_model=request.judge_model,
         user_variability=request.user_variability,
         reasoning_enabled=request.reasoning_enabled,
+        checkpoint_path=request.checkpoint_path,
     )
 
 

@@ -27,6 +27,7 @@ async def run_dts_example(
     deep_research: bool = True,
     output_path: str = "dts_output.json",
     resume_from: str | None = None,
+    checkpoint_path: str | None = None,
 ) -> float:
     use_gpu = torch.cuda.is_available()
     model = "llama-3-8b" if use_gpu else "llama-tiny"
@@ -49,6 +50,7 @@ async def run_dts_example(
         scoring_mode="comparative",
         prune_threshold=5.0,
         deep_research=deep_research,
+        checkpoint_path=checkpoint_path,
         seed=0,
     )
     researcher = DeepResearcher(llm, cache_dir=config.research_cache_dir)
@@ -66,7 +68,10 @@ if __name__ == "__main__":
     ap = argparse.ArgumentParser()
     ap.add_argument("--rounds", type=int, default=2)
     ap.add_argument("--resume", default=None, help="exploration JSON to resume")
+    ap.add_argument("--checkpoint", default=None,
+                    help="write the tree state here after every round")
     a = ap.parse_args()
     start = time.time()
-    asyncio.run(run_dts_example(rounds=a.rounds, resume_from=a.resume))
+    asyncio.run(run_dts_example(rounds=a.rounds, resume_from=a.resume,
+                                checkpoint_path=a.checkpoint))
     print(f"Completed in {time.time() - start:.1f}s")
