@@ -228,6 +228,11 @@ class ServingEngine:
             else:
                 self.eager_decode_steps += 1
         sampled_seqs = batch._sampled_seqs  # type: ignore[attr-defined]
+        # NOTE on timer attribution: kernel launches above are async, so
+        # t_forward_* measures launch/replay time only; the first device
+        # sync happens inside sample(), so t_sample absorbs the GPU
+        # execution wait of the whole step. Per-kernel truth lives in
+        # rocprof (profiles/), not these host timers.
         tokens = self.sampler.sample(logits, sampled_seqs) if sampled_seqs else []
         t2 = _time.perf_counter()
         self.t_sample += t2 - t1
