@@ -267,7 +267,9 @@ class TrajectoryEvaluator:
         seed = None
         if self.seed is not None:
             self._judge_counter += 1
-            seed = hash((self.seed, "judge", self._judge_counter)) & 0x7FFFFFFF
+            from dts_amd.utils.seeding import stable_seed
+
+            seed = stable_seed(self.seed, "judge", self._judge_counter)
         completion = await self.llm.complete(
             [Message.system(system), Message.user(user)],
             model=self.model,

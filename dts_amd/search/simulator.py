@@ -317,8 +317,17 @@ class ConversationSimulator:
         async with self._sem:
             seed = None
             if self.seed is not None:
-                # derive a distinct deterministic seed per call
-                seed = hash((self.seed, phase, len(messages), attempt)) & 0x7FFFFFFF
+                # distinct, run-stable seed per call; the last message's
+                # content decorrelates sibling branches at the same turn
+                from dts_amd.utils.seeding import stable_seed
+
+                seed = stable_seed(
+                    self.seed,
+                    phase,
+                    len(messages),
+                    attempt,
+                    messages[-1].content if messages else "",
+                )
             completion = await self.llm.complete(
                 messages,
                 model=self.model,

@@ -107,7 +107,10 @@ class Sampler:
 
             seeds = torch.tensor(
                 [
-                    (hash((s.params.seed or 0, s.seq_id, len(s.tokens))) & 0x7FFFFFFF)
+                    # (seed, position) only — the stream must not depend
+                    # on seq_id (submission order / co-batched traffic)
+                    # and stays identical across a preempt+recompute
+                    (hash((s.params.seed or 0, len(s.tokens))) & 0x7FFFFFFF)
                     if s.params.seed is not None
                     else _random.getrandbits(31)
                     for s in free_seqs
