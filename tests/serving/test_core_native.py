@@ -202,3 +202,14 @@ def test_lockstep_under_memory_pressure(seed):
         nat.abort(nat._seqs[ids[0]])
     assert py.bm.num_free() == py.bm.num_blocks, "python scheduler leaked blocks"
     assert nat.num_free() == py.bm.num_free(), "native scheduler leaked blocks"
+
+
+def test_core_api_tolerates_bad_ids():
+    core = load_core()
+    cs = core.CoreScheduler(8, 4, 1024, 8)
+    cs.add(1, [1, 2, 3])
+    cs.finish(1)
+    cs.finish(1)  # double finish: no-op, no crash, no double free
+    cs.abort(99)  # unknown ids: no-op
+    cs.finish(42)
+    assert cs.num_free() == 8

@@ -58,3 +58,12 @@ class TestChatTemplate:
     def test_stop_tokens(self, tok):
         tpl = ChatTemplate(tok)
         assert tok.eot_id in tpl.stop_token_ids
+
+
+def test_unicode_roundtrip():
+    """UTF-8 byte-level: multibyte scripts and emoji round-trip exactly."""
+    from dts_amd.serving.tokenizer import SyntheticTokenizer
+
+    tok = SyntheticTokenizer(128256)
+    for text in ["héllo wörld", "日本語テスト", "emoji 🎉 test", ""]:
+        assert tok.decode(tok.encode(text)) == text
