@@ -33,7 +33,12 @@ from dts_amd.serving.kv_cache import BlockManager, KVCachePool
 from dts_amd.serving.sampler import Sampler
 from dts_amd.serving.scheduler import Scheduler
 from dts_amd.serving.sequence import Sequence
-from dts_amd.serving.tokenizer import ChatTemplate, SyntheticTokenizer
+from dts_amd.serving.tokenizer import (
+    ChatTemplate,
+    HFChatTemplate,
+    HFTokenizer,
+    load_tokenizer,
+)
 from dts_amd.utils.logging import logger
 
 
@@ -59,6 +64,7 @@ class ServingEngine:
         max_running: int = 256,
         weight_seed: int = 0,
         model: Optional[object] = None,
+        tokenizer_path: Optional[str] = None,
     ) -> None:
         self.spec: ModelSpec = get_model_spec(model_name)
         if device is None:
@@ -118,8 +124,19 @@ class ServingEngine:
         if self.scheduler is None:
             self.scheduler = Scheduler(self.block_manager, max_batch_tokens, max_running)
         self.sampler = Sampler(device)
-        self.tokenizer = SyntheticTokenizer(self.spec.vocab_size)
-        self.template = ChatTemplate(self.tokenizer)
+        # a real tokenizer JSON (HF `tokenizers` format) when given —
+        # its vocab must fit the model's embedding width; else the
+        # synthetic byte-level tokenizer at full vocab width
+        self.tokenizer = load_tokenizer(self.spec.vocab_size, tokenizer_path)
+        if isinstance(self.tokenizer, HFTokenizer):
+            if self.tokenizer.vocab_size > self.spec.vocab_size:
+                raise ValueError(
+                    f"tokenizer vocab {self.tokenizer.vocab_size} exceeds "
+                    f"model vocab {self.spec.vocab_size}"
+                )
+            self.template = HFChatTemplate(self.tokenizer)
+        else:
+            self.template = ChatTemplate(self.tokenizer)
 
         self._graph_runner = None
         if device.startswith("cuda") and os.environ.get("DTS_NO_HIPGRAPH") != "1":

@@ -82,12 +82,33 @@ class FormGuide:
 
     def __init__(self, tokenizer, segments: list) -> None:
         self.tok = tokenizer
+        # byte-level masks translate through the tokenizer when its ids
+        # are not raw bytes (HFTokenizer); identity for the synthetic one
+        bmap = getattr(tokenizer, "byte_token_map", None)
+        self._b2t = dict(bmap) if bmap is not None else None
+        self._t2b = {t: b for b, t in bmap.items()} if bmap is not None else None
         self.segments: list = []
         for seg in segments:
             if isinstance(seg, list):
                 self.segments.extend(seg)
             else:
                 self.segments.append(seg)
+        if self._b2t is not None:
+            need = set()
+            for seg in self.segments:
+                if isinstance(seg, Free):
+                    need.update(seg.charset)
+                    if seg.stop is not None:
+                        need.add(seg.stop)
+                elif isinstance(seg, Choice):
+                    for c in seg.choices:
+                        need.update(c.encode("utf-8"))
+            missing = sorted(b for b in need if b not in self._b2t)
+            if missing:
+                raise ValueError(
+                    "tokenizer cannot express form charset as single "
+                    f"tokens (missing bytes: {missing[:8]}...)"
+                )
         self._i = 0  # current segment
         self._free_count = 0
         self._choice_state: Optional[list] = None  # remaining candidate strs
@@ -157,6 +178,8 @@ class FormGuide:
             if seg.stop is not None and self._free_count >= seg.min_tokens:
                 if seg.stop not in allowed:
                     allowed.append(seg.stop)
+            if self._b2t is not None:
+                allowed = [self._b2t[b] for b in allowed]
             return allowed
         if isinstance(seg, Choice):
             nxt = set()
@@ -164,6 +187,8 @@ class FormGuide:
                 bs = c.encode("utf-8")
                 if self._choice_pos < len(bs):
                     nxt.add(bs[self._choice_pos])
+            if self._b2t is not None:
+                return sorted(self._b2t[b] for b in nxt) or None
             return sorted(nxt) or None
         return None
 
@@ -171,6 +196,8 @@ class FormGuide:
         """Advance; returns forced tokens to append after this one."""
         if self._done:
             return []
+        if self._t2b is not None:
+            tok = self._t2b.get(tok, -1)  # compare in byte space below
         seg = self.segments[self._i]
         if isinstance(seg, Free):
             self._free_count += 1
@@ -185,7 +212,10 @@ class FormGuide:
                 # budget exhausted before terminator: close the segment by
                 # forcing the stop byte
                 self._i += 1
-                return [seg.stop] + self._collect_forced()
+                stop_tok = (
+                    self._b2t[seg.stop] if self._b2t is not None else seg.stop
+                )
+                return [stop_tok] + self._collect_forced()
             if ended:
                 self._i += 1
                 return self._collect_forced()
@@ -343,6 +373,15 @@ class _RankingGuide(FormGuide):
             segs += [Fixed(', "reason": '), JsonString(40), Fixed("}")]
         segs += [Fixed('], "ranking_confidence": "'), Choice(CONFIDENCE), Fixed('"}')]
         super().__init__(tok, segs)
+        # the id choices materialize lazily (markers), so validate their
+        # bytes against the byte-token map here
+        if self._b2t is not None:
+            need = {b for i in ids for b in i.encode("utf-8")}
+            missing = sorted(b for b in need if b not in self._b2t)
+            if missing:
+                raise ValueError(
+                    f"tokenizer cannot express ranking ids (bytes {missing[:8]})"
+                )
 
     def _collect_forced(self) -> list:
         forced = super()._collect_forced()

@@ -101,10 +101,96 @@ class ChatTemplate:
         return [self.tok.eot_id, self.tok.eos_id]
 
 
-def load_tokenizer(spec_vocab_size: int, path: Optional[str] = None):
-    """Return an HF tokenizer when a local file is given, else synthetic."""
-    if path:
+class HFTokenizer:
+    """Adapter over a local `tokenizers` JSON file with the same duck-type
+    surface the engine uses (encode -> list[int], decode -> str, special
+    ids, role headers). Special tokens are probed by name so both
+    Llama-3-style (<|begin_of_text|>/<|eot_id|>) and classic BPE
+    (<s>/</s>) vocabularies work."""
+
+    def __init__(self, path: str) -> None:
         from tokenizers import Tokenizer  # local wheel, no network
 
-        return Tokenizer.from_file(path)
+        self.tk = Tokenizer.from_file(path)
+        self.vocab_size = self.tk.get_vocab_size()
+
+        def probe(*names):
+            for n in names:
+                i = self.tk.token_to_id(n)
+                if i is not None:
+                    return i
+            return None
+
+        self.bos_id = probe("<|begin_of_text|>", "<s>", "<bos>")
+        self.eos_id = probe("<|end_of_text|>", "</s>", "<eos>")
+        self.eot_id = probe("<|eot_id|>", "<|im_end|>")
+        if self.eot_id is None:
+            self.eot_id = self.eos_id
+        self._hdr_start = probe("<|start_header_id|>", "<|im_start|>")
+        self._hdr_end = probe("<|end_header_id|>")
+        pad = probe("<pad>", "<|pad|>")
+        self.pad_id = pad if pad is not None else (self.eos_id or 0)
+
+    def encode(self, text: str, add_bos: bool = False) -> list:
+        ids = self.tk.encode(text, add_special_tokens=False).ids
+        if add_bos and self.bos_id is not None:
+            return [self.bos_id] + ids
+        return ids
+
+    def decode(self, ids) -> str:
+        return self.tk.decode(list(ids), skip_special_tokens=True)
+
+    @property
+    def byte_token_map(self) -> dict:
+        """byte value -> single token id, for every printable byte the
+        vocab can express as ONE token. Constrained-JSON guides
+        (serving/structured.py) translate their byte-level masks through
+        this; a form whose charset is not covered raises loudly."""
+        if not hasattr(self, "_byte_map"):
+            m = {}
+            for b in range(32, 127):
+                ids = self.tk.encode(chr(b), add_special_tokens=False).ids
+                if len(ids) == 1:
+                    m[b] = ids[0]
+            self._byte_map = m
+        return self._byte_map
+
+
+class HFChatTemplate:
+    """Chat layout over an HFTokenizer: the Llama-3 header layout when the
+    header specials exist, else a plain-text role header. Token-exact so
+    block-aligned prefix reuse stays deterministic."""
+
+    def __init__(self, tok: HFTokenizer) -> None:
+        self.tok = tok
+
+    def _header(self, role: str) -> list:
+        t = self.tok
+        if t._hdr_start is not None and t._hdr_end is not None:
+            return [t._hdr_start] + t.encode(role) + [t._hdr_end] + t.encode("\n\n")
+        return t.encode(f"<{role}>\n")
+
+    def render(self, messages: list, add_generation_prompt: bool = True) -> list:
+        t = self.tok
+        ids = [t.bos_id] if t.bos_id is not None else []
+        for m in messages:
+            ids.extend(self._header(m.role))
+            ids.extend(t.encode(m.content or ""))
+            if t.eot_id is not None:
+                ids.append(t.eot_id)
+        if add_generation_prompt:
+            ids.extend(self._header("assistant"))
+        return ids
+
+    @property
+    def stop_token_ids(self) -> list:
+        out = [i for i in (self.tok.eot_id, self.tok.eos_id) if i is not None]
+        return out or [0]
+
+
+def load_tokenizer(spec_vocab_size: int, path: Optional[str] = None):
+    """Return an HFTokenizer when a local JSON file is given, else the
+    synthetic byte-level tokenizer at the spec's vocab width."""
+    if path:
+        return HFTokenizer(path)
     return SyntheticTokenizer(spec_vocab_size)
