@@ -218,6 +218,8 @@ def main() -> None:
     p.add_argument("--port", type=int, default=8000)
     p.add_argument("--model", default="llama-3-8b")
     p.add_argument("--device", default=None)
+    p.add_argument("--tokenizer", default=None,
+                   help="local HF tokenizers JSON (default: synthetic)")
     args = p.parse_args()
 
     def factory() -> LLM:
@@ -226,7 +228,9 @@ def main() -> None:
         from dts_amd.serving import LocalBackend, ServingEngine
 
         device = args.device or ("cuda" if torch.cuda.is_available() else "cpu")
-        engine = ServingEngine(model_name=args.model, device=device)
+        engine = ServingEngine(
+            model_name=args.model, device=device, tokenizer_path=args.tokenizer
+        )
         backend = LocalBackend.single(engine, name=args.model)
         return LLM(backend, default_model=args.model)
 
