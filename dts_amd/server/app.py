@@ -220,6 +220,8 @@ def main() -> None:
     p.add_argument("--device", default=None)
     p.add_argument("--tokenizer", default=None,
                    help="local HF tokenizers JSON (default: synthetic)")
+    p.add_argument("--weights", default=None,
+                   help="HF-layout safetensors dir (default: random-init)")
     args = p.parse_args()
 
     def factory() -> LLM:
@@ -229,7 +231,10 @@ def main() -> None:
 
         device = args.device or ("cuda" if torch.cuda.is_available() else "cpu")
         engine = ServingEngine(
-            model_name=args.model, device=device, tokenizer_path=args.tokenizer
+            model_name=args.model,
+            device=device,
+            tokenizer_path=args.tokenizer,
+            weights_path=args.weights,
         )
         backend = LocalBackend.single(engine, name=args.model)
         return LLM(backend, default_model=args.model)

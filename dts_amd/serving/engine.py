@@ -65,6 +65,7 @@ class ServingEngine:
         weight_seed: int = 0,
         model: Optional[object] = None,
         tokenizer_path: Optional[str] = None,
+        weights_path: Optional[str] = None,
     ) -> None:
         self.spec: ModelSpec = get_model_spec(model_name)
         if device is None:
@@ -78,7 +79,18 @@ class ServingEngine:
             self.model = model
         else:
             self.model = build_model(self.spec, dtype=dtype, device=device)
-            self.model.random_init(seed=weight_seed)
+            if weights_path:
+                if self.spec.arch != "llama":
+                    raise NotImplementedError(
+                        f"safetensors loading for arch {self.spec.arch!r} "
+                        "— pass a pre-loaded model= instead"
+                    )
+                from dts_amd.models.weights import load_llama_safetensors
+
+                n = load_llama_safetensors(self.model, weights_path)
+                logger.info("loaded %d tensors from %s", n, weights_path)
+            else:
+                self.model.random_init(seed=weight_seed)
 
         kv_heads = getattr(self.model, "num_kv_heads_local", None)
         if kv_heads is None:

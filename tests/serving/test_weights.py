@@ -22,3 +22,50 @@ def test_safetensors_roundtrip(tmp_path):
     ):
         assert n1 == n2
         assert torch.equal(p1, p2), n1
+
+
+def test_engine_loads_safetensors_checkpoint(tmp_path):
+    """ServingEngine(weights_path=...) serves a real HF-layout checkpoint:
+    outputs must match an engine running the source model directly."""
+    import torch
+
+    from dts_amd.llm.types import SamplingParams
+    from dts_amd.models.config import get_model_spec
+    from dts_amd.models.llama import LlamaModel
+    from dts_amd.models.weights import save_llama_safetensors
+    from dts_amd.serving import ServingEngine
+
+    src = LlamaModel(get_model_spec("llama-tiny"), dtype=torch.float32, device="cpu")
+    src.random_init(seed=77)
+    save_llama_safetensors(src, str(tmp_path))
+
+    def run(eng):
+        f = eng.submit_tokens(
+            list(range(1, 40)), SamplingParams(max_tokens=8, temperature=0.0)
+        )
+        eng.run_until_idle()
+        out = f.result(timeout=10).token_ids
+        eng.stop()
+        return out
+
+    direct = run(
+        ServingEngine(
+            model_name="llama-tiny",
+            device="cpu",
+            dtype=torch.float32,
+            num_blocks=128,
+            block_size=8,
+            model=src,
+        )
+    )
+    loaded = run(
+        ServingEngine(
+            model_name="llama-tiny",
+            device="cpu",
+            dtype=torch.float32,
+            num_blocks=128,
+            block_size=8,
+            weights_path=str(tmp_path),
+        )
+    )
+    assert direct == loaded
