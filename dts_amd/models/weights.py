@@ -122,3 +122,106 @@ def save_llama_safetensors(model, path: str) -> None:
     out["lm_head.weight"] = model.lm_head.weight.data.clone()
     Path(path).mkdir(parents=True, exist_ok=True)
     save_file(out, str(Path(path) / "model.safetensors"))
+
+
+def load_mixtral_safetensors(model, path: str) -> int:
+    """Load an HF-format Mixtral checkpoint into a MixtralModel.
+
+    HF names: self_attn.{q,k,v,o}_proj (TP-sharded like Llama),
+    block_sparse_moe.gate (router, replicated) and
+    block_sparse_moe.experts.{e}.{w1,w3,w2} fused into the grouped
+    [E_local, 2I, H] / [E_local, H, I] expert tensors (EP-sharded).
+    """
+    from safetensors import safe_open
+
+    tp = model.tp
+    files = sorted(Path(path).glob("*.safetensors"))
+    if not files:
+        raise FileNotFoundError(f"no .safetensors under {path}")
+    tensors: dict = {}
+    for f in files:
+        with safe_open(str(f), framework="pt") as sf:
+            for name in sf.keys():
+                tensors[name] = sf.get_tensor(name)
+
+    def get(name: str) -> torch.Tensor:
+        if name not in tensors:
+            raise KeyError(f"missing tensor {name}")
+        return tensors[name].to(torch.float32)
+
+    consumed = 0
+    with torch.no_grad():
+        model.embed.copy_(get("model.embed_tokens.weight").to(model.dtype))
+        consumed += 1
+        for i, layer in enumerate(model.layers):
+            p = f"model.layers.{i}."
+            layer.input_norm_w.copy_(get(p + "input_layernorm.weight").to(model.dtype))
+            layer.post_norm_w.copy_(
+                get(p + "post_attention_layernorm.weight").to(model.dtype)
+            )
+            q = _shard(get(p + "self_attn.q_proj.weight"), 0, tp.rank, tp.size)
+            k = _shard(get(p + "self_attn.k_proj.weight"), 0, tp.rank, tp.size)
+            v = _shard(get(p + "self_attn.v_proj.weight"), 0, tp.rank, tp.size)
+            layer.attn.qkv_proj.weight.copy_(
+                torch.cat([q, k, v], dim=0).to(model.dtype)
+            )
+            layer.attn.o_proj.weight.copy_(
+                _shard(get(p + "self_attn.o_proj.weight"), 1, tp.rank, tp.size).to(
+                    model.dtype
+                )
+            )
+            moe = layer.moe
+            moe.router_w.copy_(
+                get(p + "block_sparse_moe.gate.weight").to(model.dtype)
+            )
+            consumed += 6
+            for el in range(moe.experts_local):
+                e = moe.expert_offset + el
+                ep = p + f"block_sparse_moe.experts.{e}."
+                w1 = get(ep + "w1.weight")  # [I, H] gate
+                w3 = get(ep + "w3.weight")  # [I, H] up
+                w2 = get(ep + "w2.weight")  # [H, I] down
+                moe.gate_up_w[el].copy_(torch.cat([w1, w3], dim=0).to(model.dtype))
+                moe.down_w[el].copy_(w2.to(model.dtype))
+                consumed += 3
+        model.final_norm_w.copy_(get("model.norm.weight").to(model.dtype))
+        lm = tensors.get("lm_head.weight", tensors.get("model.embed_tokens.weight"))
+        model.lm_head.weight.copy_(
+            _shard(lm.to(torch.float32), 0, tp.rank, tp.size).to(model.dtype)
+        )
+        consumed += 2
+    logger.info("loaded %d tensors from %s", consumed, path)
+    return consumed
+
+
+def save_mixtral_safetensors(model, path: str) -> None:
+    """Write a MixtralModel out in HF Mixtral naming (single shard,
+    TP/EP size 1)."""
+    from safetensors.torch import save_file
+
+    out = {"model.embed_tokens.weight": model.embed.detach().cpu()}
+    spec = model.spec
+    for i, layer in enumerate(model.layers):
+        p = f"model.layers.{i}."
+        out[p + "input_layernorm.weight"] = layer.input_norm_w.detach().cpu()
+        out[p + "post_attention_layernorm.weight"] = layer.post_norm_w.detach().cpu()
+        qkv = layer.attn.qkv_proj.weight.detach().cpu()
+        qs = spec.num_heads * spec.head_dim
+        ks = spec.num_kv_heads * spec.head_dim
+        out[p + "self_attn.q_proj.weight"] = qkv[:qs]
+        out[p + "self_attn.k_proj.weight"] = qkv[qs : qs + ks]
+        out[p + "self_attn.v_proj.weight"] = qkv[qs + ks :]
+        out[p + "self_attn.o_proj.weight"] = layer.attn.o_proj.weight.detach().cpu()
+        moe = layer.moe
+        out[p + "block_sparse_moe.gate.weight"] = moe.router_w.detach().cpu()
+        for e in range(spec.num_experts):
+            ep = p + f"block_sparse_moe.experts.{e}."
+            gu = moe.gate_up_w[e].detach().cpu()
+            out[ep + "w1.weight"] = gu[: spec.intermediate_size].clone()
+            out[ep + "w3.weight"] = gu[spec.intermediate_size :].clone()
+            out[ep + "w2.weight"] = moe.down_w[e].detach().cpu().clone()
+    out["model.norm.weight"] = model.final_norm_w.detach().cpu()
+    out["lm_head.weight"] = model.lm_head.weight.detach().cpu()
+    out = {k: v.contiguous() for k, v in out.items()}
+    Path(path).mkdir(parents=True, exist_ok=True)
+    save_file(out, str(Path(path) / "model.safetensors"))

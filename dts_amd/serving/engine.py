@@ -80,15 +80,19 @@ class ServingEngine:
         else:
             self.model = build_model(self.spec, dtype=dtype, device=device)
             if weights_path:
-                if self.spec.arch != "llama":
+                if self.spec.arch == "llama":
+                    from dts_amd.models.weights import load_llama_safetensors
+
+                    load_llama_safetensors(self.model, weights_path)
+                elif self.spec.arch == "mixtral":
+                    from dts_amd.models.weights import load_mixtral_safetensors
+
+                    load_mixtral_safetensors(self.model, weights_path)
+                else:
                     raise NotImplementedError(
                         f"safetensors loading for arch {self.spec.arch!r} "
                         "— pass a pre-loaded model= instead"
                     )
-                from dts_amd.models.weights import load_llama_safetensors
-
-                n = load_llama_safetensors(self.model, weights_path)
-                logger.info("loaded %d tensors from %s", n, weights_path)
             else:
                 self.model.random_init(seed=weight_seed)
 

@@ -69,3 +69,52 @@ def test_engine_loads_safetensors_checkpoint(tmp_path):
         )
     )
     assert direct == loaded
+
+
+def test_mixtral_safetensors_roundtrip(tmp_path):
+    """Save a random-init Mixtral in HF naming, reload via the engine,
+    outputs token-identical to the source model."""
+    import torch
+
+    from dts_amd.llm.types import SamplingParams
+    from dts_amd.models.config import get_model_spec
+    from dts_amd.models.mixtral import MixtralModel
+    from dts_amd.models.weights import save_mixtral_safetensors
+    from dts_amd.serving import ServingEngine
+
+    src = MixtralModel(
+        get_model_spec("mixtral-tiny"), dtype=torch.float32, device="cpu"
+    )
+    src.random_init(seed=5)
+    save_mixtral_safetensors(src, str(tmp_path))
+
+    def run(eng):
+        f = eng.submit_tokens(
+            list(range(1, 30)), SamplingParams(max_tokens=6, temperature=0.0)
+        )
+        eng.run_until_idle()
+        out = f.result(timeout=10).token_ids
+        eng.stop()
+        return out
+
+    direct = run(
+        ServingEngine(
+            model_name="mixtral-tiny",
+            device="cpu",
+            dtype=torch.float32,
+            num_blocks=128,
+            block_size=8,
+            model=src,
+        )
+    )
+    loaded = run(
+        ServingEngine(
+            model_name="mixtral-tiny",
+            device="cpu",
+            dtype=torch.float32,
+            num_blocks=128,
+            block_size=8,
+            weights_path=str(tmp_path),
+        )
+    )
+    assert direct == loaded
