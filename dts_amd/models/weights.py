@@ -225,3 +225,83 @@ def save_mixtral_safetensors(model, path: str) -> None:
     out = {k: v.contiguous() for k, v in out.items()}
     Path(path).mkdir(parents=True, exist_ok=True)
     save_file(out, str(Path(path) / "model.safetensors"))
+
+
+def load_gpt2_safetensors(model, path: str) -> int:
+    """Load an HF-format GPT-2 checkpoint into a GPT2Model.
+
+    HF GPT-2 stores linear weights as transposed Conv1D ([in, out]);
+    they transpose into this framework's [out, in] layout.
+    """
+    from safetensors import safe_open
+
+    files = sorted(Path(path).glob("*.safetensors"))
+    if not files:
+        raise FileNotFoundError(f"no .safetensors under {path}")
+    tensors: dict = {}
+    for f in files:
+        with safe_open(str(f), framework="pt") as sf:
+            for name in sf.keys():
+                tensors[name] = sf.get_tensor(name)
+
+    def get(name: str) -> torch.Tensor:
+        # HF publishes both with and without the "transformer." prefix
+        for cand in (name, "transformer." + name):
+            if cand in tensors:
+                return tensors[cand].to(torch.float32)
+        raise KeyError(f"missing tensor {name}")
+
+    consumed = 0
+    with torch.no_grad():
+        model.wte.copy_(get("wte.weight").to(model.dtype))
+        model.wpe.copy_(get("wpe.weight").to(model.dtype))
+        consumed += 2
+        for i, layer in enumerate(model.layers):
+            p = f"h.{i}."
+            layer.ln1_w.copy_(get(p + "ln_1.weight").to(model.dtype))
+            layer.ln1_b.copy_(get(p + "ln_1.bias").to(model.dtype))
+            layer.ln2_w.copy_(get(p + "ln_2.weight").to(model.dtype))
+            layer.ln2_b.copy_(get(p + "ln_2.bias").to(model.dtype))
+            layer.qkv_w.copy_(get(p + "attn.c_attn.weight").t().to(model.dtype))
+            layer.qkv_b.copy_(get(p + "attn.c_attn.bias").to(model.dtype))
+            layer.o_w.copy_(get(p + "attn.c_proj.weight").t().to(model.dtype))
+            layer.o_b.copy_(get(p + "attn.c_proj.bias").to(model.dtype))
+            layer.fc_w.copy_(get(p + "mlp.c_fc.weight").t().to(model.dtype))
+            layer.fc_b.copy_(get(p + "mlp.c_fc.bias").to(model.dtype))
+            layer.proj_w.copy_(get(p + "mlp.c_proj.weight").t().to(model.dtype))
+            layer.proj_b.copy_(get(p + "mlp.c_proj.bias").to(model.dtype))
+            consumed += 12
+        model.lnf_w.copy_(get("ln_f.weight").to(model.dtype))
+        model.lnf_b.copy_(get("ln_f.bias").to(model.dtype))
+        consumed += 2
+    logger.info("loaded %d tensors from %s", consumed, path)
+    return consumed
+
+
+def save_gpt2_safetensors(model, path: str) -> None:
+    """Write a GPT2Model out in HF GPT-2 naming (Conv1D-transposed)."""
+    from safetensors.torch import save_file
+
+    out = {
+        "wte.weight": model.wte.detach().cpu(),
+        "wpe.weight": model.wpe.detach().cpu(),
+        "ln_f.weight": model.lnf_w.detach().cpu(),
+        "ln_f.bias": model.lnf_b.detach().cpu(),
+    }
+    for i, layer in enumerate(model.layers):
+        p = f"h.{i}."
+        out[p + "ln_1.weight"] = layer.ln1_w.detach().cpu()
+        out[p + "ln_1.bias"] = layer.ln1_b.detach().cpu()
+        out[p + "ln_2.weight"] = layer.ln2_w.detach().cpu()
+        out[p + "ln_2.bias"] = layer.ln2_b.detach().cpu()
+        out[p + "attn.c_attn.weight"] = layer.qkv_w.detach().cpu().t()
+        out[p + "attn.c_attn.bias"] = layer.qkv_b.detach().cpu()
+        out[p + "attn.c_proj.weight"] = layer.o_w.detach().cpu().t()
+        out[p + "attn.c_proj.bias"] = layer.o_b.detach().cpu()
+        out[p + "mlp.c_fc.weight"] = layer.fc_w.detach().cpu().t()
+        out[p + "mlp.c_fc.bias"] = layer.fc_b.detach().cpu()
+        out[p + "mlp.c_proj.weight"] = layer.proj_w.detach().cpu().t()
+        out[p + "mlp.c_proj.bias"] = layer.proj_b.detach().cpu()
+    out = {k: v.contiguous() for k, v in out.items()}
+    Path(path).mkdir(parents=True, exist_ok=True)
+    save_file(out, str(Path(path) / "model.safetensors"))

@@ -88,6 +88,10 @@ class ServingEngine:
                     from dts_amd.models.weights import load_mixtral_safetensors
 
                     load_mixtral_safetensors(self.model, weights_path)
+                elif self.spec.arch == "gpt2":
+                    from dts_amd.models.weights import load_gpt2_safetensors
+
+                    load_gpt2_safetensors(self.model, weights_path)
                 else:
                     raise NotImplementedError(
                         f"safetensors loading for arch {self.spec.arch!r} "

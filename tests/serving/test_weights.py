@@ -118,3 +118,48 @@ def test_mixtral_safetensors_roundtrip(tmp_path):
         )
     )
     assert direct == loaded
+
+
+def test_gpt2_safetensors_roundtrip(tmp_path):
+    import torch
+
+    from dts_amd.llm.types import SamplingParams
+    from dts_amd.models.config import get_model_spec
+    from dts_amd.models.gpt2 import GPT2Model
+    from dts_amd.models.weights import save_gpt2_safetensors
+    from dts_amd.serving import ServingEngine
+
+    src = GPT2Model(get_model_spec("gpt2-tiny"), dtype=torch.float32, device="cpu")
+    src.random_init(seed=11)
+    save_gpt2_safetensors(src, str(tmp_path))
+
+    def run(eng):
+        f = eng.submit_tokens(
+            list(range(1, 30)), SamplingParams(max_tokens=6, temperature=0.0)
+        )
+        eng.run_until_idle()
+        out = f.result(timeout=10).token_ids
+        eng.stop()
+        return out
+
+    direct = run(
+        ServingEngine(
+            model_name="gpt2-tiny",
+            device="cpu",
+            dtype=torch.float32,
+            num_blocks=128,
+            block_size=8,
+            model=src,
+        )
+    )
+    loaded = run(
+        ServingEngine(
+            model_name="gpt2-tiny",
+            device="cpu",
+            dtype=torch.float32,
+            num_blocks=128,
+            block_size=8,
+            weights_path=str(tmp_path),
+        )
+    )
+    assert direct == loaded
