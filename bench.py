@@ -100,7 +100,7 @@ async def run_search(llm, cfg, dp, rounds=1):
     scored = sum(
         1 for n in result.all_nodes if n.strategy is not None and n.stats.judge_scores
     )
-    return scored, result
+    return scored, engine.phase_times
 
 
 def main():
@@ -162,14 +162,18 @@ def main():
         cfg = type(cfg)(**{**cfg.__dict__,
                            "goal": f"{GOAL} (scenario {step_idx})",
                            "first_message": f"{FIRST_MESSAGE} (case {step_idx})"})
-        scored, _ = asyncio.run(run_search(llm, cfg, dp, rounds=args.rounds))
+        scored, phases = asyncio.run(run_search(llm, cfg, dp, rounds=args.rounds))
+        for k, v in phases.items():
+            phase_totals[k] = phase_totals.get(k, 0.0) + v
         return scored
 
     # warmup
+    phase_totals: dict = {}
     for w in range(args.warmup):
         one_step(10_000 + w)
 
     sync()
+    phase_totals.clear()  # keep only the timed steps' phase wall-clock
     t0 = time.perf_counter()
     total_scored = 0
     for s in range(args.steps):
@@ -221,6 +225,12 @@ def main():
                         "tok/turn, judge prompt ~order-10k tok)",
                         "parallelism": f"dp{n_gpus}",
                         "engine": stats,
+                        # rank-0 wall per search phase over the timed
+                        # steps: exposes the init(strategy)/expand/score
+                        # split for scaling analysis
+                        "search_phases": {
+                            k: round(v, 2) for k, v in phase_totals.items()
+                        },
                     },
                 },
             )

@@ -91,6 +91,9 @@ class DTSEngine:
         self._researcher = researcher
 
         self._tree: Optional[DialogueTree] = None
+        # wall-clock per search phase (init/expand/score), cumulative over
+        # rounds — dumped by bench.py to expose Amdahl pieces at scale
+        self.phase_times: dict = {"init_s": 0.0, "expand_s": 0.0, "score_s": 0.0}
         self._event_callback: Optional[Callable] = None
         self._research_report: Optional[str] = None
 
@@ -138,10 +141,14 @@ class DTSEngine:
         )
 
         self._emit("phase", {"phase": "initializing", "message": "Creating tree structure..."})
+        import time as _time
+
+        _t0 = _time.perf_counter()
         if resume_from is not None:
             tree = self._load_tree(resume_from)
         else:
             tree = await self._initialize_tree()
+        self.phase_times["init_s"] += _time.perf_counter() - _t0
         self._tree = tree
 
         total_pruned = 0
@@ -186,6 +193,7 @@ class DTSEngine:
                 async def generate_intents_fn(_history, _count):  # type: ignore[misc]
                     return [FIXED_INTENT]
 
+            _t0 = _time.perf_counter()
             expanded = await self._simulator.expand_nodes(
                 expandable,
                 turns=cfg.turns_per_branch,
@@ -193,6 +201,7 @@ class DTSEngine:
                 tree=tree,
                 generate_intents=generate_intents_fn,
             )
+            self.phase_times["expand_s"] += _time.perf_counter() - _t0
             log_phase("EXPAND", f"Completed {len(expanded)} expansions", indent=1)
 
             for node in expanded:
@@ -207,10 +216,12 @@ class DTSEngine:
                     "scoring_mode": cfg.scoring_mode,
                 },
             )
+            _t0 = _time.perf_counter()
             if cfg.scoring_mode == "comparative":
                 scores = await self._evaluator.evaluate_comparative(expanded)
             else:
                 scores = await self._evaluator.evaluate_absolute(expanded)
+            self.phase_times["score_s"] += _time.perf_counter() - _t0
 
             for node in expanded:
                 if node.id in scores:
