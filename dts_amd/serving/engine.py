@@ -361,7 +361,12 @@ class ServingEngine:
                 if self._stop:
                     return
             try:
-                self.step()
+                if not self.step():
+                    # no batch this step but work remains (e.g. stuck
+                    # drain): yield briefly instead of spinning hot
+                    import time as _t
+
+                    _t.sleep(0.001)
             except Exception as e:  # noqa: BLE001
                 logger.exception("engine step failed: %s", e)
                 with self._lock:
