@@ -75,6 +75,7 @@ class DTSEngine:
             on_event=self._emit_async,
             budget=config.budget,
             seed=config.seed,
+            reasoning_enabled=config.reasoning_enabled,
         )
         self._evaluator = TrajectoryEvaluator(
             llm=llm,

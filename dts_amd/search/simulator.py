@@ -64,10 +64,12 @@ class ConversationSimulator:
         on_event: Optional[Callable[[str, dict], Any]] = None,
         budget: Optional[GenerationBudget] = None,
         seed: Optional[int] = None,
+        reasoning_enabled: bool = False,
     ) -> None:
         self.llm = llm
         self.goal = goal
         self.model = model
+        self.reasoning_enabled = reasoning_enabled
         self.temperature = temperature
         self.budget = budget or GenerationBudget()
         self.seed = seed
@@ -272,9 +274,21 @@ class ConversationSimulator:
             strategy_tagline=strategy.tagline if strategy else "",
             strategy_description=strategy.description if strategy else "",
         )
+        max_tokens = self.budget.assistant
+        if self.reasoning_enabled:
+            # local semantics for the wire flag the reference drops
+            # (SURVEY.md §4.1.1): the actor may think in <think> spans,
+            # which LLM.complete strips from the returned content; the
+            # budget grows to cover the hidden reasoning
+            system += (
+                "\n\nBefore answering, reason privately inside "
+                "<think>...</think> tags; the user sees only what follows "
+                "the closing tag."
+            )
+            max_tokens *= 2
         messages = [Message.system(system)] + history + [Message.user(user)]
         return await self._call_with_retry(
-            messages, phase="assistant", max_tokens=self.budget.assistant
+            messages, phase="assistant", max_tokens=max_tokens
         )
 
     async def _call_with_retry(

@@ -149,3 +149,49 @@ class TestForking:
         result = run_async(sim._expand_with_intent(node, turns=1, first_intent=INTENT))
         assert result.messages[0].content == "original opening"
         assert result.messages[1].content == "assistant reply"
+
+
+class TestReasoningEnabled:
+    """reasoning_enabled (the wire flag the reference silently drops,
+    SURVEY.md §4.1.1): assistant turns get a think-span instruction and
+    <think> spans are stripped from the stored trajectory."""
+
+    def test_think_instruction_and_strip(self):
+        import asyncio
+
+        from dts_amd.llm import LLM
+        from dts_amd.llm.fake import FakeBackend
+        from dts_amd.search.simulator import ConversationSimulator
+        from dts_amd.search.types import DialogueNode, Strategy
+
+        backend = FakeBackend()
+        orig = backend._respond
+
+        def with_think(system, user, messages):
+            text = orig(system, user, messages)
+            if "[dts:assistant]" in system:
+                return f"<think>hidden chain of thought</think>{text}"
+            return text
+
+        backend._respond = with_think
+        llm = LLM(backend, default_model="fake")
+        sim = ConversationSimulator(
+            llm, goal="g", reasoning_enabled=True, seed=1
+        )
+        node = DialogueNode(
+            id="n1",
+            strategy=Strategy(tagline="t", description="d"),
+            messages=[],
+        )
+        from dts_amd.llm.types import Message
+
+        history = [Message.user("hi there can you help me?")]
+        ok = asyncio.run(sim._run_turn(node, history, 0, skip_user_simulation=True))
+        assert ok
+        assistant = history[-1]
+        assert assistant.role == "assistant"
+        assert "<think>" not in (assistant.content or "")
+        assert "hidden chain of thought" not in (assistant.content or "")
+        # the instruction reached the model
+        sys_prompts = [c["system"] for c in backend.calls if "[dts:assistant]" in c["system"]]
+        assert sys_prompts and "<think>" in sys_prompts[-1]
