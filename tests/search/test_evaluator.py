@@ -147,3 +147,41 @@ def test_comparative_pass_votes():
         passed=True,
     )
     assert s.passed
+
+
+class TestOversizedTrajectory:
+    """A trajectory whose flattened prompt exceeds the serving context
+    must degrade to the zero-score fallback, never crash the round."""
+
+    def test_zero_score_fallback_on_context_overflow(self):
+        import asyncio
+
+        import torch
+
+        from dts_amd.llm import LLM
+        from dts_amd.llm.types import Message
+        from dts_amd.search.evaluator import TrajectoryEvaluator
+        from dts_amd.search.types import DialogueNode, Strategy
+        from dts_amd.serving import LocalBackend, ServingEngine
+
+        eng = ServingEngine(
+            model_name="llama-tiny",
+            device="cpu",
+            dtype=torch.float32,
+            num_blocks=1024,
+            block_size=8,
+            weight_seed=1,
+        )
+        backend = LocalBackend.single(eng, name="llama-tiny")
+        llm = LLM(backend, default_model="llama-tiny")
+        ev = TrajectoryEvaluator(llm, goal="g", prune_threshold=5.0)
+        huge = "word " * 3000  # far beyond llama-tiny's max_position
+        node = DialogueNode(
+            id="big",
+            strategy=Strategy(tagline="t", description="d"),
+            messages=[Message.user(huge), Message.assistant(huge)],
+        )
+        scores = asyncio.run(ev.evaluate_absolute([node]))
+        assert scores["big"].aggregated_score == 0.0
+        assert not scores["big"].passed
+        backend.shutdown()
