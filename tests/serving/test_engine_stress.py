@@ -255,3 +255,46 @@ class TestBatchCompositionInvariance:
         eng2.stop()
 
         assert solo == cobatched
+
+
+class TestThreadedLoopConcurrency:
+    """The production mode: the engine's background step thread serving
+    concurrent submitters (the async search fans out from multiple
+    tasks). Every request must resolve; no lock-discipline races."""
+
+    def test_concurrent_submitters_against_running_loop(self):
+        import threading
+
+        eng = make_engine("1", num_blocks=1024)
+        eng.start()
+        results = []
+        errors = []
+
+        def submitter(tid):
+            rng = random.Random(tid)
+            try:
+                for i in range(8):
+                    prompt = [rng.randrange(1, 500) for _ in range(rng.randrange(6, 80))]
+                    f = eng.submit_tokens(
+                        prompt,
+                        SamplingParams(
+                            max_tokens=rng.randrange(1, 10),
+                            temperature=0.7,
+                            seed=tid * 100 + i,
+                        ),
+                    )
+                    r = f.result(timeout=30)
+                    results.append(r.completion_tokens)
+            except Exception as e:  # noqa: BLE001
+                errors.append(e)
+
+        threads = [threading.Thread(target=submitter, args=(t,)) for t in range(6)]
+        for t in threads:
+            t.start()
+        for t in threads:
+            t.join(timeout=60)
+            assert not t.is_alive(), "submitter thread hung"
+        assert not errors, errors
+        assert len(results) == 48
+        eng.stop()
+        assert not eng._futures
