@@ -188,6 +188,9 @@ class ServingEngine:
         self.t_post = 0.0
         self.eager_decode_steps = 0
         self.prefill_steps = 0
+        self.req_count = 0
+        self.req_latency_sum = 0.0
+        self.req_latency_max = 0.0
 
     # ------------------------------------------------------------------
     def submit_tokens(
@@ -218,6 +221,9 @@ class ServingEngine:
             seq.tokens.extend(forced)
             seq.output_tokens.extend(forced)
         fut: concurrent.futures.Future = concurrent.futures.Future()
+        import time as _time
+
+        seq.submit_ts = _time.perf_counter()  # type: ignore[attr-defined]
         with self._work:
             self._futures[seq.seq_id] = fut
             self.scheduler.add(seq)
@@ -318,6 +324,14 @@ class ServingEngine:
             self._finish(seq, "length")
 
     def _finish(self, seq: Sequence, reason: str) -> None:
+        import time as _time
+
+        ts = getattr(seq, "submit_ts", None)
+        if ts is not None:
+            lat = _time.perf_counter() - ts
+            self.req_count += 1
+            self.req_latency_sum += lat
+            self.req_latency_max = max(self.req_latency_max, lat)
         self.scheduler.finish(seq, reason)
         self.sampler.release(seq)
         fut = self._futures.pop(seq.seq_id, None)
@@ -413,6 +427,11 @@ class ServingEngine:
             "tokens_prefilled": self.tokens_prefilled,
             "native_scheduler": type(self.scheduler).__name__ == "NativeScheduler",
             "preemptions": getattr(self.scheduler, "preemptions", 0),
+            "requests": self.req_count,
+            "req_latency_mean_s": round(
+                self.req_latency_sum / max(1, self.req_count), 3
+            ),
+            "req_latency_max_s": round(self.req_latency_max, 3),
             "eager_decode_steps": self.eager_decode_steps,
             "prefill_steps": self.prefill_steps,
             "t_forward_graph_s": round(self.t_forward_graph, 2),
