@@ -151,11 +151,14 @@ class DistributedDTSEngine(DTSEngine):
     async def _initialize_tree(self) -> DialogueTree:
         if not self.dp.enabled:
             return await super()._initialize_tree()
+        import os as _os
+
         cfg = self.config
         if (
             self.dp.world >= 2
             and cfg.init_branches >= self.dp.world
             and cfg.init_branches % self.dp.world == 0
+            and _os.environ.get("DTS_SHARDED_INIT", "1") != "0"
         ):
             return await self._initialize_tree_sharded()
         return await self._initialize_tree_rank0()
