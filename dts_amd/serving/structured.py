@@ -174,12 +174,19 @@ class FormGuide:
             return None
         seg = self.segments[self._i]
         if isinstance(seg, Free):
+            # identical until the segment (or the stop-eligibility flag)
+            # changes — cache it: this runs every decode step per guided
+            # row (~0.13 ms/row/step measured uncached, profiles/)
+            key = (self._i, seg.stop is not None and self._free_count >= seg.min_tokens)
+            cached = getattr(self, "_allowed_cache", None)
+            if cached is not None and cached[0] == key:
+                return cached[1]
             allowed = list(seg.charset)
-            if seg.stop is not None and self._free_count >= seg.min_tokens:
-                if seg.stop not in allowed:
-                    allowed.append(seg.stop)
+            if key[1] and seg.stop not in allowed:
+                allowed.append(seg.stop)
             if self._b2t is not None:
                 allowed = [self._b2t[b] for b in allowed]
+            self._allowed_cache = (key, allowed)
             return allowed
         if isinstance(seg, Choice):
             nxt = set()
