@@ -19,6 +19,9 @@ class Sampler:
     def __init__(self, device: str = "cpu") -> None:
         self.device = device
         self._generators: dict = {}  # seq_id -> torch.Generator
+        # (temps, top_ps) device tensors cached per batch composition —
+        # stable across decode stretches; saves two H2D uploads per step
+        self._param_cache: dict = {}
 
     def _generator_for(self, seq) -> Optional[torch.Generator]:
         if seq.params.seed is None:
@@ -97,10 +100,19 @@ class Sampler:
             if len(free_rows) == S
             else logits[torch.as_tensor(free_rows, device=logits.device)]
         )
-        temps = torch.tensor(
-            [s.params.temperature for s in free_seqs], dtype=torch.float32
-        )
-        top_ps = torch.tensor([s.params.top_p for s in free_seqs], dtype=torch.float32)
+        key = tuple((s.params.temperature, s.params.top_p) for s in free_seqs)
+        cached = self._param_cache.get(key)
+        if cached is None:
+            temps = torch.tensor([t for t, _ in key], dtype=torch.float32)
+            top_ps = torch.tensor([p for _, p in key], dtype=torch.float32)
+            if logits.is_cuda:
+                temps = temps.to(logits.device)
+                top_ps = top_ps.to(logits.device)
+            if len(self._param_cache) > 256:
+                self._param_cache.clear()
+            self._param_cache[key] = (temps, top_ps)
+            cached = (temps, top_ps)
+        temps, top_ps = cached
 
         if logits.is_cuda:
             import random as _random
@@ -122,8 +134,8 @@ class Sampler:
             # gather for all guided rows; a single .cpu() syncs both
             toks_gpu = ops.top_p_sample(
                 free_logits.contiguous(),
-                temps.to(logits.device),
-                top_ps.to(logits.device),
+                temps,
+                top_ps,
                 seeds=seeds,
             )
             if guided:
