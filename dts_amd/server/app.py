@@ -26,12 +26,16 @@ from dts_amd.server import service as dts_service
 from dts_amd.server.schemas import SearchRequest
 from dts_amd.utils.logging import logger
 
+from dts_amd.utils.config import settings as _settings
+
+# /config defaults reflect the env-driven Settings (ref utils/config.py
+# pydantic-settings semantics: .env overrides, ref server.py:156-169)
 DEFAULT_CONFIG = {
-    "init_branches": 6,
-    "turns_per_branch": 5,
+    "init_branches": _settings.init_branches,
+    "turns_per_branch": _settings.turns_per_branch,
     "user_intents_per_branch": 3,
-    "scoring_mode": "comparative",
-    "prune_threshold": 6.5,
+    "scoring_mode": _settings.scoring_mode,
+    "prune_threshold": _settings.prune_threshold,
     "rounds": 1,
     "deep_research": False,
     "user_variability": False,
@@ -214,10 +218,10 @@ def main() -> None:
     import uvicorn
 
     p = argparse.ArgumentParser()
-    p.add_argument("--host", default="0.0.0.0")
-    p.add_argument("--port", type=int, default=8000)
-    p.add_argument("--model", default="llama-3-8b")
-    p.add_argument("--device", default=None)
+    p.add_argument("--host", default=_settings.server_host)
+    p.add_argument("--port", type=int, default=_settings.server_port)
+    p.add_argument("--model", default=_settings.model_name)
+    p.add_argument("--device", default=None if _settings.device == "auto" else _settings.device)
     p.add_argument("--tokenizer", default=None,
                    help="local HF tokenizers JSON (default: synthetic)")
     p.add_argument("--weights", default=None,
