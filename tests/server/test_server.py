@@ -104,10 +104,24 @@ class TestWebSocket:
                     }
                 )
             )
+            from dts_amd.server import schemas as sch
+
+            validators = {
+                "search_started": sch.SearchStartedData,
+                "phase": sch.PhaseData,
+                "strategy_generated": sch.StrategyGeneratedData,
+                "node_added": sch.NodeAddedData,
+                "node_updated": sch.NodeUpdatedData,
+                "round_started": sch.RoundStartedData,
+                "error": sch.ErrorData,
+            }
             events = []
             while True:
                 msg = json.loads(ws.receive_text())
                 events.append(msg["type"])
+                model = validators.get(msg["type"])
+                if model is not None:
+                    model(**msg["data"])  # event data matches the wire schema
                 if msg["type"] in ("complete", "error"):
                     final = msg
                     break
