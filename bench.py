@@ -236,6 +236,10 @@ def main():
             )
         )
     backend.shutdown()
+    if world > 1:
+        from dts_amd.parallel.dist import destroy
+
+        destroy()
 
 
 if __name__ == "__main__":
