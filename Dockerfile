@@ -7,7 +7,7 @@ WORKDIR /app
 COPY . /app
 
 ENV PYTORCH_ROCM_ARCH=gfx950
-RUN python -m dts_amd.ops.build
+RUN python -m dts_amd.ops.build && python -m dts_amd.core.build
 
 EXPOSE 8000
 HEALTHCHECK --interval=30s --timeout=5s CMD curl -sf http://localhost:8000/health || exit 1
