@@ -104,3 +104,46 @@ class TestHFTokenizer:
         d = json.loads(res.text)
         assert len(d["nodes"]) == 2
         eng.stop()
+
+
+def test_plain_bpe_without_chat_specials(tmp_path):
+    """A vocab with only <s>/</s> (no Llama-3 headers) uses the
+    plain-text role-header fallback and still generates."""
+    import torch
+    from tokenizers import Tokenizer, decoders, models, pre_tokenizers, trainers
+
+    from dts_amd.llm.types import Message, SamplingParams
+    from dts_amd.serving import ServingEngine
+    from dts_amd.serving.tokenizer import HFChatTemplate, HFTokenizer
+
+    tok = Tokenizer(models.BPE(unk_token=None))
+    tok.pre_tokenizer = pre_tokenizers.ByteLevel(add_prefix_space=False)
+    tok.decoder = decoders.ByteLevel()
+    trainer = trainers.BpeTrainer(
+        vocab_size=400,
+        special_tokens=["<s>", "</s>"],
+        initial_alphabet=pre_tokenizers.ByteLevel.alphabet(),
+    )
+    tok.train_from_iterator(["plain corpus text for a classic bpe " * 4], trainer)
+    path = tmp_path / "plain.json"
+    tok.save(str(path))
+
+    t = HFTokenizer(str(path))
+    assert t.bos_id is not None and t._hdr_start is None
+    tpl = HFChatTemplate(t)
+    ids = tpl.render([Message.user("hello")])
+    assert ids[0] == t.bos_id and t.eot_id in ids  # eot falls back to </s>
+
+    eng = ServingEngine(
+        model_name="llama-tiny",
+        device="cpu",
+        dtype=torch.float32,
+        num_blocks=256,
+        block_size=8,
+        weight_seed=1,
+        tokenizer_path=str(path),
+    )
+    fut = eng.submit_tokens(ids, SamplingParams(max_tokens=6, temperature=0.0))
+    eng.run_until_idle()
+    assert fut.result(timeout=10).completion_tokens >= 1
+    eng.stop()
