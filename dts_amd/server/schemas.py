@@ -92,3 +92,26 @@ class NodeUpdatedData(BaseModel):
 class RoundStartedData(BaseModel):
     round: int
     total_rounds: int
+
+
+class NodesPrunedData(BaseModel):
+    ids: list
+    reasons: dict
+
+
+class IntentGeneratedData(BaseModel):
+    strategy: str
+    index: int
+    total: int
+    label: str
+    emotional_tone: str
+    cognitive_stance: str
+
+
+class TokenUpdateData(BaseModel):
+    totals: dict
+
+
+class CompleteData(BaseModel):
+    summary: dict
+    exploration: dict

@@ -113,6 +113,9 @@ class TestWebSocket:
                 "node_added": sch.NodeAddedData,
                 "node_updated": sch.NodeUpdatedData,
                 "round_started": sch.RoundStartedData,
+                "nodes_pruned": sch.NodesPrunedData,
+                "intent_generated": sch.IntentGeneratedData,
+                "token_update": sch.TokenUpdateData,
                 "error": sch.ErrorData,
             }
             events = []
