@@ -113,5 +113,8 @@ class TokenUpdateData(BaseModel):
 
 
 class CompleteData(BaseModel):
-    summary: dict
+    best_node_id: Optional[str]
+    best_score: float
+    pruned_count: int
+    total_rounds: int
     exploration: dict

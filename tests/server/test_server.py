@@ -129,6 +129,7 @@ class TestWebSocket:
                     final = msg
                     break
             assert final["type"] == "complete"
+            sch.CompleteData(**final["data"])
             assert "search_started" in events
             assert "node_added" in events
             assert "node_updated" in events
