@@ -170,3 +170,28 @@ def test_metrics_endpoint():
     assert 'dts_engine_stat{model="llama-tiny",stat="free_blocks"}' in r.text
     assert 'stat="preemptions"' in r.text
     backend.shutdown()
+
+
+def test_frontend_js_syntax():
+    """The bundled visualizer's script must be valid JavaScript."""
+    import shutil
+    import subprocess
+    from pathlib import Path
+
+    node = shutil.which("node")
+    if node is None:
+        import pytest
+
+        pytest.skip("node not available")
+    html = (
+        Path(__file__).resolve().parents[2]
+        / "dts_amd"
+        / "server"
+        / "static"
+        / "index.html"
+    ).read_text()
+    script = html.split("<script>")[1].split("</script>")[0]
+    p = subprocess.run(
+        [node, "--check", "/dev/stdin"], input=script, capture_output=True, text=True
+    )
+    assert p.returncode == 0, p.stderr
