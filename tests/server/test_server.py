@@ -191,7 +191,10 @@ def test_frontend_js_syntax():
         / "index.html"
     ).read_text()
     script = html.split("<script>")[1].split("</script>")[0]
-    p = subprocess.run(
-        [node, "--check", "/dev/stdin"], input=script, capture_output=True, text=True
-    )
+    import tempfile
+
+    with tempfile.NamedTemporaryFile("w", suffix=".js", delete=False) as f:
+        f.write(script)
+        path = f.name
+    p = subprocess.run([node, "--check", path], capture_output=True, text=True)
     assert p.returncode == 0, p.stderr
