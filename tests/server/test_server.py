@@ -198,3 +198,29 @@ def test_frontend_js_syntax():
         path = f.name
     p = subprocess.run([node, "--check", path], capture_output=True, text=True)
     assert p.returncode == 0, p.stderr
+
+
+def test_frontend_event_handling_headless():
+    """Drive the visualizer's WS event handlers in node with a stub DOM:
+    node_added/node_updated/nodes_pruned must build the expected tree."""
+    import shutil
+    import subprocess
+    from pathlib import Path
+
+    node = shutil.which("node")
+    if node is None:
+        import pytest
+
+        pytest.skip("node not available")
+    root = Path(__file__).resolve().parents[2]
+    p = subprocess.run(
+        [
+            node,
+            str(Path(__file__).parent / "fe_harness.js"),
+            str(root / "dts_amd" / "server" / "static" / "index.html"),
+        ],
+        capture_output=True,
+        text=True,
+    )
+    assert p.returncode == 0, p.stderr
+    assert "frontend logic OK" in p.stdout
