@@ -49,6 +49,9 @@ def parse_args():
     p.add_argument("--rounds", type=int, default=1)
     p.add_argument("--scoring", type=str, default="comparative",
                    choices=["comparative", "absolute"])
+    p.add_argument("--intents", type=int, default=1,
+                   help=">1 enables user-intent forking (BASELINE config 3: "
+                        "user_variability=True, K intents per branch)")
     p.add_argument("--device", type=str, default=None)
     p.add_argument("--kv-frac", type=float, default=0.75)
     return p.parse_args()
@@ -73,8 +76,8 @@ def build_config(args, world, seed):
         first_message=FIRST_MESSAGE,
         init_branches=args.branches_per_gpu * world,
         turns_per_branch=args.turns,
-        user_intents_per_branch=1,
-        user_variability=False,
+        user_intents_per_branch=args.intents,
+        user_variability=args.intents > 1,
         scoring_mode=args.scoring,
         prune_threshold=6.5,
         min_survivors=1,
@@ -223,7 +226,8 @@ def main():
                         "global_batch": args.branches_per_gpu * world,
                         "seq_len": "chat-scale (user<=160, assistant<=256 "
                         "tok/turn, judge prompt ~order-10k tok)",
-                        "parallelism": f"dp{n_gpus}",
+                        "parallelism": f"dp{n_gpus}"
+                        + (f"+fork{args.intents}" if args.intents > 1 else ""),
                         "engine": stats,
                         # rank-0 wall per search phase over the timed
                         # steps: exposes the init(strategy)/expand/score
