@@ -129,8 +129,17 @@ class CoreScheduler {
   }
 
   std::vector<int64_t> take_stuck() {
+    // stuck seqs are terminal: the engine fails their futures and never
+    // calls finish/abort, so erase their state here or it leaks for the
+    // process lifetime
     auto out = stuck_;
     stuck_.clear();
+    for (int64_t sid : out) {
+      auto it = seqs_.find(sid);
+      if (it == seqs_.end()) continue;
+      release_blocks(it->second);
+      seqs_.erase(it);
+    }
     return out;
   }
 

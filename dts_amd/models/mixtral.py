@@ -153,6 +153,11 @@ class MixtralLayer(nn.Module):
 
 class MixtralModel(nn.Module):
     arch = "mixtral"
+    # MoE expert gather uses boolean-mask indexing (nonzero() → stream
+    # sync), which aborts hipGraph capture with
+    # hipErrorStreamCaptureUnsupported; decode runs eager until the
+    # capture-safe indirect grouped GEMV lands.
+    graph_capturable = False
 
     def __init__(
         self,

@@ -24,6 +24,22 @@ def _shard(t: torch.Tensor, dim: int, rank: int, size: int) -> torch.Tensor:
     return t.narrow(dim, rank * chunk, chunk)
 
 
+def _shard_kv(
+    t: torch.Tensor, num_kv_heads: int, head_dim: int, rank: int, size: int
+) -> torch.Tensor:
+    """Shard a k/v projection [nkv*hd, H] over TP ranks.
+
+    When size > num_kv_heads the model replicates KV heads
+    (LlamaAttention: num_kv_heads_local = max(1, nkv//size)); each rank
+    takes the single head it shares — plain narrowing would produce a
+    fraction of a head and fail to load.
+    """
+    if size <= num_kv_heads:
+        return _shard(t, 0, rank, size)
+    head = rank * num_kv_heads // size
+    return t.narrow(0, head * head_dim, head_dim)
+
+
 def load_llama_safetensors(model, path: str) -> int:
     """Load an HF-format Llama checkpoint directory into a LlamaModel.
 
@@ -60,8 +76,14 @@ def load_llama_safetensors(model, path: str) -> int:
                 get(p + "post_attention_layernorm.weight").to(model.dtype)
             )
             q = _shard(get(p + "self_attn.q_proj.weight"), 0, tp.rank, tp.size)
-            k = _shard(get(p + "self_attn.k_proj.weight"), 0, tp.rank, tp.size)
-            v = _shard(get(p + "self_attn.v_proj.weight"), 0, tp.rank, tp.size)
+            k = _shard_kv(
+                get(p + "self_attn.k_proj.weight"),
+                spec.num_kv_heads, spec.head_dim, tp.rank, tp.size,
+            )
+            v = _shard_kv(
+                get(p + "self_attn.v_proj.weight"),
+                spec.num_kv_heads, spec.head_dim, tp.rank, tp.size,
+            )
             layer.attn.qkv_proj.weight.copy_(
                 torch.cat([q, k, v], dim=0).to(model.dtype)
             )
@@ -134,6 +156,7 @@ def load_mixtral_safetensors(model, path: str) -> int:
     """
     from safetensors import safe_open
 
+    spec = model.spec
     tp = model.tp
     files = sorted(Path(path).glob("*.safetensors"))
     if not files:
@@ -160,8 +183,14 @@ def load_mixtral_safetensors(model, path: str) -> int:
                 get(p + "post_attention_layernorm.weight").to(model.dtype)
             )
             q = _shard(get(p + "self_attn.q_proj.weight"), 0, tp.rank, tp.size)
-            k = _shard(get(p + "self_attn.k_proj.weight"), 0, tp.rank, tp.size)
-            v = _shard(get(p + "self_attn.v_proj.weight"), 0, tp.rank, tp.size)
+            k = _shard_kv(
+                get(p + "self_attn.k_proj.weight"),
+                spec.num_kv_heads, spec.head_dim, tp.rank, tp.size,
+            )
+            v = _shard_kv(
+                get(p + "self_attn.v_proj.weight"),
+                spec.num_kv_heads, spec.head_dim, tp.rank, tp.size,
+            )
             layer.attn.qkv_proj.weight.copy_(
                 torch.cat([q, k, v], dim=0).to(model.dtype)
             )

@@ -159,7 +159,11 @@ class ServingEngine:
             self.template = ChatTemplate(self.tokenizer)
 
         self._graph_runner = None
-        if device.startswith("cuda") and os.environ.get("DTS_NO_HIPGRAPH") != "1":
+        if (
+            device.startswith("cuda")
+            and os.environ.get("DTS_NO_HIPGRAPH") != "1"
+            and getattr(self.model, "graph_capturable", True)
+        ):
             from dts_amd.serving.graph_runner import DecodeGraphRunner
 
             self._graph_runner = DecodeGraphRunner(
@@ -561,24 +565,49 @@ class LocalBackend:
         buf = bytearray()
         tok = engine.tokenizer
 
-        def emit(toks: list) -> str:
-            parts = []
-            for t in toks:
-                if t in stops:
-                    continue
-                if t < 256:
-                    buf.append(t)
-                    try:
-                        parts.append(buf.decode("utf-8"))
-                        buf.clear()
-                    except UnicodeDecodeError:
+        if isinstance(tok, HFTokenizer):
+            # incremental detokenization: decode the growing id list and
+            # emit the new suffix — per-token decode would mangle
+            # multi-token UTF-8 sequences, and ids < 256 are ordinary
+            # vocab entries for an HF tokenizer, not raw bytes
+            ids: list = []
+            emitted = [0]  # chars already yielded
+
+            def emit(toks: list) -> str:
+                ids.extend(t for t in toks if t not in stops)
+                text = tok.decode(ids)
+                # hold back a trailing replacement char: it usually marks
+                # an incomplete multibyte sequence that the next token
+                # completes
+                safe = len(text)
+                while safe > emitted[0] and text[safe - 1] == "�":
+                    safe -= 1
+                delta = text[emitted[0] : safe]
+                emitted[0] = safe
+                return delta
+
+        else:
+
+            def emit(toks: list) -> str:
+                # synthetic byte-level tokenizer: ids < 256 ARE raw UTF-8
+                # bytes; buffer until a whole codepoint decodes
+                parts = []
+                for t in toks:
+                    if t in stops:
                         continue
-                else:
-                    if buf:
-                        parts.append(buf.decode("utf-8", errors="replace"))
-                        buf.clear()
-                    parts.append(tok.decode([t]))
-            return "".join(parts)
+                    if t < 256:
+                        buf.append(t)
+                        try:
+                            parts.append(buf.decode("utf-8"))
+                            buf.clear()
+                        except UnicodeDecodeError:
+                            continue
+                    else:
+                        if buf:
+                            parts.append(buf.decode("utf-8", errors="replace"))
+                            buf.clear()
+                        parts.append(tok.decode([t]))
+                return "".join(parts)
 
         while True:
             get_task = asyncio.ensure_future(queue.get())
@@ -598,5 +627,9 @@ class LocalBackend:
                 if buf:  # trailing incomplete UTF-8 sequence
                     yield buf.decode("utf-8", errors="replace")
                     buf.clear()
+                if isinstance(tok, HFTokenizer) and ids:
+                    tail = tok.decode(ids)[emitted[0] :]
+                    if tail:
+                        yield tail
                 await afut  # surface exceptions
                 return

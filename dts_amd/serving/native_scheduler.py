@@ -77,7 +77,9 @@ class NativeScheduler:
     def schedule(self) -> Optional[ForwardBatch]:
         d = self.core.schedule()
         for sid in self.core.take_stuck():
-            seq = self._seqs.get(sid)
+            # stuck = terminal; drop our entry too or it leaks (the engine
+            # fails the future without calling finish/abort)
+            seq = self._seqs.pop(sid, None)
             if seq is not None:
                 self.stuck.append(seq)
         if not d:

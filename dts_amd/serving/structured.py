@@ -60,13 +60,15 @@ def JsonString(max_tokens: int = 64) -> list:
     return [Fixed('"'), Free(max_tokens=max_tokens, stop=ord('"'))]
 
 
+_SCORE_CHOICES = [f"{a}.{b}" for a in range(10) for b in range(10)] + ["10.0"]
+
+
 def Score() -> list:
-    """A d.d score in [0.0-9.9]: three constrained byte samples."""
-    return [
-        Free(max_tokens=1, charset=tuple(_DIGIT_BYTES), stop=None, min_tokens=1),
-        Fixed("."),
-        Free(max_tokens=1, charset=tuple(_DIGIT_BYTES), stop=None, min_tokens=1),
-    ]
+    """A score in [0.0, 10.0] with one decimal: byte-trie Choice over
+    '0.0'..'9.9' plus '10.0' — the full reference rubric range
+    (ref prompts.py:246-257 criteria 0-1, totals 0-10; a d.d-only form
+    made 10.0 unreachable)."""
+    return [Choice(_SCORE_CHOICES)]
 
 
 class FormGuide:
@@ -247,6 +249,18 @@ class FormGuide:
             if not self._choice_state:  # should be impossible under masking
                 self._done = True
                 return []
+            if len(self._choice_state) == 1:
+                # single candidate left: its tail is fully determined —
+                # force it as a prefill chunk instead of decoding byte by
+                # byte (e.g. '1.0'/'10.0' tails in Score choices)
+                only = self._choice_state[0]
+                tail = only.encode("utf-8")[self._choice_pos :]
+                self.last_choice = only
+                self._i += 1
+                forced = [
+                    self._b2t[b] if self._b2t is not None else b for b in tail
+                ]
+                return forced + self._collect_forced()
             return []
         raise AssertionError("on_token on Fixed segment")
 
@@ -324,8 +338,9 @@ def absolute_judge_form(tok) -> FormGuide:
         if i:
             segs.append(Fixed(", "))
         segs += [
-            Fixed(f'"{name}": {{"score": 0.'),
-            Free(max_tokens=1, charset=tuple(_DIGIT_BYTES), stop=None, min_tokens=1),
+            # criterion scores span the full 0.0-1.0 rubric (1.0 included)
+            Fixed(f'"{name}": {{"score": '),
+            Choice([f"0.{d}" for d in range(10)] + ["1.0"]),
             Fixed(', "rationale": '),
             JsonString(48),
             Fixed("}"),

@@ -1,0 +1,191 @@
+"""Regression tests for the round-1 advisor findings.
+
+One test per finding: Mixtral graph-capture flag, HF streaming
+detokenization, native stuck-sequence leak, KV-head-replicated weight
+sharding, and the full 0-1 / 0-10 judge score range.
+"""
+
+import asyncio
+
+import pytest
+import torch
+
+from dts_amd.serving.structured import (
+    Choice,
+    FormGuide,
+    Score,
+    absolute_judge_form,
+)
+from dts_amd.serving.tokenizer import SyntheticTokenizer
+
+
+@pytest.fixture(scope="module")
+def tok():
+    return SyntheticTokenizer(1024)
+
+
+class TestScoreRange:
+    """ADVICE low: 1.0 criteria and 10.0 totals must be reachable."""
+
+    def test_score_reaches_ten(self, tok):
+        g = FormGuide(tok, Score())
+        assert not g.initial_forced()
+        allowed = g.allowed_tokens()
+        assert ord("1") in allowed and ord("9") in allowed
+        forced = g.on_token(ord("1"))
+        # '1' narrows to {1.0..1.9, 10.0}; next byte may be '.' or '0'
+        assert forced == []
+        allowed = g.allowed_tokens()
+        assert ord(".") in allowed and ord("0") in allowed
+        forced = g.on_token(ord("0"))
+        # only '10.0' remains: the '.0' tail is forced, not decoded
+        assert forced == [ord("."), ord("0")]
+        assert g.done()
+
+    def test_score_plain_values_still_work(self, tok):
+        g = FormGuide(tok, Score())
+        g.initial_forced()
+        g.on_token(ord("6"))
+        forced = g.on_token(ord("."))
+        assert forced == []
+        g.on_token(ord("5"))
+        assert g.done()
+
+    def test_criterion_reaches_one(self, tok):
+        g = absolute_judge_form(tok)
+        choice_segs = [
+            s
+            for s in g.segments
+            if isinstance(s, Choice) and "1.0" in s.choices and "0.9" in s.choices
+        ]
+        # 10 criterion choices (0.0-1.0) + the total Score choice
+        assert len(choice_segs) >= 10
+
+
+class TestKVShardReplication:
+    """ADVICE low: loading with tp.size > num_kv_heads must replicate
+    whole KV heads, not narrow to head fractions."""
+
+    def test_shard_kv_replicates(self):
+        from dts_amd.models.weights import _shard_kv
+
+        nkv, hd = 2, 4
+        k = torch.arange(nkv * hd * 3, dtype=torch.float32).reshape(nkv * hd, 3)
+        shards = [_shard_kv(k, nkv, hd, r, 4) for r in range(4)]
+        for s in shards:
+            assert s.shape == (hd, 3)  # one whole head each
+        assert torch.equal(shards[0], k[:hd])
+        assert torch.equal(shards[1], k[:hd])
+        assert torch.equal(shards[2], k[hd:])
+        assert torch.equal(shards[3], k[hd:])
+
+    def test_shard_kv_normal_path(self):
+        from dts_amd.models.weights import _shard_kv
+
+        nkv, hd = 4, 2
+        k = torch.arange(nkv * hd * 3, dtype=torch.float32).reshape(nkv * hd, 3)
+        s = _shard_kv(k, nkv, hd, 1, 2)
+        assert torch.equal(s, k[4:8])
+
+
+class TestStuckLeak:
+    """ADVICE low: stuck (never-fitting) requests must not leak their
+    Seq entries in the native core or the adapter."""
+
+    def test_native_stuck_erased(self):
+        pytest.importorskip("dts_amd.core", reason="native core not built")
+        from dts_amd.core import load_core
+
+        if load_core() is None:
+            pytest.skip("native core not built")
+        from dts_amd.llm.types import SamplingParams
+        from dts_amd.serving.native_scheduler import NativeScheduler
+        from dts_amd.serving.sequence import Sequence
+
+        ns = NativeScheduler(num_blocks=4, block_size=4, max_batch_tokens=64)
+        big = Sequence(tokens=list(range(100)), params=SamplingParams())
+        ns.add(big)
+        assert ns.schedule() is None
+        assert [s.seq_id for s in ns.stuck] == [big.seq_id]
+        # terminal: no leak in the adapter map, blocks all free again
+        assert big.seq_id not in ns._seqs
+        assert ns.num_free() == 4
+        # a follow-up request still schedules fine
+        ok = Sequence(tokens=[1, 2, 3], params=SamplingParams(max_tokens=2))
+        ns.add(ok)
+        assert ns.schedule() is not None
+
+
+class TestMixtralGraphFlag:
+    """ADVICE high: MoE masked-gather breaks hipGraph capture — the
+    engine must not wrap Mixtral in the DecodeGraphRunner."""
+
+    def test_flag(self):
+        from dts_amd.models.mixtral import MixtralModel
+
+        assert MixtralModel.graph_capturable is False
+
+    def test_llama_still_capturable(self):
+        from dts_amd.models.llama import LlamaModel
+
+        assert getattr(LlamaModel, "graph_capturable", True) is True
+
+
+class TestHFStreaming:
+    """ADVICE medium: streamed deltas with an HF tokenizer must use
+    incremental detokenization, not the byte-buffer path."""
+
+    def test_stream_matches_complete_hf(self, tmp_path):
+        from tokenizers import Tokenizer, decoders, models, pre_tokenizers, trainers
+
+        tkz = Tokenizer(models.BPE(unk_token=None))
+        tkz.pre_tokenizer = pre_tokenizers.ByteLevel(add_prefix_space=False)
+        tkz.decoder = decoders.ByteLevel()
+        trainer = trainers.BpeTrainer(
+            vocab_size=480,
+            special_tokens=[
+                "<|begin_of_text|>",
+                "<|end_of_text|>",
+                "<|eot_id|>",
+                "<|start_header_id|>",
+                "<|end_header_id|>",
+            ],
+            initial_alphabet=pre_tokenizers.ByteLevel.alphabet(),
+        )
+        tkz.train_from_iterator(["hello world, how are you today? " * 8], trainer)
+        path = tmp_path / "bpe.json"
+        tkz.save(str(path))
+
+        from dts_amd.llm import LLM
+        from dts_amd.llm.types import Message
+        from dts_amd.serving import LocalBackend, ServingEngine
+
+        eng = ServingEngine(
+            model_name="llama-tiny",
+            device="cpu",
+            dtype=torch.float32,
+            num_blocks=512,
+            block_size=8,
+            weight_seed=3,
+            tokenizer_path=str(path),
+        )
+        backend = LocalBackend.single(eng, name="m")
+        try:
+            msgs = [Message.user("hello there")]
+
+            async def main():
+                llm = LLM(backend, default_model="m")
+                chunks = []
+                async for d in llm.stream(msgs, max_tokens=16, seed=7):
+                    chunks.append(d)
+                whole = await llm.complete(msgs, max_tokens=16, seed=7)
+                return "".join(chunks), whole.message.content
+
+            streamed, whole = asyncio.run(main())
+            # the old byte-buffer path treated HF ids < 256 as raw UTF-8
+            # bytes and mangled the text; incremental detokenization must
+            # reproduce the complete() text exactly (random-init output
+            # may legitimately contain replacement chars in BOTH)
+            assert streamed.strip() == whole.strip()
+        finally:
+            backend.shutdown()
