@@ -41,6 +41,10 @@ struct Block {
 
 enum class Status : int { WAITING = 0, RUNNING = 1, FINISHED = 2, ABORTED = 3 };
 
+static inline uint64_t bigram_key(int32_t a, int32_t b) {
+  return ((uint64_t)(uint32_t)a << 32) | (uint32_t)b;
+}
+
 struct Seq {
   int64_t id;
   std::vector<int32_t> tokens;
@@ -54,16 +58,53 @@ struct Seq {
   int64_t arrival = 0;
   Status status = Status::WAITING;
   uint64_t prompt_key = 0;
+  // --- prompt-lookup speculative decoding (dts_amd/serving/spec.py is
+  // the Python reference twin): bigram -> (latest, prev) continuation
+  // positions over this sequence's own tokens
+  std::unordered_map<uint64_t, std::pair<int, int>> ngram;
+  int ngram_n = 0;
+  bool allow_spec = true;
+  std::vector<int32_t> draft;  // draft rows scheduled this step
+
+  void index_tokens() {
+    const int L = (int)tokens.size();
+    for (int i = std::max(1, ngram_n); i < L; ++i) {
+      uint64_t key = bigram_key(tokens[i - 1], tokens[i]);
+      auto it = ngram.find(key);
+      if (it == ngram.end())
+        ngram.emplace(key, std::make_pair(i + 1, 0));
+      else {
+        it->second.second = it->second.first;
+        it->second.first = i + 1;
+      }
+    }
+    ngram_n = L;
+  }
+
+  std::vector<int32_t> propose(int max_k) const {
+    const int L = (int)tokens.size();
+    if (L < 3 || max_k <= 0) return {};
+    auto it = ngram.find(bigram_key(tokens[L - 2], tokens[L - 1]));
+    if (it == ngram.end()) return {};
+    // the latest continuation is the tail itself (self-match) — draft
+    // from the previous occurrence
+    int cont = (it->second.first >= L) ? it->second.second : it->second.first;
+    if (cont <= 0 || cont >= L) return {};
+    int k = std::min(max_k, L - cont);
+    return std::vector<int32_t>(tokens.begin() + cont, tokens.begin() + cont + k);
+  }
 };
 
 class CoreScheduler {
  public:
   CoreScheduler(int num_blocks, int block_size, int64_t max_batch_tokens,
-                int max_running)
+                int max_running, int spec_k = 0, int max_spec_rows = 16)
       : num_blocks_(num_blocks),
         block_size_(block_size),
         max_batch_tokens_(max_batch_tokens),
         max_running_(max_running),
+        spec_k_(spec_k),
+        max_spec_rows_(max_spec_rows),
         blocks_(num_blocks) {
     free_ids_.reserve(num_blocks);
     for (int i = num_blocks - 1; i >= 0; --i) free_ids_.push_back(i);
@@ -77,24 +118,35 @@ class CoreScheduler {
   int num_free() const { return (int)(free_ids_.size() + evictable_.size()); }
 
   // ---- sequence lifecycle ------------------------------------------------
-  void add(int64_t seq_id, const std::vector<int32_t>& tokens) {
+  void add(int64_t seq_id, const std::vector<int32_t>& tokens,
+           bool allow_spec = true) {
     Seq s;
     s.id = seq_id;
     s.tokens = tokens;
     s.num_prompt = (int)tokens.size();
     s.arrival = arrival_++;
+    s.allow_spec = allow_spec;
     s.prompt_key = chunk_hash(0xABCD, tokens.data(), (int)tokens.size());
+    if (spec_k_ > 0 && s.allow_spec) s.index_tokens();
     seqs_.emplace(seq_id, std::move(s));
     waiting_.push_back(seq_id);
   }
 
   void append_token(int64_t seq_id, int32_t tok) {
-    seqs_.at(seq_id).tokens.push_back(tok);
+    Seq& s = seqs_.at(seq_id);
+    s.tokens.push_back(tok);
+    if (spec_k_ > 0 && s.allow_spec) s.index_tokens();
   }
 
   void extend_tokens(int64_t seq_id, const std::vector<int32_t>& toks) {
-    auto& t = seqs_.at(seq_id).tokens;
-    t.insert(t.end(), toks.begin(), toks.end());
+    Seq& s = seqs_.at(seq_id);
+    s.tokens.insert(s.tokens.end(), toks.begin(), toks.end());
+    if (spec_k_ > 0 && s.allow_spec) s.index_tokens();
+  }
+
+  void set_sched_chunk(int64_t seq_id, int chunk) {
+    auto it = seqs_.find(seq_id);
+    if (it != seqs_.end()) it->second.sched_chunk = chunk;
   }
 
   int64_t num_tokens(int64_t seq_id) const {
@@ -149,6 +201,7 @@ class CoreScheduler {
     admit();
     std::vector<Seq*> prefills, decodes;
     int64_t budget = max_batch_tokens_;
+    int spec_rows = 0;
     // iterate a snapshot: preempt_youngest erases victims from running_
     // mid-loop (mutating a range-for'd container is UB), and a victim
     // visited later must be skipped — growing its released block table
@@ -176,17 +229,32 @@ class CoreScheduler {
       }
       int chunk = std::min<int64_t>(remaining, budget);
       if (chunk <= 0) continue;
-      budget -= chunk;
-      s.sched_chunk = chunk;
       // mark immediately: a seq already placed in this step's batch must
       // not become a preemption victim for a later seq in this loop
       // (running order != arrival order after re-admission), or
       // build_batch would index its released block table
       s.in_flight = true;
-      if (chunk == 1)
+      if (remaining == 1) {
+        // true decode row (the tail token) — maybe add draft rows.
+        // (a budget-starved chunk==1 MID-prompt is a prefill chunk;
+        // classifying it as decode would sample a bogus mid-prompt token)
+        s.draft.clear();
+        if (spec_k_ > 0 && s.allow_spec &&
+            spec_rows + 1 + spec_k_ <= max_spec_rows_ && budget > 1) {
+          s.draft = s.propose((int)std::min<int64_t>(spec_k_, budget - 1));
+          if (!s.draft.empty() &&
+              !ensure_blocks(s, (int)s.tokens.size() + (int)s.draft.size()))
+            s.draft.clear();  // never preempt for draft rows
+        }
+        spec_rows += 1 + (int)s.draft.size();
+        budget -= 1 + (int64_t)s.draft.size();
+        s.sched_chunk = 1;
         decodes.push_back(&s);
-      else
+      } else {
+        budget -= chunk;
+        s.sched_chunk = chunk;
         prefills.push_back(&s);
+      }
       if (budget <= 0) break;
     }
     if (prefills.empty() && decodes.empty()) return py::dict();
@@ -397,7 +465,9 @@ class CoreScheduler {
   py::dict build_batch(std::vector<Seq*>& prefills, std::vector<Seq*>& decodes) {
     int64_t T = 0;
     for (Seq* s : prefills) T += s->sched_chunk;
-    T += (int64_t)decodes.size();
+    int64_t D_rows = 0;
+    for (Seq* s : decodes) D_rows += 1 + (int64_t)s->draft.size();
+    T += D_rows;
 
     auto opts_i64 = torch::TensorOptions().dtype(torch::kInt64);
     auto opts_i32 = torch::TensorOptions().dtype(torch::kInt32);
@@ -408,7 +478,9 @@ class CoreScheduler {
     auto* pos = positions.data_ptr<int64_t>();
     auto* slt = slots.data_ptr<int64_t>();
 
-    std::vector<int64_t> sample_idx, sampled_ids, scheduled_ids;
+    std::vector<int64_t> sample_idx, sampled_ids, scheduled_ids, sample_pos;
+    py::list row_groups;  // (seq_id, n_rows) in sampled-row order
+    py::list spec;        // (seq_id, [draft tokens]) for spec'd seqs
     int64_t cursor = 0;
 
     const int P = (int)prefills.size();
@@ -439,31 +511,49 @@ class CoreScheduler {
       if (end == (int)s->tokens.size()) {
         sample_idx.push_back(cursor - 1);
         sampled_ids.push_back(s->id);
+        sample_pos.push_back((int64_t)s->tokens.size());
+        row_groups.append(py::make_tuple(s->id, 1));
       }
       s->in_flight = true;
       scheduled_ids.push_back(s->id);
     }
     int64_t num_prefill_tokens = cursor;
 
-    const int D = (int)decodes.size();
     int max_dc_blocks = 1;
     for (Seq* s : decodes)
       max_dc_blocks = std::max(max_dc_blocks, (int)s->block_table.size());
-    torch::Tensor dc_tables = torch::zeros({D, max_dc_blocks}, opts_i32);
-    torch::Tensor dc_kv = torch::empty({D}, opts_i32);
-    for (int i = 0; i < D; ++i) {
-      Seq* s = decodes[i];
-      int p = s->num_computed;
-      tid[cursor] = s->tokens[p];
-      pos[cursor] = p;
-      slt[cursor] =
-          (int64_t)s->block_table[p / block_size_] * block_size_ + p % block_size_;
-      auto* rp = dc_tables[i].data_ptr<int32_t>();
-      for (size_t b = 0; b < s->block_table.size(); ++b) rp[b] = s->block_table[b];
-      dc_kv.data_ptr<int32_t>()[i] = p + 1;
-      sample_idx.push_back(cursor);
-      sampled_ids.push_back(s->id);
-      cursor++;
+    torch::Tensor dc_tables = torch::zeros({D_rows, max_dc_blocks}, opts_i32);
+    torch::Tensor dc_kv = torch::empty({D_rows}, opts_i32);
+    int64_t row = 0;
+    for (Seq* s : decodes) {
+      const int p = s->num_computed;
+      const int k = (int)s->draft.size();
+      // row 0 = the real tail token; rows 1..k = draft candidates, each
+      // an independent 1-token decode row with its own position/kv_len
+      // (rope_kv_append writes all rows' KV before attention runs)
+      for (int j = 0; j <= k; ++j) {
+        const int q = p + j;
+        tid[cursor] = (j == 0) ? s->tokens[p] : s->draft[j - 1];
+        pos[cursor] = q;
+        slt[cursor] =
+            (int64_t)s->block_table[q / block_size_] * block_size_ +
+            q % block_size_;
+        auto* rp = dc_tables[row].data_ptr<int32_t>();
+        for (size_t b = 0; b < s->block_table.size(); ++b)
+          rp[b] = s->block_table[b];
+        dc_kv.data_ptr<int32_t>()[row] = q + 1;
+        sample_idx.push_back(cursor);
+        sampled_ids.push_back(s->id);
+        sample_pos.push_back((int64_t)s->tokens.size() + j);
+        cursor++;
+        row++;
+      }
+      row_groups.append(py::make_tuple(s->id, k + 1));
+      if (k) {
+        py::list dl;
+        for (int32_t d : s->draft) dl.append((int64_t)d);
+        spec.append(py::make_tuple(s->id, dl));
+      }
       s->in_flight = true;
       scheduled_ids.push_back(s->id);
     }
@@ -480,8 +570,8 @@ class CoreScheduler {
       out["prefill_block_tables"] = pf_tables;
       out["prefill_kv_lens"] = pf_kv;
     }
-    out["num_decode_seqs"] = D;
-    if (D) {
+    out["num_decode_seqs"] = D_rows;  // decode ROWS (>= seqs with spec on)
+    if (D_rows) {
       out["decode_block_tables"] = dc_tables;
       out["decode_kv_lens"] = dc_kv;
     }
@@ -489,6 +579,9 @@ class CoreScheduler {
         torch::tensor(sample_idx, opts_i64);
     out["sampled_ids"] = sampled_ids;
     out["scheduled_ids"] = scheduled_ids;
+    out["sample_pos"] = sample_pos;
+    out["row_groups"] = row_groups;
+    out["spec"] = spec;
     return out;
   }
 
@@ -496,6 +589,8 @@ class CoreScheduler {
   int block_size_;
   int64_t max_batch_tokens_;
   int max_running_;
+  int spec_k_;
+  int max_spec_rows_;
   std::vector<Block> blocks_;
   std::vector<int> free_ids_;
   std::list<int> evictable_;
@@ -513,10 +608,13 @@ class CoreScheduler {
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   py::class_<CoreScheduler>(m, "CoreScheduler")
-      .def(py::init<int, int, int64_t, int>(), py::arg("num_blocks"),
+      .def(py::init<int, int, int64_t, int, int, int>(), py::arg("num_blocks"),
            py::arg("block_size"), py::arg("max_batch_tokens"),
-           py::arg("max_running"))
-      .def("add", &CoreScheduler::add)
+           py::arg("max_running"), py::arg("spec_k") = 0,
+           py::arg("max_spec_rows") = 16)
+      .def("add", &CoreScheduler::add, py::arg("seq_id"), py::arg("tokens"),
+           py::arg("allow_spec") = true)
+      .def("set_sched_chunk", &CoreScheduler::set_sched_chunk)
       .def("append_token", &CoreScheduler::append_token)
       .def("extend_tokens", &CoreScheduler::extend_tokens)
       .def("num_tokens", &CoreScheduler::num_tokens)

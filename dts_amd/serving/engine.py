@@ -62,6 +62,7 @@ class ServingEngine:
         block_size: int = 16,
         max_batch_tokens: int = 8192,
         max_running: int = 256,
+        spec_k: Optional[int] = None,
         weight_seed: int = 0,
         model: Optional[object] = None,
         tokenizer_path: Optional[str] = None,
@@ -130,6 +131,12 @@ class ServingEngine:
         # last block reserved as the hipGraph pad-row scratch target
         self.block_manager = BlockManager(max(1, num_blocks - 1), block_size)
         self._scratch_block = num_blocks - 1
+        # prompt-lookup speculative decoding (serving/spec.py): k draft
+        # rows per decode seq; exact-match verification keeps the output
+        # stream identical, so it is on by default (DTS_SPEC_K=0 disables)
+        if spec_k is None:
+            spec_k = int(os.environ.get("DTS_SPEC_K", "4"))
+        self.spec_k = spec_k
         use_native = os.environ.get("DTS_NATIVE_CORE", "1") != "0"
         self.scheduler = None
         if use_native:
@@ -137,12 +144,18 @@ class ServingEngine:
                 from dts_amd.serving.native_scheduler import NativeScheduler
 
                 self.scheduler = NativeScheduler(
-                    max(1, num_blocks - 1), block_size, max_batch_tokens, max_running
+                    max(1, num_blocks - 1),
+                    block_size,
+                    max_batch_tokens,
+                    max_running,
+                    spec_k=spec_k,
                 )
             except Exception as e:  # noqa: BLE001
                 logger.warning("native core unavailable (%s); Python scheduler", e)
         if self.scheduler is None:
-            self.scheduler = Scheduler(self.block_manager, max_batch_tokens, max_running)
+            self.scheduler = Scheduler(
+                self.block_manager, max_batch_tokens, max_running, spec_k=spec_k
+            )
         self.sampler = Sampler(device)
         # a real tokenizer JSON (HF `tokenizers` format) when given —
         # its vocab must fit the model's embedding width; else the
@@ -192,6 +205,8 @@ class ServingEngine:
         self.t_post = 0.0
         self.eager_decode_steps = 0
         self.prefill_steps = 0
+        self.spec_draft_tokens = 0
+        self.spec_accepted_tokens = 0
         self.req_count = 0
         self.req_latency_sum = 0.0
         self.req_latency_max = 0.0
@@ -280,17 +295,62 @@ class ServingEngine:
         # sync happens inside sample(), so t_sample absorbs the GPU
         # execution wait of the whole step. Per-kernel truth lives in
         # rocprof (profiles/), not these host timers.
-        tokens = self.sampler.sample(logits, sampled_seqs) if sampled_seqs else []
+        tokens = (
+            self.sampler.sample(
+                logits, sampled_seqs, positions=getattr(batch, "_sample_pos", None)
+            )
+            if sampled_seqs
+            else []
+        )
         t2 = _time.perf_counter()
         self.t_sample += t2 - t1
         with self._lock:
-            self.scheduler.advance_computed(batch)
             self.tokens_prefilled += batch.num_prefill_tokens
-            self.tokens_sampled += len(tokens)
-            for seq, tok in zip(sampled_seqs, tokens):
-                self._handle_sampled(seq, tok)
+            row_groups = getattr(batch, "_row_groups", None)
+            if row_groups is None:
+                # one sampled row per seq (no speculation)
+                self.tokens_sampled += len(tokens)
+                for seq, tok in zip(sampled_seqs, tokens):
+                    self._handle_sampled(seq, tok)
+            else:
+                spec_drafts = getattr(batch, "_spec_drafts", None) or {}
+                r = 0
+                for seq, n_rows in row_groups:
+                    if n_rows == 1:
+                        self.tokens_sampled += 1
+                        self._handle_sampled(seq, tokens[r])
+                    else:
+                        draft = spec_drafts[seq.seq_id]
+                        emitted = self._verify_and_emit(
+                            seq, tokens[r : r + n_rows], draft
+                        )
+                        self.spec_draft_tokens += n_rows - 1
+                        self.spec_accepted_tokens += emitted - 1
+                        self.scheduler.set_accepted(seq, emitted)
+                    r += n_rows
+            # advance AFTER appends: speculative chunks must register
+            # block content that exists, and finished seqs are skipped
+            self.scheduler.advance_computed(batch)
         self.t_post += _time.perf_counter() - t2
         return True
+
+    def _verify_and_emit(self, seq: Sequence, toks: list, draft: list) -> int:
+        """Exact-match speculative verification (serving/spec.py): emit
+        sampled row j while sampled j-1 matched draft j-1 — every emitted
+        token is the model's own sample conditioned on accepted history,
+        so the stream equals non-speculative decoding token for token."""
+        from dts_amd.serving.sequence import SeqStatus
+
+        emitted = 0
+        for j, tok in enumerate(toks):
+            if j > 0 and toks[j - 1] != draft[j - 1]:
+                break
+            self.tokens_sampled += 1
+            self._handle_sampled(seq, tok)
+            emitted += 1
+            if seq.status != SeqStatus.RUNNING:
+                break  # finished (stop/length) mid-group: drop the rest
+        return emitted
 
     def _handle_sampled(self, seq: Sequence, tok: int) -> None:
         params = seq.params
@@ -438,6 +498,8 @@ class ServingEngine:
             "req_latency_max_s": round(self.req_latency_max, 3),
             "eager_decode_steps": self.eager_decode_steps,
             "prefill_steps": self.prefill_steps,
+            "spec_draft_tokens": self.spec_draft_tokens,
+            "spec_accepted_tokens": self.spec_accepted_tokens,
             "t_forward_graph_s": round(self.t_forward_graph, 2),
             "t_forward_eager_s": round(self.t_forward_eager, 2),
             "t_sample_s": round(self.t_sample, 2),

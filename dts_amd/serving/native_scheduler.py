@@ -22,6 +22,8 @@ class NativeScheduler:
         block_size: int,
         max_batch_tokens: int = 8192,
         max_running: int = 256,
+        spec_k: int = 0,
+        max_spec_rows: int = 16,
     ) -> None:
         from dts_amd.core import load_core
 
@@ -29,7 +31,12 @@ class NativeScheduler:
         if core_mod is None:
             raise RuntimeError("_dts_core extension not built")
         self.core = core_mod.CoreScheduler(
-            num_blocks, block_size, max_batch_tokens, max_running
+            num_blocks,
+            block_size,
+            max_batch_tokens,
+            max_running,
+            spec_k,
+            max_spec_rows,
         )
         self.block_size = block_size
         self._seqs: dict = {}
@@ -55,7 +62,15 @@ class NativeScheduler:
     def add(self, seq: Sequence) -> None:
         self._seqs[seq.seq_id] = seq
         seq.status = SeqStatus.WAITING
-        self.core.add(seq.seq_id, [int(t) for t in seq.tokens])
+        # guided (constrained-form) seqs never speculate: drafts would
+        # fight the forced-token machinery
+        self.core.add(
+            seq.seq_id, [int(t) for t in seq.tokens], seq.guide is None
+        )
+
+    def set_accepted(self, seq: Sequence, accepted: int) -> None:
+        """Engine feedback after speculative verification."""
+        self.core.set_sched_chunk(seq.seq_id, int(accepted))
 
     def append_token(self, seq: Sequence, tok: int) -> None:
         seq.append_token(tok)
@@ -104,6 +119,13 @@ class NativeScheduler:
         )
         batch._scheduled = scheduled  # type: ignore[attr-defined]
         batch._sampled_seqs = sampled  # type: ignore[attr-defined]
+        batch._sample_pos = d["sample_pos"]  # type: ignore[attr-defined]
+        batch._row_groups = [  # type: ignore[attr-defined]
+            (self._seqs[sid], n) for sid, n in d["row_groups"]
+        ]
+        batch._spec_drafts = {  # type: ignore[attr-defined]
+            sid: [int(t) for t in toks] for sid, toks in d["spec"]
+        }
         return batch
 
     def advance_computed(self, batch: ForwardBatch) -> None:

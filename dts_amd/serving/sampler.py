@@ -74,10 +74,20 @@ class Sampler:
         sub = row[idx].float().cpu()
         return self._pick_constrained(sub, seq, allowed)
 
-    def sample(self, logits: torch.Tensor, seqs: list) -> list:
-        """logits [S, V] fp32 for the sampled rows; returns token ids."""
+    def sample(
+        self, logits: torch.Tensor, seqs: list, positions: Optional[list] = None
+    ) -> list:
+        """logits [S, V] fp32 for the sampled rows; returns token ids.
+
+        positions[i] = token index row i draws (defaults to
+        len(seq.tokens)); speculative draft rows pass len+j so the seeded
+        stream is identical with speculation on or off — seeding is
+        stateless in (seed, position) on CPU and GPU alike.
+        """
         S, V = logits.shape
         assert S == len(seqs)
+        if positions is None:
+            positions = [len(s.tokens) for s in seqs]
 
         out: list = [None] * S
         free_rows: list = []
@@ -95,6 +105,7 @@ class Sampler:
             return out
 
         free_seqs = [seqs[i] for i in free_rows]
+        free_pos = [positions[i] for i in free_rows]
         free_logits = (
             logits
             if len(free_rows) == S
@@ -122,10 +133,11 @@ class Sampler:
                     # (seed, position) only — the stream must not depend
                     # on seq_id (submission order / co-batched traffic)
                     # and stays identical across a preempt+recompute
-                    (hash((s.params.seed or 0, len(s.tokens))) & 0x7FFFFFFF)
+                    # or a speculative re-derivation of the same position
+                    (hash((s.params.seed or 0, p)) & 0x7FFFFFFF)
                     if s.params.seed is not None
                     else _random.getrandbits(31)
-                    for s in free_seqs
+                    for s, p in zip(free_seqs, free_pos)
                 ],
                 dtype=torch.long,
                 device=logits.device,
@@ -151,7 +163,17 @@ class Sampler:
         else:
             for i, seq, allowed in guided:
                 out[i] = self._sample_constrained(logits[i], seq, allowed)
-            gens = [self._generator_for(s) for s in free_seqs]
+            # stateless per-(seed, position) generators, matching the GPU
+            # kernel's seeding discipline — a persistent generator would
+            # advance on unaccepted draft rows and diverge with spec on
+            gens = []
+            for s, p in zip(free_seqs, free_pos):
+                if s.params.seed is None:
+                    gens.append(None)
+                else:
+                    g = torch.Generator(device="cpu")
+                    g.manual_seed(hash((int(s.params.seed), p)) & 0x7FFFFFFF)
+                    gens.append(g)
             toks = ops.top_p_sample(free_logits.cpu(), temps, top_ps, generators=gens)
         for i, t in zip(free_rows, toks):
             out[i] = int(t)

@@ -33,11 +33,18 @@ class Scheduler:
         block_manager: BlockManager,
         max_batch_tokens: int = 8192,
         max_running: int = 256,
+        spec_k: int = 0,
+        max_spec_rows: int = 16,
     ) -> None:
         self.bm = block_manager
         self.block_size = block_manager.block_size
         self.max_batch_tokens = max_batch_tokens
         self.max_running = max_running
+        # speculative decoding: up to spec_k prompt-lookup draft tokens
+        # per decode seq (serving/spec.py), capped so total decode rows
+        # stay graph-bucket friendly
+        self.spec_k = spec_k
+        self.max_spec_rows = max_spec_rows
         self.waiting: deque = deque()
         self.running: list = []
         self.stuck: list = []  # seqs that can NEVER fit (engine fails them)
@@ -53,6 +60,10 @@ class Scheduler:
         seq.arrival_order = self._arrival
         self._arrival += 1
         seq.status = SeqStatus.WAITING
+        if self.spec_k > 0 and seq.guide is None:
+            from dts_amd.serving.spec import NgramIndex
+
+            seq._ngram = NgramIndex(seq.tokens)  # type: ignore[attr-defined]
         self.waiting.append(seq)
 
     def _admit(self, seq: Sequence) -> bool:
@@ -88,10 +99,21 @@ class Scheduler:
 
     def append_token(self, seq: Sequence, tok: int) -> None:
         seq.append_token(tok)
+        ng = getattr(seq, "_ngram", None)
+        if ng is not None:
+            ng.extend(seq.tokens)
 
     def extend_tokens(self, seq: Sequence, toks: list) -> None:
         seq.tokens.extend(toks)
         seq.output_tokens.extend(toks)
+        ng = getattr(seq, "_ngram", None)
+        if ng is not None:
+            ng.extend(seq.tokens)
+
+    def set_accepted(self, seq: Sequence, accepted: int) -> None:
+        """Engine feedback after speculative verification: this step
+        advanced `accepted` tokens for seq (1 ≤ accepted ≤ 1+drafts)."""
+        seq._sched_chunk = accepted  # type: ignore[attr-defined]
 
     # ------------------------------------------------------------------
     def _ensure_blocks(self, seq: Sequence, upto_tokens: int) -> bool:
@@ -163,7 +185,8 @@ class Scheduler:
 
         budget = self.max_batch_tokens
         prefills: list = []  # (seq, chunk)
-        decodes: list = []
+        decodes: list = []  # (seq, draft_tokens)
+        spec_rows = 0
         for seq in list(self.running):
             # snapshot iteration: a preemption triggered by an EARLIER seq
             # in this loop removes its victim from self.running — the stale
@@ -193,17 +216,34 @@ class Scheduler:
             chunk = min(remaining, budget)
             if chunk <= 0:
                 continue
-            budget -= chunk
             # mark immediately: a seq already placed in this step's batch
             # must not be chosen as a preemption victim by a LATER seq in
             # this same loop (running order != arrival order after a
             # re-admission), or _build_batch would index released blocks
             seq.in_flight = True
-            if remaining == 1 and seq.output_tokens:
-                decodes.append(seq)
-            elif chunk == 1:
-                decodes.append(seq)
+            if remaining == 1:
+                # true decode row (last token) — maybe add draft rows.
+                # (a budget-starved chunk==1 MID-prompt is a prefill
+                # chunk: classifying it as decode would sample a bogus
+                # mid-prompt token)
+                drafts: list = []
+                ng = getattr(seq, "_ngram", None)
+                if (
+                    ng is not None
+                    and seq.guide is None
+                    and spec_rows + 1 + self.spec_k <= self.max_spec_rows
+                    and budget > 1
+                ):
+                    drafts = ng.propose(seq.tokens, min(self.spec_k, budget - 1))
+                    if drafts and not self._ensure_blocks(
+                        seq, len(seq.tokens) + len(drafts)
+                    ):
+                        drafts = []  # never preempt for draft rows
+                spec_rows += 1 + len(drafts)
+                budget -= 1 + len(drafts)
+                decodes.append((seq, drafts))
             else:
+                budget -= chunk
                 prefills.append((seq, chunk))
             if budget <= 0:
                 break
@@ -219,6 +259,9 @@ class Scheduler:
         slots: list = []
         sample_indices: list = []
         sampled_seqs: list = []
+        sample_pos: list = []  # token index each sampled row draws
+        row_groups: list = []  # (seq, n_rows) in sampled-row order
+        spec_drafts: dict = {}
 
         cu_q = [0]
         pf_tables: list = []
@@ -236,21 +279,39 @@ class Scheduler:
             if end == len(seq.tokens):
                 sample_indices.append(len(token_ids) - 1)
                 sampled_seqs.append(seq)
+                sample_pos.append(len(seq.tokens))
+                row_groups.append((seq, 1))
             seq._sched_chunk = chunk  # type: ignore[attr-defined]
 
         num_prefill_tokens = len(token_ids)
 
         dc_tables: list = []
         dc_kv_lens: list = []
-        for seq in decodes:
+        num_decode_rows = 0
+        for seq, drafts in decodes:
             p = seq.num_computed
-            token_ids.append(seq.tokens[p])
-            positions.append(p)
-            slots.append(seq.block_table[p // bs] * bs + p % bs)
-            dc_tables.append(list(seq.block_table))
-            dc_kv_lens.append(p + 1)
-            sample_indices.append(len(token_ids) - 1)
-            sampled_seqs.append(seq)
+            # row 0 = the real tail token; rows 1..k = draft candidates.
+            # Each draft row is an ordinary 1-token decode row (own
+            # position and kv_len): rope_kv_append writes all rows' KV
+            # before attention runs, so row j attends rows < j of the
+            # same seq written this very step.
+            for j, t in enumerate([seq.tokens[p]] + drafts):
+                pos = p + j
+                token_ids.append(t)
+                positions.append(pos)
+                slots.append(seq.block_table[pos // bs] * bs + pos % bs)
+                dc_tables.append(list(seq.block_table))
+                dc_kv_lens.append(pos + 1)
+                sample_indices.append(len(token_ids) - 1)
+                sampled_seqs.append(seq)
+                sample_pos.append(len(seq.tokens) + j)
+                num_decode_rows += 1
+            row_groups.append((seq, 1 + len(drafts)))
+            if drafts:
+                spec_drafts[seq.seq_id] = drafts
+            # conservative default: callers that never verify drafts
+            # (direct scheduler use) advance by the single real token;
+            # the engine raises it via set_accepted after verification
             seq._sched_chunk = 1  # type: ignore[attr-defined]
 
         def pad_tables(tables: list) -> Optional[torch.Tensor]:
@@ -272,27 +333,37 @@ class Scheduler:
             prefill_kv_lens=(
                 torch.tensor(pf_kv_lens, dtype=torch.int32) if prefills else None
             ),
-            num_decode_seqs=len(decodes),
+            num_decode_seqs=num_decode_rows,
             decode_block_tables=pad_tables(dc_tables),
             decode_kv_lens=(
                 torch.tensor(dc_kv_lens, dtype=torch.int32) if decodes else None
             ),
             sample_indices=torch.tensor(sample_indices, dtype=torch.long),
         )
-        scheduled = [s for s, _ in prefills] + decodes
+        scheduled = [s for s, _ in prefills] + [s for s, _ in decodes]
         for s in scheduled:
             s.in_flight = True
         batch._scheduled = scheduled  # type: ignore[attr-defined]
         batch._sampled_seqs = sampled_seqs  # type: ignore[attr-defined]
+        batch._sample_pos = sample_pos  # type: ignore[attr-defined]
+        batch._row_groups = row_groups  # type: ignore[attr-defined]
+        batch._spec_drafts = spec_drafts  # type: ignore[attr-defined]
         return batch
 
     # ------------------------------------------------------------------
     def advance_computed(self, batch: ForwardBatch) -> None:
-        """After a forward: bump num_computed and register full blocks."""
+        """After a forward: bump num_computed and register full blocks.
+
+        Runs AFTER the engine appended this step's verified tokens (so
+        speculative chunks register block content that exists); finished
+        seqs released their blocks already and are skipped.
+        """
         for seq in batch._scheduled:  # type: ignore[attr-defined]
+            seq.in_flight = False
+            if seq.status != SeqStatus.RUNNING:
+                continue
             chunk = getattr(seq, "_sched_chunk", 0)
             seq.num_computed += chunk
-            seq.in_flight = False
             self._register_full_blocks(seq)
 
     def _register_full_blocks(self, seq: Sequence) -> None:
