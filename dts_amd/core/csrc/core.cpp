@@ -149,6 +149,31 @@ class CoreScheduler {
     if (it != seqs_.end()) it->second.sched_chunk = chunk;
   }
 
+  // ---- chained-decode hooks (dts_amd/serving/chain.py) ------------------
+  int64_t waiting_count() const { return (int64_t)waiting_.size(); }
+
+  bool reserve_tokens(int64_t seq_id, int upto_tokens) {
+    auto it = seqs_.find(seq_id);
+    if (it == seqs_.end()) return false;
+    return ensure_blocks(it->second, upto_tokens);
+  }
+
+  void chain_advance(int64_t seq_id, int32_t tok) {
+    auto it = seqs_.find(seq_id);
+    if (it == seqs_.end()) return;
+    Seq& s = it->second;
+    s.tokens.push_back(tok);
+    if (spec_k_ > 0 && s.allow_spec) s.index_tokens();
+    s.num_computed += 1;
+    register_full_blocks(s);
+  }
+
+  std::vector<int32_t> get_block_table(int64_t seq_id) const {
+    auto it = seqs_.find(seq_id);
+    if (it == seqs_.end()) return {};
+    return it->second.block_table;
+  }
+
   int64_t num_tokens(int64_t seq_id) const {
     return (int64_t)seqs_.at(seq_id).tokens.size();
   }
@@ -615,6 +640,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
       .def("add", &CoreScheduler::add, py::arg("seq_id"), py::arg("tokens"),
            py::arg("allow_spec") = true)
       .def("set_sched_chunk", &CoreScheduler::set_sched_chunk)
+      .def("waiting_count", &CoreScheduler::waiting_count)
+      .def("reserve_tokens", &CoreScheduler::reserve_tokens)
+      .def("chain_advance", &CoreScheduler::chain_advance)
+      .def("get_block_table", &CoreScheduler::get_block_table)
       .def("append_token", &CoreScheduler::append_token)
       .def("extend_tokens", &CoreScheduler::extend_tokens)
       .def("num_tokens", &CoreScheduler::num_tokens)

@@ -165,16 +165,32 @@ def gelu(x):
     return torch_ref.gelu(x)
 
 
-def top_p_sample(logits, temperatures, top_ps, generators=None, seeds=None):
+def top_p_sample(logits, temperatures, top_ps, generators=None, seeds=None, out=None):
     if _on_gpu(logits):
         ext = _require_hip()
         if ext is not None and seeds is not None:
-            out = torch.empty(
-                logits.shape[0], dtype=torch.long, device=logits.device
-            )
+            if out is None:
+                out = torch.empty(
+                    logits.shape[0], dtype=torch.long, device=logits.device
+                )
             ext.top_p_sample(out, logits, temperatures, top_ps, seeds)
             return out
     return torch_ref.top_p_sample(logits, temperatures, top_ps, generators)
+
+
+def derive_seeds(out, bases, positions):
+    """out[i] = mix_seed(bases[i], positions[i] + 1) — on-device for the
+    chained decode loop (no host round-trip per step)."""
+    if _on_gpu(out):
+        ext = _require_hip()
+        if ext is not None:
+            ext.derive_seeds(out, bases, positions)
+            return out
+    out.copy_(torch_ref.derive_seeds(bases, positions))
+    return out
+
+
+mix_seed = torch_ref.mix_seed
 
 
 def linear_bf16(x: torch.Tensor, weight: torch.Tensor) -> torch.Tensor:

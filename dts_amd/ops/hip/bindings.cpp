@@ -23,6 +23,8 @@ void attn_prefill_paged(torch::Tensor out, torch::Tensor q, torch::Tensor cu_q,
                         torch::Tensor kv_lens, double scale, int64_t swz);
 void top_p_sample(torch::Tensor out, torch::Tensor logits, torch::Tensor temps,
                   torch::Tensor top_ps, torch::Tensor seeds);
+void derive_seeds(torch::Tensor out, torch::Tensor bases,
+                  torch::Tensor positions);
 void gemv_bf16(torch::Tensor out, torch::Tensor x, torch::Tensor w);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -43,6 +45,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("scale"), py::arg("swz") = -1);
   m.def("top_p_sample", &top_p_sample,
         "fused temperature softmax + top-p sampling (sort-free)");
+  m.def("derive_seeds", &derive_seeds,
+        "stateless (seed, position) mix for chained decode");
   m.def("gemv_bf16", &gemv_bf16,
         "skinny-batch (M<=8) bf16 weight-streaming GEMV");
 }

@@ -39,6 +39,31 @@ DEV unsigned long long xorshift64(unsigned long long x) {
   return x;
 }
 
+// splitmix64-style stateless seed mix over (base, position): the SAME
+// formula runs on the host (dts_amd/ops/torch_ref.py mix_seed) so the
+// chained-decode path (seeds derived on device from static positions)
+// draws the exact token the per-step host path would.
+DEV unsigned long long mix_seed64(unsigned long long base,
+                                  unsigned long long pos) {
+  unsigned long long x = base ^ (pos * 0x9E3779B97F4A7C15ull);
+  x = (x ^ (x >> 30)) * 0xBF58476D1CE4E5B9ull;
+  x = (x ^ (x >> 27)) * 0x94D049BB133111EBull;
+  x ^= x >> 31;
+  return x & 0x7FFFFFFFull;
+}
+
+// out[i] = mix(base[i], positions[i] + 1): positions are QUERY positions
+// of the decode rows; the drawn token's index is one past the query.
+__global__ void derive_seeds_kernel(long* __restrict__ out,
+                                    const long* __restrict__ bases,
+                                    const long* __restrict__ positions,
+                                    int n) {
+  int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < n)
+    out[i] = (long)mix_seed64((unsigned long long)bases[i],
+                              (unsigned long long)(positions[i] + 1));
+}
+
 template <int BLOCK>
 __global__ void __launch_bounds__(BLOCK)
 top_p_sample_kernel(long* __restrict__ out, const float* __restrict__ logits,
@@ -188,6 +213,16 @@ top_p_sample_kernel(long* __restrict__ out, const float* __restrict__ logits,
     }
     out[row] = r;
   }
+}
+
+void derive_seeds(torch::Tensor out, torch::Tensor bases,
+                  torch::Tensor positions) {
+  const int n = out.size(0);
+  hipLaunchKernelGGL(derive_seeds_kernel, dim3((n + 255) / 256), dim3(256), 0,
+                     c10::hip::getCurrentHIPStream(), (long*)out.data_ptr(),
+                     (const long*)bases.data_ptr(),
+                     (const long*)positions.data_ptr(), n);
+  HIP_CHECK_LAST();
 }
 
 void top_p_sample(torch::Tensor out, torch::Tensor logits, torch::Tensor temps,

@@ -206,6 +206,36 @@ def gelu(x: torch.Tensor) -> torch.Tensor:
 # Sampling
 # ---------------------------------------------------------------------------
 
+_MASK64 = (1 << 64) - 1
+
+
+def mix_seed(base: int, pos: int) -> int:
+    """Stateless splitmix64-style seed mix over (base seed, token index).
+
+    Bit-identical to the HIP device function mix_seed64
+    (ops/hip/sampling.hip): the host per-step path and the on-device
+    chained-decode path must draw the same token for the same position.
+    """
+    x = (base ^ (pos * 0x9E3779B97F4A7C15)) & _MASK64
+    x = ((x ^ (x >> 30)) * 0xBF58476D1CE4E5B9) & _MASK64
+    x = ((x ^ (x >> 27)) * 0x94D049BB133111EB) & _MASK64
+    x ^= x >> 31
+    return x & 0x7FFFFFFF
+
+
+def derive_seeds(
+    bases: torch.Tensor, positions: torch.Tensor
+) -> torch.Tensor:
+    """CPU reference of the derive_seeds kernel: mix(base, pos + 1)."""
+    return torch.tensor(
+        [
+            mix_seed(int(b), int(p) + 1)
+            for b, p in zip(bases.tolist(), positions.tolist())
+        ],
+        dtype=torch.long,
+    )
+
+
 def top_p_sample(
     logits: torch.Tensor,  # [B, V] fp32
     temperatures: torch.Tensor,  # [B]

@@ -172,6 +172,7 @@ class ServingEngine:
             self.template = ChatTemplate(self.tokenizer)
 
         self._graph_runner = None
+        self._chain = None
         if (
             device.startswith("cuda")
             and os.environ.get("DTS_NO_HIPGRAPH") != "1"
@@ -187,6 +188,10 @@ class ServingEngine:
                 max_blocks_per_seq=self.spec.max_position // block_size,
                 max_bucket=min(256, max_running),
             )
+            if os.environ.get("DTS_NO_CHAIN") != "1":
+                from dts_amd.serving.chain import ChainRunner
+
+                self._chain = ChainRunner(self._graph_runner, device)
 
         self._lock = threading.Lock()
         self._work = threading.Condition(self._lock)
@@ -207,6 +212,8 @@ class ServingEngine:
         self.prefill_steps = 0
         self.spec_draft_tokens = 0
         self.spec_accepted_tokens = 0
+        self.chain_steps = 0
+        self.chains = 0
         self.req_count = 0
         self.req_latency_sum = 0.0
         self.req_latency_max = 0.0
@@ -273,8 +280,11 @@ class ServingEngine:
             return False
         import time as _time
 
-        self.steps += 1
         t0 = _time.perf_counter()
+        if self._chain is not None and self._chain.eligible(batch):
+            if self._run_chain(batch):
+                return True
+        self.steps += 1
         if self._graph_runner is not None and self._graph_runner.can_run(batch):
             logits = self._graph_runner.run(batch)
             self.graph_steps += 1
@@ -333,6 +343,97 @@ class ServingEngine:
             self.scheduler.advance_computed(batch)
         self.t_post += _time.perf_counter() - t2
         return True
+
+    def _run_chain(self, batch) -> bool:
+        """Chained device-resident decode (serving/chain.py): enqueue up
+        to CHAIN_MAX graph replays with on-device token feedback, process
+        tokens lagging the GPU, sync once. Returns False to fall back to
+        the normal single-step path (reserve failure / too close to the
+        context cap)."""
+        import time as _time
+
+        from dts_amd.serving.chain import BURST, CHAIN_MAX, MIN_CHAIN
+
+        seqs = batch._sampled_seqs  # type: ignore[attr-defined]
+        with self._lock:
+            W = CHAIN_MAX
+            for s in seqs:
+                W = min(W, self.spec.max_position - 8 - len(s.tokens))
+                W = min(W, s.params.max_tokens - s.num_generated + 1)
+            if W < MIN_CHAIN:
+                return False
+            for s in seqs:
+                if not self.scheduler.reserve_tokens(s, len(s.tokens) + W):
+                    return False  # KV pressure: normal path handles it
+            # reservation may have grown block tables — rebuild the rows
+            # the statics will load (the batch's tensor predates reserve)
+            tbls = [self.scheduler.block_table_of(s) for s in seqs]
+            m = max(len(t) for t in tbls)
+            batch.decode_block_tables = torch.tensor(
+                [t + [0] * (m - len(t)) for t in tbls], dtype=torch.int32
+            )
+        t0 = _time.perf_counter()
+        ctx = self._chain.prepare(batch, seqs)
+        active = [True] * len(seqs)
+        deferred: list = []  # (seq, reason) — finished AFTER the sync
+        executed = 0
+        processed = 0
+        stop = False
+        while not stop and executed < W:
+            for _ in range(min(BURST, W - executed)):
+                self._chain.launch_step(ctx, executed)
+                executed += 1
+            # catch up on whatever the GPU already finished (no blocking)
+            while processed < executed and self._chain.step_ready(processed):
+                self._process_chain_step(ctx, seqs, active, deferred, processed)
+                processed += 1
+            if not any(active) or self.scheduler.waiting_count() > 0:
+                stop = True
+        # drain: ONE real wait on the last outstanding step
+        while processed < executed:
+            self._chain.wait_step(processed)
+            self._process_chain_step(ctx, seqs, active, deferred, processed)
+            processed += 1
+        with self._lock:
+            # all in-flight KV writes are complete (last event synced) —
+            # safe to release finished seqs' blocks now
+            for seq, reason in deferred:
+                self._finish(seq, reason)
+            for s in seqs:
+                self.scheduler.set_accepted(s, 0)
+            self.scheduler.advance_computed(batch)
+        self.steps += executed
+        self.graph_steps += executed
+        self.chain_steps += executed
+        self.chains += 1
+        self.t_forward_graph += _time.perf_counter() - t0
+        return True
+
+    def _process_chain_step(
+        self, ctx: dict, seqs: list, active: list, deferred: list, i: int
+    ) -> None:
+        toks = self._chain.tokens_of(ctx, i)
+        with self._lock:
+            for j, seq in enumerate(seqs):
+                if not active[j]:
+                    continue  # finished earlier in the chain: garbage row
+                tok = toks[j]
+                self.tokens_sampled += 1
+                self.scheduler.chain_advance(seq, tok)
+                if seq.stream_cb is not None:
+                    try:
+                        seq.stream_cb([tok])
+                    except Exception:  # noqa: BLE001
+                        seq.stream_cb = None
+                if tok in self.template.stop_token_ids:
+                    active[j] = False
+                    deferred.append((seq, "stop"))
+                elif (
+                    len(seq.tokens) + 8 >= self.spec.max_position
+                    or seq.num_generated >= seq.params.max_tokens
+                ):
+                    active[j] = False
+                    deferred.append((seq, "length"))
 
     def _verify_and_emit(self, seq: Sequence, toks: list, draft: list) -> int:
         """Exact-match speculative verification (serving/spec.py): emit
@@ -500,6 +601,8 @@ class ServingEngine:
             "prefill_steps": self.prefill_steps,
             "spec_draft_tokens": self.spec_draft_tokens,
             "spec_accepted_tokens": self.spec_accepted_tokens,
+            "chain_steps": self.chain_steps,
+            "chains": self.chains,
             "t_forward_graph_s": round(self.t_forward_graph, 2),
             "t_forward_eager_s": round(self.t_forward_eager, 2),
             "t_sample_s": round(self.t_sample, 2),

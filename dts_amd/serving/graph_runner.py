@@ -112,15 +112,22 @@ class DecodeGraphRunner:
         return {"graph": graph, "static": static, "logits": logits}
 
     # ------------------------------------------------------------------
-    def run(self, batch: ForwardBatch) -> torch.Tensor:
-        n = batch.num_decode_seqs
+    def entry_for(self, n: int) -> dict:
+        """Get (or capture) the graph entry for the bucket covering n."""
         B = self._bucket_for(n)
         entry = self._graphs.get(B)
         if entry is None:
             entry = self._capture(B)
+            entry["bucket"] = B
             self._graphs[B] = entry
+        return entry
+
+    def load_batch(self, entry: dict, batch: ForwardBatch) -> None:
+        """Fill the entry's static inputs from a decode batch (padded,
+        one H2D copy per input)."""
+        n = batch.num_decode_seqs
+        B = entry["static"]["token_ids"].shape[0]
         st = entry["static"]
-        # build padded host-side tensors, one H2D copy per input
         tid = torch.full((B,), 0, dtype=torch.long)
         tid[:n] = batch.token_ids
         pos = torch.zeros(B, dtype=torch.long)
@@ -137,5 +144,10 @@ class DecodeGraphRunner:
         st["slot_mapping"].copy_(slots, non_blocking=True)
         st["block_tables"].copy_(tables, non_blocking=True)
         st["kv_lens"].copy_(kvl, non_blocking=True)
+
+    def run(self, batch: ForwardBatch) -> torch.Tensor:
+        n = batch.num_decode_seqs
+        entry = self.entry_for(n)
+        self.load_batch(entry, batch)
         entry["graph"].replay()
         return entry["logits"][:n]

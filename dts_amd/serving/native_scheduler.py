@@ -72,6 +72,23 @@ class NativeScheduler:
         """Engine feedback after speculative verification."""
         self.core.set_sched_chunk(seq.seq_id, int(accepted))
 
+    # -- chained-decode hooks (serving/chain.py) -----------------------
+    def waiting_count(self) -> int:
+        return self.core.waiting_count()
+
+    def reserve_tokens(self, seq: Sequence, upto_tokens: int) -> bool:
+        return self.core.reserve_tokens(seq.seq_id, int(upto_tokens))
+
+    def chain_advance(self, seq: Sequence, tok: int) -> None:
+        seq.append_token(tok)
+        self.core.chain_advance(seq.seq_id, int(tok))
+
+    def block_table_of(self, seq: Sequence) -> list:
+        return list(self.core.get_block_table(seq.seq_id))
+
+    def num_computed_of(self, seq: Sequence) -> int:
+        return int(self.core.num_computed(seq.seq_id))
+
     def append_token(self, seq: Sequence, tok: int) -> None:
         seq.append_token(tok)
         self.core.append_token(seq.seq_id, int(tok))

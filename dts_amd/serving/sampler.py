@@ -130,11 +130,12 @@ class Sampler:
 
             seeds = torch.tensor(
                 [
-                    # (seed, position) only — the stream must not depend
-                    # on seq_id (submission order / co-batched traffic)
-                    # and stays identical across a preempt+recompute
-                    # or a speculative re-derivation of the same position
-                    (hash((s.params.seed or 0, p)) & 0x7FFFFFFF)
+                    # mix(seed, position) only — the stream must not
+                    # depend on seq_id (submission order / co-batched
+                    # traffic), stays identical across preempt+recompute
+                    # or speculative re-derivation, and matches the
+                    # on-device derive_seeds of the chained decode loop
+                    ops.mix_seed(int(s.params.seed), p)
                     if s.params.seed is not None
                     else _random.getrandbits(31)
                     for s, p in zip(free_seqs, free_pos)
@@ -172,7 +173,7 @@ class Sampler:
                     gens.append(None)
                 else:
                     g = torch.Generator(device="cpu")
-                    g.manual_seed(hash((int(s.params.seed), p)) & 0x7FFFFFFF)
+                    g.manual_seed(ops.mix_seed(int(s.params.seed), p))
                     gens.append(g)
             toks = ops.top_p_sample(free_logits.cpu(), temps, top_ps, generators=gens)
         for i, t in zip(free_rows, toks):

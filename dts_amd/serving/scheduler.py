@@ -112,8 +112,31 @@ class Scheduler:
 
     def set_accepted(self, seq: Sequence, accepted: int) -> None:
         """Engine feedback after speculative verification: this step
-        advanced `accepted` tokens for seq (1 ≤ accepted ≤ 1+drafts)."""
+        advanced `accepted` tokens for seq (1 ≤ accepted ≤ 1+drafts).
+        The chained decode loop passes 0 — it advances per-step through
+        chain_advance instead."""
         seq._sched_chunk = accepted  # type: ignore[attr-defined]
+
+    # -- chained-decode hooks (serving/chain.py) -----------------------
+    def waiting_count(self) -> int:
+        return len(self.waiting)
+
+    def reserve_tokens(self, seq: Sequence, upto_tokens: int) -> bool:
+        """Pre-allocate blocks to cover upto_tokens; never preempts."""
+        return self._ensure_blocks(seq, upto_tokens)
+
+    def chain_advance(self, seq: Sequence, tok: int) -> None:
+        """One chained-decode step: the device already wrote this
+        position's KV; append the sampled token and advance."""
+        self.append_token(seq, tok)
+        seq.num_computed += 1
+        self._register_full_blocks(seq)
+
+    def block_table_of(self, seq: Sequence) -> list:
+        return list(seq.block_table)
+
+    def num_computed_of(self, seq: Sequence) -> int:
+        return seq.num_computed
 
     # ------------------------------------------------------------------
     def _ensure_blocks(self, seq: Sequence, upto_tokens: int) -> bool:
