@@ -257,31 +257,42 @@ class ServingEngine:
         return fut
 
     # ------------------------------------------------------------------
+    def _fail_stuck(self) -> None:
+        """Fail futures of never-fitting requests (call under _lock)."""
+        if not self.scheduler.stuck:
+            return
+        from dts_amd.llm.errors import BackendError
+
+        for seq in self.scheduler.stuck:
+            fut = self._futures.pop(seq.seq_id, None)
+            if fut is not None and not fut.done():
+                fut.set_exception(
+                    BackendError(
+                        f"request of {len(seq.tokens)} tokens exceeds "
+                        f"the KV pool ({self.block_manager.num_blocks}"
+                        f"x{self.block_manager.block_size} tokens)"
+                    )
+                )
+        self.scheduler.stuck.clear()
+
     @torch.inference_mode()
     def step(self) -> bool:
         """One scheduling + forward + sample step. Returns True if it ran."""
         with self._lock:
             batch = self.scheduler.schedule()
-            if self.scheduler.stuck:
-                from dts_amd.llm.errors import BackendError
-
-                for seq in self.scheduler.stuck:
-                    fut = self._futures.pop(seq.seq_id, None)
-                    if fut is not None and not fut.done():
-                        fut.set_exception(
-                            BackendError(
-                                f"request of {len(seq.tokens)} tokens exceeds "
-                                f"the KV pool ({self.block_manager.num_blocks}"
-                                f"x{self.block_manager.block_size} tokens)"
-                            )
-                        )
-                self.scheduler.stuck.clear()
+            self._fail_stuck()
         if batch is None:
             return False
+        return self._execute(batch)
+
+    @torch.inference_mode()
+    def _execute(self, batch, allow_chain: bool = True) -> bool:
+        """Forward + sample + postprocess one scheduled batch (the TP
+        driver calls this directly after broadcasting the batch)."""
         import time as _time
 
         t0 = _time.perf_counter()
-        if self._chain is not None and self._chain.eligible(batch):
+        if allow_chain and self._chain is not None and self._chain.eligible(batch):
             if self._run_chain(batch):
                 return True
         self.steps += 1
