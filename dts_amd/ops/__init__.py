@@ -194,35 +194,32 @@ mix_seed = torch_ref.mix_seed
 
 
 def linear_bf16(x: torch.Tensor, weight: torch.Tensor) -> torch.Tensor:
-    """y = x @ W^T. Skinny decode batches (M<=8, bf16, K%512==0) stream
-    through the custom GEMV kernel (hipBLASLt runs ~50% of the HBM
-    roofline at these M — profiles/); everything else is a hipBLASLt GEMM
-    via F.linear."""
-    # measured crossover (profiles/ GEMM sweep): hipBLASLt has a ~22 us
-    # per-GEMM floor at decode M regardless of weight size, so the custom
-    # GEMV wins (a) at M<=2 on every shape (19.4 us = 6.05 TB/s on the
-    # 117 MB gate_up), and (b) up to M=8 on the smaller qkv/o weights
-    # (<=72 MB streams in ~9-17 us); hipBLASLt keeps the large weights at
-    # M>=3 where its tiling amortizes the x re-reads.
+    """y = x @ W^T. Skinny decode batches (M<=16, bf16, K%512==0) stream
+    through custom weight-streaming kernels; everything else is a
+    hipBLASLt GEMM via F.linear.
+
+    Dispatch (measured, profiles/): the VALU GEMV (gemv.hip) hits
+    6.05 TB/s at M<=2 on every shape; past that it goes ALU-bound, and
+    hipBLASLt runs ~50% of the roofline with a ~22 us per-GEMM floor at
+    decode M — so M=3..16 takes the MFMA skinny GEMM (gemm_skinny.hip),
+    which keeps the arithmetic on the matrix cores and the reads at
+    streaming rate."""
     if (
         x.is_cuda
         and x.dtype == torch.bfloat16
         and x.dim() == 2
         and x.shape[1] % 512 == 0
         and weight.stride(1) == 1
-        and (
-            1 <= x.shape[0] <= 2
-            or (
-                x.shape[0] <= 8
-                and weight.shape[0] * weight.shape[1] * 2 <= (72 << 20)
-            )
-        )
+        and 1 <= x.shape[0] <= 16
     ):
         ext = _require_hip()
         if ext is not None:
             out = torch.empty(
                 (x.shape[0], weight.shape[0]), dtype=x.dtype, device=x.device
             )
-            ext.gemv_bf16(out, x, weight)
+            if x.shape[0] <= 2:
+                ext.gemv_bf16(out, x, weight)
+            else:
+                ext.gemm_skinny_bf16(out, x, weight)
             return out
     return torch.nn.functional.linear(x, weight)

@@ -31,6 +31,7 @@ def build(verbose: bool = True) -> Path:
         str(HIP_DIR / "attention.hip"),
         str(HIP_DIR / "sampling.hip"),
         str(HIP_DIR / "gemv.hip"),
+        str(HIP_DIR / "gemm_skinny.hip"),
     ]
     module = load(
         name="_dts_hip",

@@ -26,6 +26,7 @@ void top_p_sample(torch::Tensor out, torch::Tensor logits, torch::Tensor temps,
 void derive_seeds(torch::Tensor out, torch::Tensor bases,
                   torch::Tensor positions);
 void gemv_bf16(torch::Tensor out, torch::Tensor x, torch::Tensor w);
+void gemm_skinny_bf16(torch::Tensor out, torch::Tensor x, torch::Tensor w);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm", &rmsnorm, "RMSNorm (bf16, CDNA4)");
@@ -49,4 +50,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "stateless (seed, position) mix for chained decode");
   m.def("gemv_bf16", &gemv_bf16,
         "skinny-batch (M<=8) bf16 weight-streaming GEMV");
+  m.def("gemm_skinny_bf16", &gemm_skinny_bf16,
+        "skinny-M (M<=16) bf16 MFMA weight-streaming GEMM");
 }

@@ -102,9 +102,12 @@ top_p_sample_kernel(long* __restrict__ out, const float* __restrict__ logits,
 #pragma unroll
     for (int i = 0; i < BLOCK / WAVE; ++i) bm = fmaxf(bm, red[i]);
     if (T <= 0.f) {
-      if (tid == 0) s_argmax = -1;
+      // deterministic tie-break: LOWEST index among equal maxima (the
+      // torch reference argmax does the same; an atomicCAS race here
+      // made chained-vs-stepped greedy streams diverge on real ties)
+      if (tid == 0) s_argmax = 0x7FFFFFFF;
       __syncthreads();
-      if (mymax == bm) atomicCAS(&s_argmax, -1, myarg);
+      if (mymax == bm) atomicMin(&s_argmax, myarg);
       __syncthreads();
       if (tid == 0) out[row] = s_argmax;
       return;

@@ -134,6 +134,16 @@ def main():
                     ext.gemv_bf16(out, x, w)
                 torch.cuda.synchronize()
                 t_gemv = (time.perf_counter() - t0) / 50
+            t_skinny = None
+            if M <= 16:
+                for _ in range(3):
+                    ext.gemm_skinny_bf16(out, x, w)
+                torch.cuda.synchronize()
+                t0 = time.perf_counter()
+                for _ in range(50):
+                    ext.gemm_skinny_bf16(out, x, w)
+                torch.cuda.synchronize()
+                t_skinny = (time.perf_counter() - t0) / 50
             for _ in range(3):
                 torch.nn.functional.linear(x, w)
             torch.cuda.synchronize()
@@ -145,6 +155,10 @@ def main():
             print(json.dumps({
                 "probe": f"gemm_M{M}_K{K}_N{N}",
                 "gemv_us": round(t_gemv * 1e6, 1) if t_gemv else None,
+                "skinny_us": round(t_skinny * 1e6, 1) if t_skinny else None,
+                "skinny_TBps": (
+                    round(N * K * 2 / t_skinny / 1e12, 2) if t_skinny else None
+                ),
                 "hipblaslt_us": round(t_blas * 1e6, 1),
                 "blas_TBps": round(N * K * 2 / t_blas / 1e12, 2),
             }))

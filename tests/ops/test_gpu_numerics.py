@@ -255,6 +255,38 @@ class TestGemv:
         torch.testing.assert_close(out.cpu().float(), ref, atol=3e-2, rtol=3e-2)
 
 
+class TestGemmSkinny:
+    @pytest.mark.parametrize(
+        "M,K,N",
+        [
+            (3, 4096, 4096),
+            (6, 4096, 14336),
+            (8, 14336, 4096),
+            (16, 4096, 6144),
+            (5, 4096, 128256),
+            (16, 4096, 100),  # N not a multiple of the 16-wide tile
+        ],
+    )
+    def test_skinny_matches_dense(self, M, K, N):
+        x = bf(torch.randn(M, K)).to(DEV)
+        w = bf(torch.randn(N, K) / math.sqrt(K)).to(DEV)
+        out = torch.empty(M, N, dtype=torch.bfloat16, device=DEV)
+        ext.gemm_skinny_bf16(out, x, w)
+        ref = x.cpu().float() @ w.cpu().float().T
+        torch.testing.assert_close(out.cpu().float(), ref, atol=3e-2, rtol=3e-2)
+
+    def test_skinny_strided_x_and_out(self):
+        M, K, N = 6, 4096, 512
+        big = bf(torch.randn(M, 2 * K)).to(DEV)
+        x = big[:, :K]  # strided rows (fused qkv views)
+        w = bf(torch.randn(N, K) / math.sqrt(K)).to(DEV)
+        out_big = torch.empty(M, 2 * N, dtype=torch.bfloat16, device=DEV)
+        out = out_big[:, :N]
+        ext.gemm_skinny_bf16(out, x, w)
+        ref = x.cpu().float() @ w.cpu().float().T
+        torch.testing.assert_close(out.cpu().float(), ref, atol=3e-2, rtol=3e-2)
+
+
 class TestSampling:
     def test_greedy_matches_argmax(self):
         S, V = 7, 128256
