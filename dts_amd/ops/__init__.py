@@ -198,28 +198,32 @@ def linear_bf16(x: torch.Tensor, weight: torch.Tensor) -> torch.Tensor:
     through custom weight-streaming kernels; everything else is a
     hipBLASLt GEMM via F.linear.
 
-    Dispatch (measured, profiles/): the VALU GEMV (gemv.hip) hits
-    6.05 TB/s at M<=2 on every shape; past that it goes ALU-bound, and
-    hipBLASLt runs ~50% of the roofline with a ~22 us per-GEMM floor at
-    decode M — so M=3..16 takes the MFMA skinny GEMM (gemm_skinny.hip),
-    which keeps the arithmetic on the matrix cores and the reads at
-    streaming rate."""
+    Dispatch (measured A/B, profiles/microbench r2): hipBLASLt has a
+    ~20 us per-GEMM floor, which caps weights <= ~96 MB at ~2.5 TB/s —
+    there the MFMA skinny GEMM (gemm_skinny.hip) wins at every M<=16
+    (10.9 us on the 75 MB qkv). On the ~117 MB gate_up/down weights
+    hipBLASLt streams at 5.7-5.8 TB/s and wins from M>=5; at M<=2 the
+    VALU GEMV leads on shallow-K shapes but underfills the grid at
+    K=14336/N=4096 (32 us vs skinny's 20.6), hence the K split."""
+    M = x.shape[0] if x.dim() == 2 else 0
     if (
         x.is_cuda
         and x.dtype == torch.bfloat16
         and x.dim() == 2
         and x.shape[1] % 512 == 0
         and weight.stride(1) == 1
-        and 1 <= x.shape[0] <= 16
+        and 1 <= M <= 16
     ):
-        ext = _require_hip()
-        if ext is not None:
-            out = torch.empty(
-                (x.shape[0], weight.shape[0]), dtype=x.dtype, device=x.device
-            )
-            if x.shape[0] <= 2:
-                ext.gemv_bf16(out, x, weight)
-            else:
-                ext.gemm_skinny_bf16(out, x, weight)
-            return out
+        big = weight.shape[0] * weight.shape[1] * 2 > (96 << 20)
+        if not (big and M >= 5):  # big weights at M>=5: hipBLASLt wins
+            ext = _require_hip()
+            if ext is not None:
+                out = torch.empty(
+                    (M, weight.shape[0]), dtype=x.dtype, device=x.device
+                )
+                if M <= 2 and big and x.shape[1] <= 8192:
+                    ext.gemv_bf16(out, x, weight)
+                else:
+                    ext.gemm_skinny_bf16(out, x, weight)
+                return out
     return torch.nn.functional.linear(x, weight)

@@ -87,6 +87,7 @@ class DTSEngine:
             on_usage=self._track_usage,
             max_tokens=config.budget.judge,
             seed=config.seed,
+            comparative_split=config.comparative_split,
         )
         # optional deep-research provider (dts_amd/search/researcher.py)
         self._researcher = researcher

@@ -43,6 +43,7 @@ class TrajectoryEvaluator:
         deep_research_context: Optional[str] = None,
         max_tokens: int = 1024,
         seed: Optional[int] = None,
+        comparative_split: bool = False,
     ) -> None:
         self.llm = llm
         self.goal = goal
@@ -63,6 +64,12 @@ class TrajectoryEvaluator:
         # would overflow its providers' context the same way; chunked
         # ranking preserves the forced-ranking semantics within chunks.
         self.max_comparative_group = 8
+        # split comparative mode: one parallel critique call per sibling
+        # (same full-context prompt — shared KV prefix makes the repeated
+        # prefill free) plus one compact ranking-only call. Cuts the
+        # score phase's sequential decode depth ~4x vs the single
+        # combined guided generation (round-1 VERDICT weak #5).
+        self.comparative_split = comparative_split
 
     def set_research_context(self, context: Optional[str]) -> None:
         self.deep_research_context = context
@@ -196,6 +203,9 @@ class TrajectoryEvaluator:
             }
             for node in group
         ]
+        if self.comparative_split and len(group) > 1:
+            return await self._judge_group_split(group, trajectories)
+
         system, user = prompts.comparative_trajectory_judge(
             conversation_goal=self.goal,
             trajectories=trajectories,
@@ -209,7 +219,51 @@ class TrajectoryEvaluator:
 
         if not result or "ranking" not in result:
             return await self._fallback_absolute(group)
+        return self._apply_ranking(group, result)
 
+    async def _judge_group_split(self, group: list, trajectories: list) -> dict:
+        """Split comparative judging: n parallel critique calls + one
+        ranking-only call, all sharing the (goal + trajectories) prompt
+        prefix so only one real prefill hits the engine."""
+        crit_prompts = [
+            prompts.comparative_critique_judge(
+                conversation_goal=self.goal,
+                trajectories=trajectories,
+                target_id=node.id,
+                deep_research_context=self.deep_research_context,
+            )
+            for node in group
+        ]
+        rank_prompt = prompts.comparative_ranking_judge(
+            conversation_goal=self.goal,
+            trajectories=trajectories,
+            deep_research_context=self.deep_research_context,
+        )
+        results = await asyncio.gather(
+            *[self._call_json(s, u) for s, u in crit_prompts],
+            self._call_json(*rank_prompt),
+            return_exceptions=True,
+        )
+        rank_res = results[-1]
+        if (
+            isinstance(rank_res, Exception)
+            or not isinstance(rank_res, dict)
+            or "ranking" not in rank_res
+        ):
+            logger.warning("Split ranking failed; absolute fallback")
+            return await self._fallback_absolute(group)
+        critiques: dict = {}
+        for node, r in zip(group, results):
+            if isinstance(r, dict):
+                critiques[node.id] = r
+        merged = {
+            "ranking": rank_res.get("ranking"),
+            "critiques": critiques,
+            "ranking_confidence": rank_res.get("ranking_confidence"),
+        }
+        return self._apply_ranking(group, merged)
+
+    def _apply_ranking(self, group: list, result: dict) -> dict:
         scores: dict = {}
         critiques = result.get("critiques", {}) or {}
         for entry in result.get("ranking", []):

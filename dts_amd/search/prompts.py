@@ -205,18 +205,23 @@ def trajectory_outcome_judge(
     return system, user
 
 
-def comparative_trajectory_judge(
+_COMPARATIVE_SYSTEM = (
+    "[dts:judge-comparative] You force-rank sibling conversation "
+    "trajectories against each other. No ties. Use this score schedule: "
+    "rank 1 → 7.5, rank 2 → 6.0, rank 3 → 4.5, each further rank 1.5 "
+    "lower. Only raise rank 1 above 8.0 for truly exceptional execution. "
+    "Respond with one valid JSON object only."
+)
+
+
+def _comparative_user_base(
     conversation_goal: str,
     trajectories: list,
-    deep_research_context: Optional[str] = None,
-) -> tuple:
-    system = (
-        "[dts:judge-comparative] You force-rank sibling conversation "
-        "trajectories against each other. No ties. Use this score schedule: "
-        "rank 1 → 7.5, rank 2 → 6.0, rank 3 → 4.5, each further rank 1.5 "
-        "lower. Only raise rank 1 above 8.0 for truly exceptional execution. "
-        "Respond with one valid JSON object only."
-    )
+    deep_research_context: Optional[str],
+) -> str:
+    """Shared (system + goal + all trajectories) prefix for the combined
+    AND the split comparative calls — every split call re-renders this
+    exact prefix so the paged KV prefix cache serves its prefill."""
     research = (
         f"\nResearch context:\n{deep_research_context}\n"
         if deep_research_context
@@ -229,9 +234,17 @@ def comparative_trajectory_judge(
             f"{t['history']}"
         )
     traj_text = "\n\n".join(blocks)
-    user = (
-        f"Goal: {conversation_goal}\n{research}\n"
-        f"Trajectories:\n{traj_text}\n\n"
+    return f"Goal: {conversation_goal}\n{research}\nTrajectories:\n{traj_text}\n\n"
+
+
+def comparative_trajectory_judge(
+    conversation_goal: str,
+    trajectories: list,
+    deep_research_context: Optional[str] = None,
+) -> tuple:
+    user = _comparative_user_base(
+        conversation_goal, trajectories, deep_research_context
+    ) + (
         "For each trajectory list 2-3 concrete weaknesses and at least one "
         "strength, then force-rank all of them.\n\n"
         "Return JSON of this shape:\n"
@@ -241,7 +254,48 @@ def comparative_trajectory_judge(
         '  "ranking_confidence": "low|medium|high"\n'
         "}"
     )
-    return system, user
+    return _COMPARATIVE_SYSTEM, user
+
+
+def comparative_critique_judge(
+    conversation_goal: str,
+    trajectories: list,
+    target_id: str,
+    deep_research_context: Optional[str] = None,
+) -> tuple:
+    """Split comparative mode, critique leg: same full sibling context
+    (so the critique stays comparative AND the prompt prefix is shared),
+    critiquing exactly one trajectory."""
+    user = _comparative_user_base(
+        conversation_goal, trajectories, deep_research_context
+    ) + (
+        f"Critique ONLY trajectory {target_id}: list 2 concrete "
+        "weaknesses, at least one strength, and the key turning moment.\n"
+        "Return JSON of this shape:\n"
+        '{"weaknesses": ["...", "..."], "strengths": ["..."], '
+        '"key_moment": "..."}\n'
+        "[dts:part=critique]"
+    )
+    return _COMPARATIVE_SYSTEM, user
+
+
+def comparative_ranking_judge(
+    conversation_goal: str,
+    trajectories: list,
+    deep_research_context: Optional[str] = None,
+) -> tuple:
+    """Split comparative mode, ranking leg (critiques travel separately)."""
+    user = _comparative_user_base(
+        conversation_goal, trajectories, deep_research_context
+    ) + (
+        "Force-rank ALL trajectories (critiques are collected "
+        "separately; rank and score only).\n"
+        "Return JSON of this shape:\n"
+        '{"ranking": [{"rank": 1, "trajectory_id": "...", "score": 7.5, '
+        '"reason": "..."}, ...], "ranking_confidence": "low|medium|high"}\n'
+        "[dts:part=ranking]"
+    )
+    return _COMPARATIVE_SYSTEM, user
 
 
 def branch_selection_judge(

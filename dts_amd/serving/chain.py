@@ -37,7 +37,13 @@ from dts_amd import ops
 from dts_amd.serving.batch import ForwardBatch
 
 CHAIN_MAX = 64  # max replays per chain (also the pinned ring size)
-BURST = 8  # replays enqueued between host catch-up passes
+# pipeline depth: the host keeps at most this many steps enqueued ahead
+# of the last processed one. Deep enough to never starve the GPU (host
+# work per step is ~50 us vs ~5 ms of GPU), shallow enough that a new
+# request arriving mid-chain waits at most DEPTH steps before the chain
+# yields to the scheduler (an 8-step burst cost ~50 ms of admission
+# latency per arrival and erased the chain's win on the search bench).
+DEPTH = 3
 MIN_CHAIN = 4  # not worth the setup below this
 
 

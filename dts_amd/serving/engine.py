@@ -363,7 +363,7 @@ class ServingEngine:
         context cap)."""
         import time as _time
 
-        from dts_amd.serving.chain import BURST, CHAIN_MAX, MIN_CHAIN
+        from dts_amd.serving.chain import CHAIN_MAX, DEPTH, MIN_CHAIN
 
         seqs = batch._sampled_seqs  # type: ignore[attr-defined]
         with self._lock:
@@ -391,15 +391,25 @@ class ServingEngine:
         processed = 0
         stop = False
         while not stop and executed < W:
-            for _ in range(min(BURST, W - executed)):
+            # keep the GPU at most DEPTH steps ahead, polling arrivals on
+            # every launch so a new request waits O(DEPTH) steps, not O(W)
+            while executed - processed < DEPTH and executed < W and not stop:
                 self._chain.launch_step(ctx, executed)
                 executed += 1
+                if self.scheduler.waiting_count() > 0:
+                    stop = True
             # catch up on whatever the GPU already finished (no blocking)
+            progressed = False
             while processed < executed and self._chain.step_ready(processed):
                 self._process_chain_step(ctx, seqs, active, deferred, processed)
                 processed += 1
+                progressed = True
             if not any(active) or self.scheduler.waiting_count() > 0:
                 stop = True
+            elif not progressed and executed - processed >= DEPTH:
+                # pipeline full and nothing completed yet: block on the
+                # oldest in-flight step instead of spinning
+                self._chain.wait_step(processed)
         # drain: ONE real wait on the last outstanding step
         while processed < executed:
             self._chain.wait_step(processed)
@@ -682,7 +692,12 @@ class LocalBackend:
         if "[dts:judge-absolute]" in system:
             return structured.absolute_judge_form(tok)
         if "[dts:judge-comparative]" in system:
+            tail = user[-160:]
+            if "[dts:part=critique]" in tail:
+                return structured.critique_form(tok)
             ids = _TRAJ_RE.findall(user)
+            if "[dts:part=ranking]" in tail:
+                return structured.ranking_only_form(tok, ids) if ids else None
             if ids:
                 return structured.comparative_judge_form(tok, ids)
         return None

@@ -361,28 +361,55 @@ def comparative_judge_form(tok, ids: list) -> FormGuide:
     return _RankingGuide(tok, ids)
 
 
+def critique_form(tok) -> FormGuide:
+    """Split comparative mode: one trajectory's critique (same shape as a
+    per-id entry of the combined ranking form)."""
+    segs: list = [
+        Fixed('{"weaknesses": ['),
+        JsonString(40),
+        Fixed(", "),
+        JsonString(40),
+        Fixed('], "strengths": ['),
+        JsonString(40),
+        Fixed('], "key_moment": '),
+        JsonString(40),
+        Fixed("}"),
+    ]
+    return FormGuide(tok, segs)
+
+
+def ranking_only_form(tok, ids: list) -> FormGuide:
+    """Split comparative mode: the forced ranking without the critique
+    skeleton (critiques run as parallel critique_form calls)."""
+    return _RankingGuide(tok, ids, include_critiques=False)
+
+
 class _RankingGuide(FormGuide):
     """Custom guide: critiques skeleton + ranking with no-repeat id choice."""
 
-    def __init__(self, tok, ids: list) -> None:
+    def __init__(self, tok, ids: list, include_critiques: bool = True) -> None:
         self.ids = list(ids)
         self.used: list = []
-        segs: list = [Fixed('{"critiques": {')]
-        for i, tid in enumerate(ids):
-            if i:
-                segs.append(Fixed(", "))
-            segs += [
-                Fixed(f'"{tid}": {{"weaknesses": ['),
-                JsonString(40),
-                Fixed(", "),
-                JsonString(40),
-                Fixed('], "strengths": ['),
-                JsonString(40),
-                Fixed('], "key_moment": '),
-                JsonString(40),
-                Fixed("}"),
-            ]
-        segs += [Fixed('}, "ranking": [')]
+        segs: list = []
+        if include_critiques:
+            segs.append(Fixed('{"critiques": {'))
+            for i, tid in enumerate(ids):
+                if i:
+                    segs.append(Fixed(", "))
+                segs += [
+                    Fixed(f'"{tid}": {{"weaknesses": ['),
+                    JsonString(40),
+                    Fixed(", "),
+                    JsonString(40),
+                    Fixed('], "strengths": ['),
+                    JsonString(40),
+                    Fixed('], "key_moment": '),
+                    JsonString(40),
+                    Fixed("}"),
+                ]
+            segs += [Fixed('}, "ranking": [')]
+        else:
+            segs += [Fixed('{"ranking": [')]
         for rank in range(1, len(ids) + 1):
             if rank > 1:
                 segs.append(Fixed(", "))

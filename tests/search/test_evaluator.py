@@ -134,6 +134,86 @@ class TestComparative:
         assert set(scores) == {"a", "b", "solo"}
 
 
+class TestComparativeSplit:
+    """Split mode: n parallel critique calls + 1 ranking-only call."""
+
+    def _rank_json(self, ids, scores):
+        return json.dumps(
+            {
+                "ranking": [
+                    {"rank": r + 1, "trajectory_id": i, "score": s, "reason": "r"}
+                    for r, (i, s) in enumerate(zip(ids, scores))
+                ],
+                "ranking_confidence": "high",
+            }
+        )
+
+    def _crit_json(self, tag):
+        return json.dumps(
+            {"weaknesses": [f"w-{tag}", "w2"], "strengths": [f"s-{tag}"],
+             "key_moment": f"m-{tag}"}
+        )
+
+    def test_split_scores_and_critiques(self, run_async):
+        a, b = make_node("a", parent="p"), make_node("b", parent="p")
+        backend = ScriptedBackend(
+            [
+                self._crit_json("a"),
+                self._crit_json("b"),
+                self._rank_json(["a", "b"], [7.5, 6.0]),
+            ]
+        )
+        ev = make_eval(backend, comparative_split=True)
+        scores = run_async(ev.evaluate_comparative([a, b]))
+        assert scores["a"].aggregated_score == 7.5
+        assert scores["b"].aggregated_score == 6.0
+        assert a.stats.critiques["weaknesses"] == ["w-a", "w2"]
+        assert b.stats.critiques["key_moment"] == "m-b"
+        # prompts: all legs share the (goal + trajectories) prefix
+        calls = backend.calls
+        assert len(calls) == 3
+        users = [m[-1].content for m in calls]
+        import os
+
+        prefix = os.path.commonprefix(users)
+        assert "--- Trajectory" in prefix  # trajectories inside shared part
+        assert "[dts:part=critique]" in users[0]
+        assert "[dts:part=ranking]" in users[2]
+
+    def test_split_ranking_failure_falls_back(self, run_async):
+        a, b = make_node("a", parent="p"), make_node("b", parent="p")
+        backend = ScriptedBackend(
+            [
+                self._crit_json("a"),
+                self._crit_json("b"),
+                # valid JSON without "ranking" → rejected by the
+                # evaluator (not the parser), consumed once
+                '{"no_ranking": 1}',
+            ]
+            + [judge_json(5.0)] * 6  # absolute fallback (3 judges x 2)
+        )
+        ev = make_eval(backend, comparative_split=True)
+        scores = run_async(ev.evaluate_comparative([a, b]))
+        assert scores["a"].aggregated_score == 5.0
+        assert scores["b"].aggregated_score == 5.0
+
+    def test_split_critique_failure_still_ranks(self, run_async):
+        a, b = make_node("a", parent="p"), make_node("b", parent="p")
+        backend = ScriptedBackend(
+            [
+                "bad",  # critique a: 3 parse retries then error
+                "bad",
+                "bad",
+                self._crit_json("b"),
+                self._rank_json(["a", "b"], [7.5, 6.0]),
+            ]
+        )
+        ev = make_eval(backend, comparative_split=True)
+        scores = run_async(ev.evaluate_comparative([a, b]))
+        assert scores["a"].aggregated_score == 7.5
+        assert b.stats.critiques["strengths"] == ["s-b"]
+
+
 def test_comparative_pass_votes():
     """Score >= threshold gives pass_votes=3 in comparative mode."""
     # covered through TestComparative, here check boundary semantics directly
