@@ -58,6 +58,7 @@ struct Seq {
   int64_t arrival = 0;
   Status status = Status::WAITING;
   uint64_t prompt_key = 0;
+  uint64_t prefix_key = 0;  // hash of first min(512, len) prompt tokens
   // --- prompt-lookup speculative decoding (dts_amd/serving/spec.py is
   // the Python reference twin): bigram -> (latest, prev) continuation
   // positions over this sequence's own tokens
@@ -127,6 +128,8 @@ class CoreScheduler {
     s.arrival = arrival_++;
     s.allow_spec = allow_spec;
     s.prompt_key = chunk_hash(0xABCD, tokens.data(), (int)tokens.size());
+    s.prefix_key =
+        chunk_hash(0xBEEF, tokens.data(), std::min<int>(512, (int)tokens.size()));
     if (spec_k_ > 0 && s.allow_spec) s.index_tokens();
     seqs_.emplace(seq_id, std::move(s));
     waiting_.push_back(seq_id);
@@ -411,18 +414,26 @@ class CoreScheduler {
 
   // ---- admission ----------------------------------------------------------
   void admit() {
-    // duplicate-prefill holdback (identical prompts of in-flight prefills)
+    // duplicate-prefill holdback (identical prompts of in-flight
+    // prefills) + shared-prefix holdback: requests whose first 512
+    // prompt tokens match an in-flight prefill wait for it, then admit
+    // into a prefix-cache hit instead of re-prefilling the shared prefix
     std::unordered_map<uint64_t, int> inflight;
+    std::unordered_map<uint64_t, int> inflight_prefix;
     for (int64_t sid : running_) {
       Seq& s = seqs_.at(sid);
-      if (s.num_computed < s.num_prompt) inflight[s.prompt_key]++;
+      if (s.num_computed < s.num_prompt) {
+        inflight[s.prompt_key]++;
+        inflight_prefix[s.prefix_key]++;
+      }
     }
     std::deque<int64_t> held;
     while (!waiting_.empty() && (int)running_.size() < max_running_) {
       int64_t sid = waiting_.front();
       waiting_.pop_front();
       Seq& s = seqs_.at(sid);
-      if (inflight.count(s.prompt_key)) {
+      if (inflight.count(s.prompt_key) ||
+          (s.num_prompt >= 512 && inflight_prefix.count(s.prefix_key))) {
         held.push_back(sid);
         continue;
       }
@@ -445,7 +456,10 @@ class CoreScheduler {
       cache_miss_tokens += (int64_t)s.tokens.size() - s.num_computed;
       s.status = Status::RUNNING;
       running_.push_back(sid);
-      if (s.num_computed < s.num_prompt) inflight[s.prompt_key]++;
+      if (s.num_computed < s.num_prompt) {
+        inflight[s.prompt_key]++;
+        inflight_prefix[s.prefix_key]++;
+      }
     }
     while (!held.empty()) {
       waiting_.push_front(held.back());

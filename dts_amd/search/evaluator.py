@@ -18,6 +18,7 @@ import asyncio
 from typing import Any, Callable, Optional
 
 from dts_amd.llm.backend import LLM
+from dts_amd.llm.errors import ContextLengthError
 from dts_amd.llm.types import Message
 from dts_amd.search import prompts
 from dts_amd.search.aggregator import aggregate_majority_vote
@@ -28,6 +29,24 @@ from dts_amd.search.types import (
     format_message_history,
 )
 from dts_amd.utils.logging import log_phase, logger
+
+
+def _shrink(text: str, frac: float) -> str:
+    """Middle-out truncation: keep the opening and the (decisive) tail.
+
+    Local models have a hard context; a judge prompt that cannot fit is
+    retried with shrunk trajectories instead of silently scoring 0.0
+    (the reference just surfaces the provider's HTTP 400 —
+    ref client.py:441-442 — because remote contexts are effectively
+    unbounded for chat-scale inputs)."""
+    if frac >= 1.0 or len(text) < 256:
+        return text
+    n = max(128, int(len(text) * frac))
+    head = n // 3
+    tail = n - head
+    return (
+        text[:head] + "\n... [conversation truncated to fit context] ...\n" + text[-tail:]
+    )
 
 
 class TrajectoryEvaluator:
@@ -134,15 +153,23 @@ class TrajectoryEvaluator:
     # ------------------------------------------------------------------
     async def _judge_single(self, node: DialogueNode):
         history_str = format_message_history(node.messages)
-        system, user = prompts.trajectory_outcome_judge(
-            conversation_goal=self.goal,
-            conversation_history=history_str,
-            deep_research_context=self.deep_research_context,
-        )
-        results = await asyncio.gather(
-            *[self._call_json(system, user) for _ in range(3)],
-            return_exceptions=True,
-        )
+        results: list = []
+        for frac in (1.0, 0.5, 0.25):
+            system, user = prompts.trajectory_outcome_judge(
+                conversation_goal=self.goal,
+                conversation_history=_shrink(history_str, frac),
+                deep_research_context=self.deep_research_context,
+            )
+            results = await asyncio.gather(
+                *[self._call_json(system, user) for _ in range(3)],
+                return_exceptions=True,
+            )
+            if not all(isinstance(r, ContextLengthError) for r in results):
+                break
+            logger.warning(
+                "Judge prompt exceeds context; retrying at %.0f%% history",
+                frac * 50,
+            )
 
         scores: list = []
         judge_results: list = []
@@ -195,31 +222,48 @@ class TrajectoryEvaluator:
 
     async def _judge_group_comparative(self, parent_id: str, group: list) -> dict:
         log_phase("JUDGE", f"Ranking {len(group)} siblings...", indent=1)
-        trajectories = [
-            {
-                "id": node.id,
-                "intent_label": node.user_intent.label if node.user_intent else "unknown",
-                "history": format_message_history(node.messages),
-            }
-            for node in group
-        ]
-        if self.comparative_split and len(group) > 1:
-            return await self._judge_group_split(group, trajectories)
-
-        system, user = prompts.comparative_trajectory_judge(
-            conversation_goal=self.goal,
-            trajectories=trajectories,
-            deep_research_context=self.deep_research_context,
-        )
-        try:
-            result = await self._call_json(system, user)
-        except Exception as e:  # noqa: BLE001
-            logger.warning("Comparative judge errored (%s); absolute fallback", e)
-            result = None
-
-        if not result or "ranking" not in result:
-            return await self._fallback_absolute(group)
-        return self._apply_ranking(group, result)
+        # the group prompt embeds EVERY sibling trajectory — on a small
+        # context it is the first thing to overflow, so retry the whole
+        # group judgment at shrinking history fractions before giving up
+        for frac in (1.0, 0.5, 0.25):
+            trajectories = [
+                {
+                    "id": node.id,
+                    "intent_label": (
+                        node.user_intent.label if node.user_intent else "unknown"
+                    ),
+                    "history": _shrink(format_message_history(node.messages), frac),
+                }
+                for node in group
+            ]
+            try:
+                if self.comparative_split and len(group) > 1:
+                    return await self._judge_group_split(group, trajectories)
+                system, user = prompts.comparative_trajectory_judge(
+                    conversation_goal=self.goal,
+                    trajectories=trajectories,
+                    deep_research_context=self.deep_research_context,
+                )
+                try:
+                    result = await self._call_json(system, user)
+                except ContextLengthError:
+                    raise
+                except Exception as e:  # noqa: BLE001
+                    logger.warning(
+                        "Comparative judge errored (%s); absolute fallback", e
+                    )
+                    result = None
+                if not result or "ranking" not in result:
+                    return await self._fallback_absolute(group)
+                return self._apply_ranking(group, result)
+            except ContextLengthError:
+                logger.warning(
+                    "Group ranking prompt exceeds context; retrying at "
+                    "%.0f%% history",
+                    frac * 50,
+                )
+                continue
+        return await self._fallback_absolute(group)
 
     async def _judge_group_split(self, group: list, trajectories: list) -> dict:
         """Split comparative judging: n parallel critique calls + one
@@ -245,6 +289,8 @@ class TrajectoryEvaluator:
             return_exceptions=True,
         )
         rank_res = results[-1]
+        if isinstance(rank_res, ContextLengthError):
+            raise rank_res  # caller retries the group at a smaller frac
         if (
             isinstance(rank_res, Exception)
             or not isinstance(rank_res, dict)

@@ -226,11 +226,6 @@ class ServingEngine:
         guide: Optional[object] = None,
         stream_cb: Optional[object] = None,
     ) -> concurrent.futures.Future:
-        if len(prompt_ids) + 16 > self.spec.max_position:
-            raise ContextLengthError(
-                f"prompt of {len(prompt_ids)} tokens exceeds max_position "
-                f"{self.spec.max_position}"
-            )
         if guide is not None and hasattr(guide, "token_budget"):
             # a finite form defines its own output size; never let a
             # free-text phase budget truncate it mid-form
@@ -239,6 +234,21 @@ class ServingEngine:
                 import dataclasses
 
                 params = dataclasses.replace(params, max_tokens=need)
+            # context guard must cover the WHOLE form: forced segments
+            # extend the sequence in multi-token chunks, so a prompt that
+            # fits but whose form does not would push positions past the
+            # rope table mid-generation (observed as an index crash at
+            # max_position on the llama-tiny CPU rehearsal)
+            if len(prompt_ids) + need + 8 > self.spec.max_position:
+                raise ContextLengthError(
+                    f"prompt of {len(prompt_ids)} tokens + structured form "
+                    f"of {need} exceeds max_position {self.spec.max_position}"
+                )
+        elif len(prompt_ids) + 16 > self.spec.max_position:
+            raise ContextLengthError(
+                f"prompt of {len(prompt_ids)} tokens exceeds max_position "
+                f"{self.spec.max_position}"
+            )
         seq = Sequence(tokens=list(prompt_ids), params=params)
         seq.guide = guide
         seq.stream_cb = stream_cb

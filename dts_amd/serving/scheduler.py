@@ -94,6 +94,15 @@ class Scheduler:
     def _prompt_key(self, seq: Sequence) -> int:
         return hash(tuple(seq.tokens[: seq.num_prompt_tokens]))
 
+    def _prefix_key(self, seq: Sequence) -> int:
+        """First-512-token key: requests sharing a long prompt prefix
+        (e.g. the n+1 split-judge calls over one sibling group) are held
+        back while one of them prefills, then admit into a prefix-cache
+        hit instead of all prefilling the shared 10k tokens in parallel.
+        Branch rollout prompts diverge well before 512 tokens (strategy
+        text), so this only serializes genuinely shared prefixes."""
+        return hash(tuple(seq.tokens[: min(512, seq.num_prompt_tokens)]))
+
     def has_work(self) -> bool:
         return bool(self.waiting or self.running)
 
@@ -177,16 +186,20 @@ class Scheduler:
     def schedule(self) -> Optional[ForwardBatch]:
         # admit — hold back requests whose exact prompt is already being
         # prefilled by a running sequence (duplicate-prefill dedup)
-        inflight_prefills = {
-            self._prompt_key(s)
-            for s in self.running
-            if s.num_computed < s.num_prompt_tokens
-        }
+        inflight_prefills = set()
+        inflight_prefixes = set()
+        for s in self.running:
+            if s.num_computed < s.num_prompt_tokens:
+                inflight_prefills.add(self._prompt_key(s))
+                inflight_prefixes.add(self._prefix_key(s))
         held: list = []
         while self.waiting and len(self.running) < self.max_running:
             seq = self.waiting.popleft()
             key = self._prompt_key(seq)
-            if key in inflight_prefills:
+            pkey = self._prefix_key(seq)
+            if key in inflight_prefills or (
+                seq.num_prompt_tokens >= 512 and pkey in inflight_prefixes
+            ):
                 held.append(seq)
                 continue
             if not self._admit(seq):
@@ -200,6 +213,7 @@ class Scheduler:
             self.running.append(seq)
             if seq.num_computed < seq.num_prompt_tokens:
                 inflight_prefills.add(key)
+                inflight_prefixes.add(pkey)
         for seq in reversed(held):
             self.waiting.appendleft(seq)
 
