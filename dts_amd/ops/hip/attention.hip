@@ -272,7 +272,277 @@ attn_decode_combine(short* __restrict__ out,        // [B, Hq, D]
 }
 
 // ---------------------------------------------------------------------------
-// Prefill (MFMA)
+// Prefill (MFMA) v2 — KSTEP=64, reg-staged K/V pipeline, transposed V LDS
+//
+// v1 (below) measured 137/180/220 TF at N=2k/4k/8k; its three structural
+// costs, per the guide's attention ladder: (1) KSTEP=32 pays two full
+// barriers per 32 keys, (2) the K/V tile load is synchronous — ~500
+// cycles of HBM latency exposed per tile, (3) the P·V B-fragment reads V
+// columns as 64 scalar ds_read_u16 per lane per tile. v2: 64-key tiles
+// halve the barrier rate; the NEXT tile's K/V global loads issue into
+// registers before the current tile's compute (T14 issue-early /
+// write-late) so HBM latency hides under the MFMAs; V stores into a
+// transposed [D][KSTEP] LDS image (row stride 68 elements: 8-byte
+// aligned for b64 reads, 4-distinct-bank store pattern) so each P·V
+// B-fragment is two contiguous ds_read_b64 instead of eight u16 gathers.
+// ---------------------------------------------------------------------------
+
+template <int G, int D>
+__global__ void __launch_bounds__(256)
+attn_prefill_kernel_v2(short* __restrict__ out,      // [Tq, Hq, D]
+                       const short* __restrict__ q,  // [Tq, Hq, D]
+                       const int* __restrict__ cu_q, // [P+1]
+                       const long* __restrict__ q_pos, // [Tq]
+                       const short* __restrict__ kcache,
+                       const short* __restrict__ vcache,
+                       const int* __restrict__ block_tables,
+                       const int* __restrict__ kv_lens, int max_blocks,
+                       int Hkv, float scale, long q_tstride) {
+  constexpr int BS = 16;
+  constexpr int KSTEP = 64;           // 4 pages per tile
+  constexpr int KCHUNKS = D / 32;     // mfma k-chunks per QK^T
+  constexpr int CTILES = D / 16;      // 16-col output tiles
+  constexpr int LDK = D + 8;          // K row stride (elements)
+  constexpr int LDV = KSTEP + 4;      // transposed-V row stride: 68
+  constexpr int ROWS = 64;
+  const int POS_PER_WG = ROWS / G;
+
+  const int seq = blockIdx.y;
+  const int kvh = blockIdx.z;
+  const int Hq = Hkv * G;
+  const int q_start = cu_q[seq];
+  const int q_len = cu_q[seq + 1] - q_start;
+  const int tile = blockIdx.x;
+  if (tile * POS_PER_WG >= q_len) return;
+
+  const int wave = threadIdx.x / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int kv_len = kv_lens[seq];
+  const int* bt = block_tables + (long)seq * max_blocks;
+
+  __shared__ short k_lds[KSTEP][LDK];
+  __shared__ short v_lds_t[D][LDV];   // transposed: [dim][key]
+  __shared__ float s_scores[4][16][KSTEP];
+  __shared__ float s_alpha[4][16];
+  __shared__ float s_rowl[4][16];
+  __shared__ short p_lds[4][16][KSTEP];
+
+  // Q A-fragments (persistent)
+  short a_frag[KCHUNKS][8];
+  {
+    const int r = lane & 15;
+    const int row_global = wave * 16 + r;
+    const int pos_local = tile * POS_PER_WG + row_global / G;
+    const int head = row_global % G;
+    const bool valid_row = pos_local < q_len;
+    const int tok = q_start + (valid_row ? pos_local : 0);
+    const short* qp = q + (long)tok * q_tstride + (long)(kvh * G + head) * D;
+#pragma unroll
+    for (int kc = 0; kc < KCHUNKS; ++kc) {
+      const int kbase = kc * 32 + (lane >> 4) * 8;
+      bf16x8 v8 = *(const bf16x8*)(qp + kbase);
+#pragma unroll
+      for (int i = 0; i < 8; ++i)
+        a_frag[kc][i] = valid_row ? v8[i] : (short)0;
+    }
+  }
+  const int sm_row = lane / 4;
+  const int sm_sub = lane & 3;  // scans KSTEP/4 = 16 keys
+  const long sm_pos = tile * POS_PER_WG + (wave * 16 + sm_row) / G;
+  const bool sm_valid = sm_pos < q_len;
+  const long sm_abs_pos = sm_valid ? q_pos[q_start + sm_pos] : -1;
+
+  f32x4 o_acc[CTILES];
+#pragma unroll
+  for (int ct = 0; ct < CTILES; ++ct) o_acc[ct] = {0.f, 0.f, 0.f, 0.f};
+  float run_m = -1e30f, run_l = 0.f;
+
+  const int last_local_pos = min(q_len, tile * POS_PER_WG + POS_PER_WG) - 1;
+  const long last_abs_pos = q_pos[q_start + last_local_pos];
+  const int kv_hi = min((long)kv_len, last_abs_pos + 1);
+
+  // ---- reg-staged tile loads: each thread owns 4 (key, d0) chunks of 8
+  // bf16 for K and the same for V (KSTEP*D / 256 threads / 8 = 4)
+  constexpr int CHUNKS_PT = KSTEP * D / (256 * 8);
+  bf16x8 k_stage[CHUNKS_PT], v_stage[CHUNKS_PT];
+  int st_key[CHUNKS_PT], st_d0[CHUNKS_PT];
+
+  auto issue_tile = [&](int kv_base) {
+#pragma unroll
+    for (int c = 0; c < CHUNKS_PT; ++c) {
+      const int base = (threadIdx.x + c * 256) * 8;
+      const int key = base / D;
+      const int d0 = base % D;
+      st_key[c] = key;
+      st_d0[c] = d0;
+      const int kglob = kv_base + key;
+      if (kglob < kv_len) {
+        const long blk = bt[kglob / BS];
+        const long off = ((blk * Hkv + kvh) * BS + kglob % BS) * (long)D + d0;
+        k_stage[c] = *(const bf16x8*)(kcache + off);
+        v_stage[c] = *(const bf16x8*)(vcache + off);
+      } else {
+        const bf16x8 z = {0, 0, 0, 0, 0, 0, 0, 0};
+        k_stage[c] = z;
+        v_stage[c] = z;
+      }
+    }
+  };
+  auto write_tile = [&]() {
+#pragma unroll
+    for (int c = 0; c < CHUNKS_PT; ++c) {
+      *(bf16x8*)(&k_lds[st_key[c]][st_d0[c]]) = k_stage[c];
+#pragma unroll
+      for (int i = 0; i < 8; ++i)
+        v_lds_t[st_d0[c] + i][st_key[c]] = v_stage[c][i];
+    }
+  };
+
+  issue_tile(0);
+  write_tile();
+  __syncthreads();
+
+  for (int kv_base = 0; kv_base < kv_hi; kv_base += KSTEP) {
+    const int kv_next = kv_base + KSTEP;
+    // T14 issue-early: next tile's HBM loads start before this tile's
+    // compute; the LDS write happens after the barrier below
+    if (kv_next < kv_hi) issue_tile(kv_next);
+
+    // ---- QK^T: four 16x16 score tiles over KCHUNKS k-chunks
+#pragma unroll
+    for (int st = 0; st < KSTEP / 16; ++st) {
+      f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+      for (int kc = 0; kc < KCHUNKS; ++kc) {
+        const int key = st * 16 + (lane & 15);
+        const int d0 = kc * 32 + (lane >> 4) * 8;
+        bf16x8 b8 = *(bf16x8*)(&k_lds[key][d0]);
+        acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            *(bf16x8*)a_frag[kc], b8, acc, 0, 0, 0);
+      }
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        const int row = (lane >> 4) * 4 + i;
+        s_scores[wave][row][st * 16 + (lane & 15)] = acc[i] * scale;
+      }
+    }
+    __builtin_amdgcn_s_waitcnt(0);
+    __builtin_amdgcn_wave_barrier();
+
+    // ---- online softmax: 4 lanes per row, each scans 16 keys
+    {
+      float tmax = -1e30f;
+      float sc[16];
+#pragma unroll
+      for (int jj = 0; jj < 16; ++jj) {
+        const int j = sm_sub * 16 + jj;
+        const long key_abs = kv_base + j;
+        float s = s_scores[wave][sm_row][j];
+        const bool ok =
+            sm_valid && key_abs <= sm_abs_pos && key_abs < (long)kv_len;
+        sc[jj] = ok ? s : -1e30f;
+        tmax = fmaxf(tmax, sc[jj]);
+      }
+      tmax = fmaxf(tmax, __shfl_xor(tmax, 1, 64));
+      tmax = fmaxf(tmax, __shfl_xor(tmax, 2, 64));
+      float m_new = fmaxf(run_m, tmax);
+      float alpha, rowsum = 0.f;
+      if (m_new > -1e30f) {
+        alpha = (run_m > -1e30f) ? __expf(run_m - m_new) : 0.f;
+#pragma unroll
+        for (int jj = 0; jj < 16; ++jj) {
+          float p = (sc[jj] > -1e30f) ? __expf(sc[jj] - m_new) : 0.f;
+          p_lds[wave][sm_row][sm_sub * 16 + jj] = f2bf(p);
+          rowsum += p;
+        }
+      } else {
+        alpha = 1.f;
+#pragma unroll
+        for (int jj = 0; jj < 16; ++jj)
+          p_lds[wave][sm_row][sm_sub * 16 + jj] = 0;
+      }
+      rowsum += __shfl_xor(rowsum, 1, 64);
+      rowsum += __shfl_xor(rowsum, 2, 64);
+      run_l = run_l * alpha + rowsum;
+      run_m = m_new;
+      if (sm_sub == 0) {
+        s_alpha[wave][sm_row] = alpha;
+        s_rowl[wave][sm_row] = run_l;
+      }
+    }
+    __builtin_amdgcn_s_waitcnt(0);
+    __builtin_amdgcn_wave_barrier();
+
+    // ---- rescale O and accumulate P*V (two k=32 chunks over 64 keys)
+    {
+      float al[4];
+#pragma unroll
+      for (int i = 0; i < 4; ++i) al[i] = s_alpha[wave][(lane >> 4) * 4 + i];
+      bf16x8 pa[2];
+      {
+        const int row = lane & 15;
+#pragma unroll
+        for (int h = 0; h < 2; ++h) {
+          const int k0 = h * 32 + (lane >> 4) * 8;
+#pragma unroll
+          for (int i = 0; i < 8; ++i) pa[h][i] = p_lds[wave][row][k0 + i];
+        }
+      }
+#pragma unroll
+      for (int ct = 0; ct < CTILES; ++ct) {
+#pragma unroll
+        for (int i = 0; i < 4; ++i) o_acc[ct][i] *= al[i];
+        const int dim = ct * 16 + (lane & 15);
+#pragma unroll
+        for (int h = 0; h < 2; ++h) {
+          // V B-frag: 8 consecutive keys of one dim = 2x ds_read_b64
+          // (the 68-element row stride is 8-byte aligned, not 16)
+          const int kk0 = h * 32 + (lane >> 4) * 8;
+          bf16x8 vb;
+          ((unsigned long long*)&vb)[0] =
+              *(const unsigned long long*)(&v_lds_t[dim][kk0]);
+          ((unsigned long long*)&vb)[1] =
+              *(const unsigned long long*)(&v_lds_t[dim][kk0 + 4]);
+          o_acc[ct] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              pa[h], vb, o_acc[ct], 0, 0, 0);
+        }
+      }
+    }
+    __syncthreads();  // everyone done READING this tile's LDS
+    if (kv_next < kv_hi) {
+      write_tile();   // T14 write-late: stage the next tile
+      __syncthreads();
+    }
+  }
+
+  // ---- epilogue: normalize + store
+  {
+    float invl[4];
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      float l = s_rowl[wave][(lane >> 4) * 4 + i];
+      invl[i] = (l > 0.f) ? 1.f / l : 0.f;
+    }
+#pragma unroll
+    for (int ct = 0; ct < CTILES; ++ct) {
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        const int row = (lane >> 4) * 4 + i;
+        const int row_global = wave * 16 + row;
+        const int pos_local = tile * POS_PER_WG + row_global / G;
+        if (pos_local >= q_len) continue;
+        const int head = row_global % G;
+        const int tok = q_start + pos_local;
+        const int dim = ct * 16 + (lane & 15);
+        out[((long)tok * Hq + kvh * G + head) * D + dim] =
+            f2bf(o_acc[ct][i] * invl[i]);
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Prefill (MFMA) v1 — KSTEP=32 reference implementation
 // ---------------------------------------------------------------------------
 
 // Rows per workgroup: 64 = (64/G) positions x G heads. Each wave owns 16
@@ -605,17 +875,15 @@ void attn_prefill_paged(torch::Tensor out, torch::Tensor q, torch::Tensor cu_q,
                         torch::Tensor q_pos, torch::Tensor kcache,
                         torch::Tensor vcache, torch::Tensor block_tables,
                         torch::Tensor kv_lens, double scale, int64_t swz) {
+  // variant select: -1 (default) -> v2 unless DTS_PREFILL_V1=1;
+  // 0 -> v1 no-swizzle, 1 -> v1 swizzled (measured -24..-30%, kept for
+  // re-runs), 2 -> v2 (KSTEP=64, reg-staged pipeline, transposed-V LDS)
   if (swz < 0) {
-    // default OFF: within-probe A/B (profiles/prefill_swizzle_ab_r01.log)
-    // measured the XOR swizzle at -24..-30% on this kernel — the +8-element
-    // row padding (272 B stride = 68 dwords, coprime-ish with the 64-bank
-    // row) already spreads the b128 lane groups, so the swizzle only adds
-    // per-access address VALU. Kept behind DTS_PREFILL_SWZ=1 for re-runs.
-    static int env_swz = [] {
-      const char* e = getenv("DTS_PREFILL_SWZ");
+    static int env_v1 = [] {
+      const char* e = getenv("DTS_PREFILL_V1");
       return (e && e[0] == '1') ? 1 : 0;
     }();
-    swz = env_swz;
+    swz = env_v1 ? 0 : 2;
   }
   const int Hq = q.size(1), D = q.size(2);
   const int Hkv = kcache.size(1);
@@ -638,7 +906,18 @@ void attn_prefill_paged(torch::Tensor out, torch::Tensor q, torch::Tensor cu_q,
   auto stream = c10::hip::getCurrentHIPStream();
 #define PREFILL_CASE(g, d)                                                    \
   do {                                                                        \
-    if (swz)                                                                  \
+    if (swz == 2)                                                             \
+      hipLaunchKernelGGL((attn_prefill_kernel_v2<g, d>), grid, block, 0,      \
+                         stream, (short*)out.data_ptr(),                      \
+                         (const short*)q.data_ptr(),                          \
+                         (const int*)cu_q.data_ptr(),                         \
+                         (const long*)q_pos.data_ptr(),                       \
+                         (const short*)kcache.data_ptr(),                     \
+                         (const short*)vcache.data_ptr(),                     \
+                         (const int*)block_tables.data_ptr(),                 \
+                         (const int*)kv_lens.data_ptr(), max_blocks, Hkv,     \
+                         (float)scale, q_tstride);                            \
+    else if (swz == 1)                                                        \
       hipLaunchKernelGGL((attn_prefill_kernel<g, d, true>), grid, block, 0,   \
                          stream, (short*)out.data_ptr(),                      \
                          (const short*)q.data_ptr(),                          \

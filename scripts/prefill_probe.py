@@ -34,7 +34,9 @@ def probe(N=4096, Hq=32, Hkv=8, D=128, iters=20):
     flops = 2 * 2 * Hq * (N * N / 2) * D
     results = {}
     # within-probe interleaved A/B (guide §5.4 rule 24): 6 rounds each
-    for swz in (0, 1, 0, 1, 0, 1):
+    # variants: 0 = v1 (KSTEP=32), 2 = v2 (KSTEP=64 + reg-staged
+    # pipeline + transposed-V LDS)
+    for swz in (0, 2, 0, 2, 0, 2):
         for _ in range(2):
             ext.attn_prefill_paged(out, q, cu_q, q_pos, kc, vc, bt, kvl,
                                    scale, swz)
@@ -46,19 +48,20 @@ def probe(N=4096, Hq=32, Hkv=8, D=128, iters=20):
         torch.cuda.synchronize()
         dt = (time.perf_counter() - t0) / iters
         results.setdefault(swz, []).append(round(flops / dt / 1e12, 1))
-    # numerics: both variants must agree bitwise (same math, layout only)
+    # numerics: v2 must match v1 closely (different tile size changes
+    # the online-softmax accumulation order, so not bitwise)
     out0 = torch.empty_like(q)
     out1 = torch.empty_like(q)
     ext.attn_prefill_paged(out0, q, cu_q, q_pos, kc, vc, bt, kvl, scale, 0)
-    ext.attn_prefill_paged(out1, q, cu_q, q_pos, kc, vc, bt, kvl, scale, 1)
-    biteq = bool(torch.equal(out0, out1))
+    ext.attn_prefill_paged(out1, q, cu_q, q_pos, kc, vc, bt, kvl, scale, 2)
+    max_err = float((out0.float() - out1.float()).abs().max())
     print(
         json.dumps(
             {
                 "probe": f"prefill_attn_N{N}",
-                "TF_noswz": results[0],
-                "TF_swz": results[1],
-                "swz_bitexact": biteq,
+                "TF_v1": results[0],
+                "TF_v2": results[2],
+                "v1_v2_max_abs_err": max_err,
             }
         )
     )
