@@ -146,6 +146,22 @@ MODEL_REGISTRY = {
         max_position=2048,
         tie_embeddings=True,
     ),
+    # small-vocab GPU test model: D=64/G=1 runs the real HIP kernels,
+    # and greedy decode on random weights enters short cycles quickly —
+    # which is what the speculative-decode GPU tests need (an 8B-class
+    # random model does not repeat a bigram within a short generation)
+    "llama-mini-gpu": ModelSpec(
+        name="llama-mini-gpu",
+        vocab_size=2048,
+        hidden_size=256,
+        intermediate_size=512,
+        num_layers=2,
+        num_heads=4,
+        num_kv_heads=4,
+        head_dim=64,
+        rope_theta=10000.0,
+        max_position=4096,
+    ),
     "mixtral-tiny": ModelSpec(
         name="mixtral-tiny",
         arch="mixtral",

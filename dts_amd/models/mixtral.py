@@ -65,11 +65,19 @@ class MoEMLP(nn.Module):
         if self.tp.size > 1:
             return self._forward_ep(x, flat_tok, flat_expert, flat_w)
 
+        if x.is_cuda and T * self.top_k <= 64 and ops.hip_available():
+            # capture-safe decode path: sort assignments by expert on
+            # device and run the grouped skinny GEMM with device-side
+            # counts — no nonzero()/masked gather, so the whole step is
+            # hipGraph-capturable (ADVICE round-1 high)
+            return self._forward_grouped(
+                x, flat_tok, flat_expert, weights, T
+            )
+
         out = torch.zeros(T, self.hidden, dtype=torch.float32, device=x.device)
-        # grouped execution: one GEMM per locally-resident expert slice.
+        # prefill path: one GEMM per locally-resident expert slice.
         # No host-side `sel.any()` early-outs: a bool() on a device tensor
-        # is a stream sync per expert per layer AND breaks hipGraph
-        # capture; zero-row GEMMs are free.
+        # is a stream sync per expert per layer; zero-row GEMMs are free.
         for e_local in range(self.experts_local):
             e = self.expert_offset + e_local
             sel = flat_expert == e
@@ -78,6 +86,28 @@ class MoEMLP(nn.Module):
             gu = F.linear(xe, self.gate_up_w[e_local])
             ye = F.linear(ops.silu_mul(gu), self.down_w[e_local])
             out.index_add_(0, toks, ye.float() * flat_w[sel].unsqueeze(1).float())
+        return out.to(x.dtype)
+
+    def _forward_grouped(self, x, flat_tok, flat_expert, weights, T):
+        """Decode MoE via the grouped per-expert MFMA GEMM: every shape
+        is static in T*k, all routing state stays on device."""
+        P = flat_tok.shape[0]
+        order = torch.argsort(flat_expert, stable=True)
+        inv = torch.empty_like(order)
+        inv.scatter_(
+            0, order, torch.arange(P, device=x.device, dtype=order.dtype)
+        )
+        counts = torch.bincount(
+            flat_expert, minlength=self.num_experts
+        ).to(torch.int32)
+        offsets = (torch.cumsum(counts, 0) - counts).to(torch.int32)
+        x_sorted = x.index_select(0, flat_tok.index_select(0, order))
+        gu = ops.moe_grouped_linear(x_sorted, self.gate_up_w, counts, offsets)
+        y = ops.moe_grouped_linear(
+            ops.silu_mul(gu), self.down_w, counts, offsets
+        )
+        y = y.index_select(0, inv).view(T, self.top_k, self.hidden)
+        out = (y.float() * weights.unsqueeze(-1).float()).sum(dim=1)
         return out.to(x.dtype)
 
     def _forward_ep(self, x, flat_tok, flat_expert, flat_w) -> torch.Tensor:
@@ -153,11 +183,15 @@ class MixtralLayer(nn.Module):
 
 class MixtralModel(nn.Module):
     arch = "mixtral"
-    # MoE expert gather uses boolean-mask indexing (nonzero() → stream
-    # sync), which aborts hipGraph capture with
-    # hipErrorStreamCaptureUnsupported; decode runs eager until the
-    # capture-safe indirect grouped GEMV lands.
-    graph_capturable = False
+
+    @property
+    def graph_capturable(self) -> bool:
+        """Single-rank decode routes through the grouped per-expert GEMM
+        (all routing state on device — capture-safe); the EP all-to-all
+        path still computes host-side splits and must stay eager."""
+        from dts_amd import ops as _ops
+
+        return self.tp.size == 1 and _ops.hip_available()
 
     def __init__(
         self,

@@ -193,6 +193,21 @@ def derive_seeds(out, bases, positions):
 mix_seed = torch_ref.mix_seed
 
 
+def moe_grouped_linear(x, w, counts, offsets):
+    """Grouped per-expert GEMM over contiguous row segments (sorted by
+    expert): out[p] = x[p] @ w[expert_of(p)]^T. counts/offsets are DEVICE
+    tensors, so a MoE decode step stays hipGraph-capturable."""
+    if _on_gpu(x):
+        ext = _require_hip()
+        if ext is not None:
+            out = torch.empty(
+                (x.shape[0], w.shape[1]), dtype=x.dtype, device=x.device
+            )
+            ext.moe_grouped_linear(out, x, w, counts, offsets)
+            return out
+    return torch_ref.moe_grouped_linear(x, w, counts, offsets)
+
+
 def linear_bf16(x: torch.Tensor, weight: torch.Tensor) -> torch.Tensor:
     """y = x @ W^T. Skinny decode batches (M<=16, bf16, K%512==0) stream
     through custom weight-streaming kernels; everything else is a

@@ -287,7 +287,7 @@ attn_decode_combine(short* __restrict__ out,        // [B, Hq, D]
 // B-fragment is two contiguous ds_read_b64 instead of eight u16 gathers.
 // ---------------------------------------------------------------------------
 
-template <int G, int D>
+template <int G, int D, int KSTEP>
 __global__ void __launch_bounds__(256)
 attn_prefill_kernel_v2(short* __restrict__ out,      // [Tq, Hq, D]
                        const short* __restrict__ q,  // [Tq, Hq, D]
@@ -299,7 +299,6 @@ attn_prefill_kernel_v2(short* __restrict__ out,      // [Tq, Hq, D]
                        const int* __restrict__ kv_lens, int max_blocks,
                        int Hkv, float scale, long q_tstride) {
   constexpr int BS = 16;
-  constexpr int KSTEP = 64;           // 4 pages per tile
   constexpr int KCHUNKS = D / 32;     // mfma k-chunks per QK^T
   constexpr int CTILES = D / 16;      // 16-col output tiles
   constexpr int LDK = D + 8;          // K row stride (elements)
@@ -346,8 +345,9 @@ attn_prefill_kernel_v2(short* __restrict__ out,      // [Tq, Hq, D]
         a_frag[kc][i] = valid_row ? v8[i] : (short)0;
     }
   }
+  constexpr int SMW = KSTEP / 4;  // keys per softmax lane
   const int sm_row = lane / 4;
-  const int sm_sub = lane & 3;  // scans KSTEP/4 = 16 keys
+  const int sm_sub = lane & 3;
   const long sm_pos = tile * POS_PER_WG + (wave * 16 + sm_row) / G;
   const bool sm_valid = sm_pos < q_len;
   const long sm_abs_pos = sm_valid ? q_pos[q_start + sm_pos] : -1;
@@ -432,10 +432,10 @@ attn_prefill_kernel_v2(short* __restrict__ out,      // [Tq, Hq, D]
     // ---- online softmax: 4 lanes per row, each scans 16 keys
     {
       float tmax = -1e30f;
-      float sc[16];
+      float sc[SMW];
 #pragma unroll
-      for (int jj = 0; jj < 16; ++jj) {
-        const int j = sm_sub * 16 + jj;
+      for (int jj = 0; jj < SMW; ++jj) {
+        const int j = sm_sub * SMW + jj;
         const long key_abs = kv_base + j;
         float s = s_scores[wave][sm_row][j];
         const bool ok =
@@ -450,16 +450,16 @@ attn_prefill_kernel_v2(short* __restrict__ out,      // [Tq, Hq, D]
       if (m_new > -1e30f) {
         alpha = (run_m > -1e30f) ? __expf(run_m - m_new) : 0.f;
 #pragma unroll
-        for (int jj = 0; jj < 16; ++jj) {
+        for (int jj = 0; jj < SMW; ++jj) {
           float p = (sc[jj] > -1e30f) ? __expf(sc[jj] - m_new) : 0.f;
-          p_lds[wave][sm_row][sm_sub * 16 + jj] = f2bf(p);
+          p_lds[wave][sm_row][sm_sub * SMW + jj] = f2bf(p);
           rowsum += p;
         }
       } else {
         alpha = 1.f;
 #pragma unroll
-        for (int jj = 0; jj < 16; ++jj)
-          p_lds[wave][sm_row][sm_sub * 16 + jj] = 0;
+        for (int jj = 0; jj < SMW; ++jj)
+          p_lds[wave][sm_row][sm_sub * SMW + jj] = 0;
       }
       rowsum += __shfl_xor(rowsum, 1, 64);
       rowsum += __shfl_xor(rowsum, 2, 64);
@@ -478,11 +478,12 @@ attn_prefill_kernel_v2(short* __restrict__ out,      // [Tq, Hq, D]
       float al[4];
 #pragma unroll
       for (int i = 0; i < 4; ++i) al[i] = s_alpha[wave][(lane >> 4) * 4 + i];
-      bf16x8 pa[2];
+      constexpr int HCH = KSTEP / 32;
+      bf16x8 pa[HCH];
       {
         const int row = lane & 15;
 #pragma unroll
-        for (int h = 0; h < 2; ++h) {
+        for (int h = 0; h < HCH; ++h) {
           const int k0 = h * 32 + (lane >> 4) * 8;
 #pragma unroll
           for (int i = 0; i < 8; ++i) pa[h][i] = p_lds[wave][row][k0 + i];
@@ -494,7 +495,7 @@ attn_prefill_kernel_v2(short* __restrict__ out,      // [Tq, Hq, D]
         for (int i = 0; i < 4; ++i) o_acc[ct][i] *= al[i];
         const int dim = ct * 16 + (lane & 15);
 #pragma unroll
-        for (int h = 0; h < 2; ++h) {
+        for (int h = 0; h < HCH; ++h) {
           // V B-frag: 8 consecutive keys of one dim = 2x ds_read_b64
           // (the 68-element row stride is 8-byte aligned, not 16)
           const int kk0 = h * 32 + (lane >> 4) * 8;
@@ -879,11 +880,14 @@ void attn_prefill_paged(torch::Tensor out, torch::Tensor q, torch::Tensor cu_q,
   // 0 -> v1 no-swizzle, 1 -> v1 swizzled (measured -24..-30%, kept for
   // re-runs), 2 -> v2 (KSTEP=64, reg-staged pipeline, transposed-V LDS)
   if (swz < 0) {
-    static int env_v1 = [] {
-      const char* e = getenv("DTS_PREFILL_V1");
-      return (e && e[0] == '1') ? 1 : 0;
+    static int env_v = [] {
+      const char* e = getenv("DTS_PREFILL_V");
+      return e ? atoi(e) : -1;
     }();
-    swz = env_v1 ? 0 : 2;
+    // default: v2 with 32-key tiles (v1's LDS budget, 5 WGs/CU, plus
+    // the staged pipeline and transposed-V reads); DTS_PREFILL_V=0
+    // selects v1, 2 the 64-key variant (2 WGs/CU — measured slower)
+    swz = (env_v >= 0) ? env_v : 3;
   }
   const int Hq = q.size(1), D = q.size(2);
   const int Hkv = kcache.size(1);
@@ -907,7 +911,18 @@ void attn_prefill_paged(torch::Tensor out, torch::Tensor q, torch::Tensor cu_q,
 #define PREFILL_CASE(g, d)                                                    \
   do {                                                                        \
     if (swz == 2)                                                             \
-      hipLaunchKernelGGL((attn_prefill_kernel_v2<g, d>), grid, block, 0,      \
+      hipLaunchKernelGGL((attn_prefill_kernel_v2<g, d, 64>), grid, block, 0,  \
+                         stream, (short*)out.data_ptr(),                      \
+                         (const short*)q.data_ptr(),                          \
+                         (const int*)cu_q.data_ptr(),                         \
+                         (const long*)q_pos.data_ptr(),                       \
+                         (const short*)kcache.data_ptr(),                     \
+                         (const short*)vcache.data_ptr(),                     \
+                         (const int*)block_tables.data_ptr(),                 \
+                         (const int*)kv_lens.data_ptr(), max_blocks, Hkv,     \
+                         (float)scale, q_tstride);                            \
+    else if (swz == 3)                                                        \
+      hipLaunchKernelGGL((attn_prefill_kernel_v2<g, d, 32>), grid, block, 0,  \
                          stream, (short*)out.data_ptr(),                      \
                          (const short*)q.data_ptr(),                          \
                          (const int*)cu_q.data_ptr(),                         \

@@ -27,6 +27,8 @@ void derive_seeds(torch::Tensor out, torch::Tensor bases,
                   torch::Tensor positions);
 void gemv_bf16(torch::Tensor out, torch::Tensor x, torch::Tensor w);
 void gemm_skinny_bf16(torch::Tensor out, torch::Tensor x, torch::Tensor w);
+void moe_grouped_linear(torch::Tensor out, torch::Tensor x, torch::Tensor w,
+                        torch::Tensor counts, torch::Tensor offsets);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm", &rmsnorm, "RMSNorm (bf16, CDNA4)");
@@ -52,4 +54,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "skinny-batch (M<=8) bf16 weight-streaming GEMV");
   m.def("gemm_skinny_bf16", &gemm_skinny_bf16,
         "skinny-M (M<=16) bf16 MFMA weight-streaming GEMM");
+  m.def("moe_grouped_linear", &moe_grouped_linear,
+        "grouped per-expert skinny GEMM (capture-safe MoE decode)");
 }

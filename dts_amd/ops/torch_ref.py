@@ -236,6 +236,25 @@ def derive_seeds(
     )
 
 
+def moe_grouped_linear(
+    x: torch.Tensor,  # [P, K]
+    w: torch.Tensor,  # [E, N, K]
+    counts: torch.Tensor,  # [E]
+    offsets: torch.Tensor,  # [E]
+) -> torch.Tensor:
+    """Reference of the grouped per-expert GEMM (capture-safe MoE)."""
+    P, K = x.shape
+    E, N, _ = w.shape
+    out = torch.zeros(P, N, dtype=x.dtype, device=x.device)
+    for e in range(E):
+        c = int(counts[e])
+        if c == 0:
+            continue
+        o = int(offsets[e])
+        out[o : o + c] = (x[o : o + c].float() @ w[e].float().T).to(x.dtype)
+    return out
+
+
 def top_p_sample(
     logits: torch.Tensor,  # [B, V] fp32
     temperatures: torch.Tensor,  # [B]

@@ -1,0 +1,158 @@
+"""TP=2 over real RCCL on ONE MI355X (VERDICT round-1 #2).
+
+Launch (both ranks share the single GPU — valid RCCL same-device comms;
+HSA_ENABLE_IPC_MODE_LEGACY=0 must be exported for dmabuf IPC):
+
+  python -m torch.distributed.run --nnodes=1 --nproc-per-node 2 \
+      --master-addr 127.0.0.1 scripts/tp2_gpu_check.py
+
+Rank 0 first computes a DENSE (tp=1) greedy reference on the same
+weights, then both ranks run the sharded engine: column/row-parallel
+linears with per-layer RCCL all-reduce over the packed-tensor batch
+broadcast (serving/tp_engine.py). Checks sharded == dense token streams
+for free decode AND a guided (constrained-JSON) request, and times the
+per-step overhead. Writes gpurun_out/tp2_check.json.
+"""
+
+import json
+import os
+import sys
+import time
+from pathlib import Path
+
+import torch
+import torch.distributed as dist
+
+sys.path.insert(0, str(Path(__file__).resolve().parent.parent))
+
+OUT = Path("gpurun_out")
+OUT.mkdir(exist_ok=True)
+
+MODEL = os.environ.get("TP2_MODEL", "llama-3-8b-2l")
+
+
+def main():
+    rank = int(os.environ["RANK"])
+    world = int(os.environ["WORLD_SIZE"])
+    assert world == 2
+    torch.cuda.set_device(0)  # both ranks share the one GPU
+    dist.init_process_group("nccl", rank=rank, world_size=world)
+
+    from dts_amd.llm.types import SamplingParams
+    from dts_amd.models.config import get_model_spec
+    from dts_amd.models.llama import LlamaModel
+    from dts_amd.models.weights import load_llama_safetensors, save_llama_safetensors
+    from dts_amd.parallel.tp import TPContext
+    from dts_amd.serving import ServingEngine
+    from dts_amd.serving.kv_cache import KVCachePool
+    from dts_amd.serving.structured import strategy_form
+    from dts_amd.serving.tp_engine import TPDriverMixin, run_tp_worker
+
+    spec = get_model_spec(MODEL)
+    ckpt = Path("/tmp/tp2_ckpt")
+    if rank == 0:
+        dense = LlamaModel(spec, dtype=torch.bfloat16, device="cuda")
+        dense.random_init(seed=77)
+        save_llama_safetensors(dense.to("cpu"), str(ckpt))
+        del dense
+        torch.cuda.empty_cache()
+    dist.barrier()
+
+    prompt = [300 + (i * 13) % 5000 for i in range(64)]
+    guided_prompt = [900 + (i * 7) % 4000 for i in range(48)]
+
+    ref = None
+    if rank == 0:
+        # dense reference on the same checkpoint
+        eng = ServingEngine(
+            model_name=MODEL,
+            device="cuda:0",
+            dtype=torch.bfloat16,
+            kv_memory_bytes=4 << 30,
+            weights_path=str(ckpt),
+        )
+        fut = eng.submit_tokens(
+            list(prompt), SamplingParams(max_tokens=32, temperature=0.0, seed=0)
+        )
+        gfut = eng.submit_tokens(
+            list(guided_prompt),
+            SamplingParams(max_tokens=4096, temperature=0.0, seed=0),
+            guide=strategy_form(eng.tokenizer, 2),
+        )
+        eng.run_until_idle()
+        ref = (fut.result(timeout=120).token_ids, gfut.result(timeout=120).text)
+        eng.stop()
+        del eng
+        torch.cuda.empty_cache()
+    dist.barrier()
+
+    # ---- sharded engine over RCCL
+    tp = TPContext.from_world()
+    model = LlamaModel(spec, tp=tp, dtype=torch.bfloat16, device="cuda")
+    load_llama_safetensors(model, str(ckpt))
+    dist.barrier()
+
+    if rank == 0:
+        eng = ServingEngine(
+            model_name=MODEL,
+            device="cuda:0",
+            dtype=torch.bfloat16,
+            kv_memory_bytes=4 << 30,
+            model=model,
+        )
+        TPDriverMixin.install(eng)
+        t0 = time.time()
+        fut = eng.submit_tokens(
+            list(prompt), SamplingParams(max_tokens=32, temperature=0.0, seed=0)
+        )
+        gfut = eng.submit_tokens(
+            list(guided_prompt),
+            SamplingParams(max_tokens=4096, temperature=0.0, seed=0),
+            guide=strategy_form(eng.tokenizer, 2),
+        )
+        eng.run_until_idle()
+        out = (fut.result(timeout=300).token_ids, gfut.result(timeout=300).text)
+        wall = time.time() - t0
+        # decode-only timing: 64 more tokens on a fresh request
+        t1 = time.time()
+        fut2 = eng.submit_tokens(
+            [p + 1 for p in prompt],
+            SamplingParams(max_tokens=64, temperature=0.0, seed=0),
+        )
+        eng.run_until_idle()
+        fut2.result(timeout=300)
+        decode_wall = time.time() - t1
+        TPDriverMixin.shutdown()
+        res = {
+            "probe": "tp2_one_gpu_rccl",
+            "model": MODEL,
+            "free_match": out[0] == ref[0],
+            "guided_match": out[1] == ref[1],
+            "steps": eng.steps,
+            "wall_s": round(wall, 2),
+            "decode64_wall_s": round(decode_wall, 2),
+            "decode_ms_per_step": round(decode_wall / 64 * 1000, 2),
+        }
+        (OUT / "tp2_check.json").write_text(json.dumps(res, indent=1))
+        print(json.dumps(res, indent=1), flush=True)
+        assert out[0] == ref[0], f"free decode mismatch: {out[0][:8]} vs {ref[0][:8]}"
+        assert out[1] == ref[1], "guided decode mismatch"
+    else:
+        pool = KVCachePool(
+            spec.num_layers,
+            model.num_kv_heads_local,
+            spec.head_dim,
+            num_blocks=4096,
+            block_size=16,
+            dtype=torch.bfloat16,
+            device="cuda",
+        )
+        run_tp_worker(model, pool, "cuda")
+    dist.barrier()
+    dist.destroy_process_group()
+    if rank == 0:
+        print("TP2 CHECK OK", flush=True)
+
+
+if __name__ == "__main__":
+    main()

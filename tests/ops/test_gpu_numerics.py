@@ -255,6 +255,40 @@ class TestGemv:
         torch.testing.assert_close(out.cpu().float(), ref, atol=3e-2, rtol=3e-2)
 
 
+class TestMoeGrouped:
+    def test_grouped_matches_reference(self):
+        from dts_amd.ops import torch_ref
+
+        E, N, K, P = 4, 512, 1024, 13
+        x = bf(torch.randn(P, K)).to(DEV)
+        w = bf(torch.randn(E, N, K) / math.sqrt(K)).to(DEV)
+        counts = torch.tensor([3, 0, 6, 4], dtype=torch.int32)
+        offsets = torch.tensor([0, 3, 3, 9], dtype=torch.int32)
+        out = torch.empty(P, N, dtype=torch.bfloat16, device=DEV)
+        ext.moe_grouped_linear(out, x, w, counts.to(DEV), offsets.to(DEV))
+        ref = torch_ref.moe_grouped_linear(
+            x.cpu(), w.cpu(), counts, offsets
+        )
+        torch.testing.assert_close(
+            out.cpu().float(), ref.float(), atol=3e-2, rtol=3e-2
+        )
+
+    def test_segment_longer_than_16(self):
+        from dts_amd.ops import torch_ref
+
+        E, N, K, P = 2, 256, 512, 40
+        x = bf(torch.randn(P, K)).to(DEV)
+        w = bf(torch.randn(E, N, K) / math.sqrt(K)).to(DEV)
+        counts = torch.tensor([25, 15], dtype=torch.int32)
+        offsets = torch.tensor([0, 25], dtype=torch.int32)
+        out = torch.empty(P, N, dtype=torch.bfloat16, device=DEV)
+        ext.moe_grouped_linear(out, x, w, counts.to(DEV), offsets.to(DEV))
+        ref = torch_ref.moe_grouped_linear(x.cpu(), w.cpu(), counts, offsets)
+        torch.testing.assert_close(
+            out.cpu().float(), ref.float(), atol=3e-2, rtol=3e-2
+        )
+
+
 class TestGemmSkinny:
     @pytest.mark.parametrize(
         "M,K,N",

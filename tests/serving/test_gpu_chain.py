@@ -122,18 +122,49 @@ class TestChainDeterminism:
             eng.stop()
 
 
+def _mk_mini(**env):
+    old = {}
+    for k, v in env.items():
+        old[k] = os.environ.get(k)
+        os.environ[k] = v
+    try:
+        return ServingEngine(
+            model_name="llama-mini-gpu",  # small vocab: greedy loops fast
+            device="cuda:0",
+            dtype=torch.bfloat16,
+            kv_memory_bytes=1 << 30,
+            weight_seed=7,
+        )
+    finally:
+        for k, v in old.items():
+            if v is None:
+                os.environ.pop(k, None)
+            else:
+                os.environ[k] = v
+
+
+MINI_PROMPT = [300 + (i * 37) % 900 for i in range(32)]
+
+
 class TestSpecOnGPU:
     """Draft rows change the forward's M, so bf16 logits are not bitwise
     equal to the spec-off run and near-tie samples can legitimately flip
     (the distribution-level equivalence is pinned by the fp32 CPU tests
     in test_spec_decode.py). On GPU we pin behavior: acceptance happens,
-    steps shrink, and the spec path is self-deterministic."""
+    steps shrink, and the spec path is self-deterministic. Uses the
+    small-vocab llama-mini-gpu: greedy decode on random weights cycles
+    within ~a hundred tokens, so prompt-lookup actually fires (an
+    8B-class random model does not repeat a bigram that fast)."""
 
     def test_spec_accepts_and_saves_steps(self):
-        eng_off = _mk_engine(DTS_SPEC_K="0", DTS_NO_CHAIN="1")
-        ref = _gen(eng_off, PROMPT * 2, seed=None, temperature=0.0, max_tokens=96)
-        eng_on = _mk_engine(DTS_SPEC_K="4", DTS_NO_CHAIN="1")
-        out = _gen(eng_on, PROMPT * 2, seed=None, temperature=0.0, max_tokens=96)
+        eng_off = _mk_mini(DTS_SPEC_K="0", DTS_NO_CHAIN="1")
+        ref = _gen(
+            eng_off, MINI_PROMPT * 2, seed=None, temperature=0.0, max_tokens=256
+        )
+        eng_on = _mk_mini(DTS_SPEC_K="4", DTS_NO_CHAIN="1")
+        out = _gen(
+            eng_on, MINI_PROMPT * 2, seed=None, temperature=0.0, max_tokens=256
+        )
         assert out.completion_tokens == ref.completion_tokens
         assert eng_on.spec_draft_tokens > 0
         assert eng_on.spec_accepted_tokens > 0
@@ -143,8 +174,12 @@ class TestSpecOnGPU:
         """Same seed + same spec config → identical stream run-to-run."""
         outs = []
         for _ in range(2):
-            eng = _mk_engine(DTS_SPEC_K="4", DTS_NO_CHAIN="1")
-            outs.append(_gen(eng, PROMPT, seed=5, temperature=0.05).token_ids)
+            eng = _mk_mini(DTS_SPEC_K="4", DTS_NO_CHAIN="1")
+            outs.append(
+                _gen(
+                    eng, MINI_PROMPT * 2, seed=5, temperature=0.05, max_tokens=256
+                ).token_ids
+            )
             assert eng.spec_draft_tokens > 0
         assert outs[0] == outs[1]
 

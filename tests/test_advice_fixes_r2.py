@@ -117,13 +117,30 @@ class TestStuckLeak:
 
 
 class TestMixtralGraphFlag:
-    """ADVICE high: MoE masked-gather breaks hipGraph capture — the
-    engine must not wrap Mixtral in the DecodeGraphRunner."""
+    """ADVICE high: MoE masked-gather breaks hipGraph capture. The fix is
+    the capture-safe grouped GEMM decode path (GPU, single rank); EP and
+    no-HIP configurations must still refuse graph capture."""
 
-    def test_flag(self):
+    def test_ep_not_capturable(self):
+        from dts_amd.models.config import get_model_spec
+        from dts_amd.models.mixtral import MixtralModel
+        from dts_amd.parallel.tp import TPContext
+
+        spec = get_model_spec("mixtral-tiny")
+        m = MixtralModel(
+            spec, tp=TPContext(None, 0, 2), dtype=torch.float32, device="cpu"
+        )
+        assert m.graph_capturable is False
+
+    def test_no_hip_not_capturable(self):
+        from dts_amd.models.config import get_model_spec
         from dts_amd.models.mixtral import MixtralModel
 
-        assert MixtralModel.graph_capturable is False
+        spec = get_model_spec("mixtral-tiny")
+        m = MixtralModel(spec, dtype=torch.float32, device="cpu")
+        from dts_amd import ops
+
+        assert m.graph_capturable == (ops.hip_available())
 
     def test_llama_still_capturable(self):
         from dts_amd.models.llama import LlamaModel
