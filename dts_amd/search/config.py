@@ -60,6 +60,11 @@ class DTSConfig:
     # the score phase's sequential decode depth ~4x on the local engine.
     # False reproduces the reference's single combined ranking call.
     comparative_split: bool = True
+    # split strategy generation: one parallel call per strategy through
+    # a distinct diversity lens (shared prompt prefix) instead of the
+    # reference's single N-node form — cuts the init phase's sequential
+    # decode depth ~N x on the local engine.
+    strategy_split: bool = True
 
     def __post_init__(self) -> None:
         if self.scoring_mode not in ("absolute", "comparative"):

@@ -27,17 +27,9 @@ from dts_amd.utils.logging import log_phase
 # slice of the initial branches through a distinct angle, so the union
 # stays diverse without the single serial mega-call (one 6*world-entry
 # form decoded on rank 0 while the other GPUs idle was the Amdahl
-# bottleneck of the weak-scaling bench)
-_DIVERSITY_ANGLES = [
-    "empathy-first emotional connection",
-    "pragmatic step-by-step problem solving",
-    "data-driven evidence and benchmarks",
-    "narrative framing and storytelling",
-    "expert-authority framing with credible sourcing",
-    "collaborative co-design with the user",
-    "contrarian assumption-challenging",
-    "incremental trust-building and small commitments",
-]
+# bottleneck of the weak-scaling bench). Shared with the within-rank
+# split mode (generator.strategy_split).
+from dts_amd.search.generator import DIVERSITY_ANGLES as _DIVERSITY_ANGLES
 
 
 class _DPSimulator:
@@ -199,10 +191,16 @@ class DistributedDTSEngine(DTSEngine):
 
         local_n = cfg.init_branches // world
         angle = _DIVERSITY_ANGLES[rank % len(_DIVERSITY_ANGLES)]
-        ctx = (f"{deep}\n\n" if deep else "") + (
-            "Diversity constraint: approach ALL of your strategies through "
-            f"the lens of {angle}; other strategy sets cover other lenses."
-        )
+        if getattr(self._generator, "strategy_split", False):
+            # within-rank split mode already applies per-call lenses —
+            # rotate them by rank so the global lens set stays disjoint
+            self._generator.lens_offset = rank * local_n
+            ctx = (f"{deep}" if deep else None)
+        else:
+            ctx = (f"{deep}\n\n" if deep else "") + (
+                "Diversity constraint: approach ALL of your strategies through "
+                f"the lens of {angle}; other strategy sets cover other lenses."
+            )
         strategies = await self._generator.generate_strategies(
             cfg.first_message, local_n, ctx
         )

@@ -64,6 +64,7 @@ class DTSEngine:
             max_tokens=config.budget.strategy,
             intent_max_tokens=config.budget.intent,
             seed=config.seed,
+            strategy_split=config.strategy_split,
         )
         self._simulator = ConversationSimulator(
             llm=llm,

@@ -26,6 +26,21 @@ FIXED_INTENT = UserIntent(
     cognitive_stance="analytical, asks probing questions",
 )
 
+# per-call lenses for split strategy generation (also used per-rank by
+# the DP sharded init, search/dist_engine.py): each call covers a
+# distinct angle so the union stays diverse without a single serial
+# mega-form
+DIVERSITY_ANGLES = [
+    "empathy-first emotional connection",
+    "pragmatic step-by-step problem solving",
+    "data-driven evidence and benchmarks",
+    "narrative framing and storytelling",
+    "expert-authority framing with credible sourcing",
+    "collaborative co-design with the user",
+    "contrarian assumption-challenging",
+    "incremental trust-building and small commitments",
+]
+
 
 class StrategyGenerator:
     def __init__(
@@ -39,6 +54,7 @@ class StrategyGenerator:
         max_tokens: int = 1024,
         intent_max_tokens: Optional[int] = None,
         seed: Optional[int] = None,
+        strategy_split: bool = False,
     ) -> None:
         self.llm = llm
         self.goal = goal
@@ -47,6 +63,13 @@ class StrategyGenerator:
         self.max_tokens = max_tokens
         self.intent_max_tokens = intent_max_tokens or max_tokens
         self.seed = seed
+        # split strategy generation: one parallel call per strategy,
+        # distinct diversity lens each, shared prompt prefix — cuts the
+        # init phase's sequential decode depth ~count x on the local
+        # engine. False reproduces the reference's single N-node call.
+        self.strategy_split = strategy_split
+        # rotated by the DP sharded init so ranks' lens sets don't overlap
+        self.lens_offset = 0
         self._sem = asyncio.Semaphore(max_concurrency)
         self._on_usage = on_usage
 
@@ -56,6 +79,10 @@ class StrategyGenerator:
         count: int,
         deep_research_context: Optional[str] = None,
     ) -> list:
+        if self.strategy_split and count > 1:
+            return await self._generate_strategies_split(
+                first_message, count, deep_research_context
+            )
         system, user = prompts.conversation_tree_generator(
             num_nodes=count,
             conversation_goal=self.goal,
@@ -69,6 +96,37 @@ class StrategyGenerator:
             Strategy(tagline=tagline, description=str(desc))
             for tagline, desc in result.get("nodes", {}).items()
         ]
+        return strategies
+
+    async def _generate_strategies_split(
+        self,
+        first_message: str,
+        count: int,
+        deep_research_context: Optional[str],
+    ) -> list:
+        calls = []
+        for i in range(count):
+            system, user = prompts.conversation_tree_generator_single(
+                index=i + 1,
+                total=count,
+                conversation_goal=self.goal,
+                conversation_context=first_message,
+                lens=DIVERSITY_ANGLES[
+                    (self.lens_offset + i) % len(DIVERSITY_ANGLES)
+                ],
+                deep_research_context=deep_research_context,
+            )
+            calls.append(self._call_json(system, user, phase="strategy"))
+        results = await asyncio.gather(*calls, return_exceptions=True)
+        strategies: list = []
+        for i, result in enumerate(results):
+            if isinstance(result, Exception) or not isinstance(result, dict):
+                logger.warning("Split strategy call %d failed: %s", i + 1, result)
+                continue
+            for tagline, desc in (result.get("nodes") or {}).items():
+                strategies.append(Strategy(tagline=tagline, description=str(desc)))
+        if not strategies:
+            raise RuntimeError("Strategy generation failed after retries")
         return strategies
 
     async def generate_intents(self, history: list, count: int) -> list:

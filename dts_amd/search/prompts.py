@@ -56,6 +56,47 @@ def conversation_tree_generator(
     return system, user
 
 
+def conversation_tree_generator_single(
+    index: int,
+    total: int,
+    conversation_goal: str,
+    conversation_context: str,
+    lens: str,
+    deep_research_context: Optional[str] = None,
+) -> tuple:
+    """Split strategy generation: one strategy per call, all calls
+    sharing the (system + goal + context) prompt prefix; a per-call
+    diversity lens replaces the combined form's see-your-siblings
+    diversity pressure. Sequential decode depth drops from one
+    N-strategy form to one strategy."""
+    system = (
+        "[dts:strategy] You design opening strategies for goal-directed "
+        "conversations. Produce genuinely distinct approaches — different "
+        "framings, orderings and emotional registers, not paraphrases of one "
+        "idea. Respond with a single valid JSON object and nothing else: no "
+        "markdown fences, no commentary."
+    )
+    research = (
+        f"\n\nBackground research to draw on:\n{deep_research_context}\n"
+        if deep_research_context
+        else ""
+    )
+    user = (
+        f"Goal of the conversation: {conversation_goal}\n\n"
+        f"The user opens with: {conversation_context}\n{research}\n"
+        f"This is strategy {index} of {total}. Approach it strictly "
+        f"through the lens of: {lens}. Other calls cover other lenses.\n\n"
+        "Return JSON of this shape:\n"
+        "{\n"
+        '  "goal": "restated goal",\n'
+        '  "nodes": {"<short tagline>": "<2-3 sentence description of the strategy>"},\n'
+        '  "coverage_rationale": "why this angle matters"\n'
+        "}\n"
+        'The "nodes" object must contain exactly 1 entries.'
+    )
+    return system, user
+
+
 def user_intent_generator(
     num_intents: int,
     conversation_goal: str,

@@ -662,6 +662,7 @@ def build_model(spec: ModelSpec, dtype, device, tp=None):
 # ---------------------------------------------------------------------------
 
 _N_RE = re.compile(r"exactly (\d+)")
+_STRAT_SPLIT_RE = re.compile(r"This is strategy (\d+) of (\d+)")
 _TRAJ_RE = re.compile(r"--- Trajectory ([0-9a-fA-F-]+)")
 
 
@@ -694,6 +695,9 @@ class LocalBackend:
         user = next((m.content for m in reversed(messages) if m.role == "user"), "") or ""
         tok = engine.tokenizer
         if "[dts:strategy]" in system:
+            ms = _STRAT_SPLIT_RE.search(user)
+            if ms:  # split mode: one strategy per call, numbered keys
+                return structured.strategy_form(tok, 1, start=int(ms.group(1)))
             m = _N_RE.search(user)
             return structured.strategy_form(tok, int(m.group(1)) if m else 4)
         if "[dts:intent]" in system:

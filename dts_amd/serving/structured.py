@@ -296,14 +296,16 @@ JUDGE_CRITERIA = (
 )
 
 
-def strategy_form(tok, n: int) -> FormGuide:
+def strategy_form(tok, n: int, start: int = 1) -> FormGuide:
+    """`start` numbers the fixed key prefixes — split strategy calls
+    each produce one node and need distinct taglines across calls."""
     segs: list = [Fixed('{"goal": '), JsonString(48), Fixed(', "nodes": {')]
     for i in range(n):
         if i:
             segs.append(Fixed(", "))
         # unique fixed key prefix guarantees n distinct dict keys
         segs += [
-            Fixed(f'"Strategy {i + 1}: '),
+            Fixed(f'"Strategy {start + i}: '),
             Free(max_tokens=24, stop=ord('"')),
             Fixed(": "),
             JsonString(96),

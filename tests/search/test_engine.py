@@ -164,3 +164,67 @@ class TestPrunePolicy:
         assert extra.status == NodeStatus.PRUNED
         assert extra.prune_reason == "scoring failed"
         assert survivors == [nodes[0]]
+
+
+class TestStrategySplit:
+    """Split strategy generation: N parallel single-strategy calls."""
+
+    def test_split_merges_strategies(self, run_async):
+        import json as _json
+
+        from dts_amd.llm import LLM, ScriptedBackend
+        from dts_amd.search.generator import StrategyGenerator
+
+        def one(i):
+            return _json.dumps(
+                {"goal": "g", "nodes": {f"Strategy {i}: tag{i}": f"desc{i}"},
+                 "coverage_rationale": "r"}
+            )
+
+        backend = ScriptedBackend([one(1), one(2), one(3)])
+        gen = StrategyGenerator(
+            LLM(backend, default_model="m"), goal="g", strategy_split=True
+        )
+        out = run_async(gen.generate_strategies("hello", 3))
+        assert len(out) == 3
+        assert {s.tagline for s in out} == {
+            "Strategy 1: tag1", "Strategy 2: tag2", "Strategy 3: tag3"
+        }
+        users = [m[-1].content for m in backend.calls]
+        assert "This is strategy 1 of 3" in users[0]
+        assert "This is strategy 3 of 3" in users[2]
+        import os as _os
+
+        prefix = _os.path.commonprefix(users)
+        assert "The user opens with" in prefix  # shared KV prefix
+
+    def test_split_partial_failure_keeps_rest(self, run_async):
+        import json as _json
+
+        from dts_amd.llm import LLM, ScriptedBackend
+        from dts_amd.search.generator import StrategyGenerator
+
+        ok = _json.dumps({"goal": "g", "nodes": {"Strategy 2: t": "d"}})
+        backend = ScriptedBackend(["bad", "bad", "bad", ok])
+        gen = StrategyGenerator(
+            LLM(backend, default_model="m"), goal="g", strategy_split=True
+        )
+        out = run_async(gen.generate_strategies("hello", 2))
+        assert len(out) == 1
+
+    def test_split_off_single_call(self, run_async):
+        import json as _json
+
+        from dts_amd.llm import LLM, ScriptedBackend
+        from dts_amd.search.generator import StrategyGenerator
+
+        payload = _json.dumps(
+            {"goal": "g", "nodes": {"A": "da", "B": "db"}}
+        )
+        backend = ScriptedBackend([payload])
+        gen = StrategyGenerator(
+            LLM(backend, default_model="m"), goal="g", strategy_split=False
+        )
+        out = run_async(gen.generate_strategies("hello", 2))
+        assert len(out) == 2
+        assert len(backend.calls) == 1
