@@ -884,10 +884,15 @@ void attn_prefill_paged(torch::Tensor out, torch::Tensor q, torch::Tensor cu_q,
       const char* e = getenv("DTS_PREFILL_V");
       return e ? atoi(e) : -1;
     }();
-    // default: v2 with 32-key tiles (v1's LDS budget, 5 WGs/CU, plus
-    // the staged pipeline and transposed-V reads); DTS_PREFILL_V=0
-    // selects v1, 2 the 64-key variant (2 WGs/CU — measured slower)
-    swz = (env_v >= 0) ? env_v : 3;
+    // default: v1. The v2 ladder steps measured SLOWER on hardware —
+    // v1 147/179/220 TF at N=2k/4k/8k vs v2(KSTEP=64) 127/167/194 and
+    // v2(KSTEP=32, same LDS budget as v1) 119/156/184
+    // (profiles/prefill_v2_ablation_r2.md): at 5 WGs/CU the cooperative
+    // loads are already TLP-hidden, so the reg-staged pipeline only adds
+    // VGPR pressure, and the transposed-V image trades scalar reads for
+    // 4-way-conflicted scalar stores. Kept behind DTS_PREFILL_V=2/3 as
+    // the measured ablation.
+    swz = (env_v >= 0) ? env_v : 0;
   }
   const int Hq = q.size(1), D = q.size(2);
   const int Hkv = kcache.size(1);
