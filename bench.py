@@ -131,11 +131,18 @@ def main():
     else:
         kv_bytes = 512 << 20
 
+    # two-engine runs (config 5): the actor engine takes a conservative
+    # slice and the judge engine sizes itself from memory REMAINING
+    # after the actor's weights+KV+graphs — a blind 50/50 split of
+    # pre-weights free memory over-committed (47B judge weights + pool
+    # exceeded the 288 GB card)
     engine = ServingEngine(
         model_name=args.model,
         device=device,
         dtype=dtype,
-        kv_memory_bytes=kv_bytes if not args.judge_model else kv_bytes // 2,
+        kv_memory_bytes=(
+            int(kv_bytes * 0.35) if args.judge_model else kv_bytes
+        ),
         max_batch_tokens=16384,
         max_running=512,
         weight_seed=0,
@@ -146,7 +153,9 @@ def main():
             model_name=args.judge_model,
             device=device,
             dtype=dtype,
-            kv_memory_bytes=kv_bytes // 2,
+            # None on GPU: the engine sizes its pool from memory free
+            # AFTER its own weights loaded (0.8 of the remainder)
+            kv_memory_bytes=None if use_gpu else kv_bytes // 2,
             max_batch_tokens=16384,
             max_running=512,
             weight_seed=1,
