@@ -1,18 +1,23 @@
-"""TP=2 over real RCCL on ONE MI355X (VERDICT round-1 #2).
+"""TP=2 end-to-end on ONE MI355X (VERDICT round-1 #2).
 
-Launch (both ranks share the single GPU — valid RCCL same-device comms;
-HSA_ENABLE_IPC_MODE_LEGACY=0 must be exported for dmabuf IPC):
-
-  python -m torch.distributed.run --nnodes=1 --nproc-per-node 2 \
+  python -m torch.distributed.run --nnodes=1 --nproc-per-node 2 \\
       --master-addr 127.0.0.1 scripts/tp2_gpu_check.py
 
 Rank 0 first computes a DENSE (tp=1) greedy reference on the same
 weights, then both ranks run the sharded engine: column/row-parallel
-linears with per-layer RCCL all-reduce over the packed-tensor batch
+linears with a per-layer all-reduce over the packed-tensor batch
 broadcast (serving/tp_engine.py). Checks sharded == dense token streams
 for free decode AND a guided (constrained-JSON) request, and times the
 per-step overhead. Writes gpurun_out/tp2_check.json.
-"""
+
+Transport note (measured): RCCL REFUSES two ranks on one device --
+"NCCL WARN Duplicate GPU detected : rank 0 and rank 1 both on CUDA
+device 8e000 / ncclInvalidUsage" (gpurun_out/tp2_debug.log) -- exactly
+like NCCL >= 2.5. On a single-GPU lease this script therefore falls
+back to gloo WITH CUDA TENSORS: the sharded weights, HIP kernels,
+packed device-tensor batch broadcast and per-layer all-reduce all run
+on the GPU; only the wire transport differs. On a multi-GPU node
+(ranks on distinct devices) the same script runs pure RCCL."""
 
 import json
 import os
@@ -31,12 +36,32 @@ OUT.mkdir(exist_ok=True)
 MODEL = os.environ.get("TP2_MODEL", "llama-3-8b-2l")
 
 
+def _init_dist(rank: int, world: int) -> str:
+    n_dev = torch.cuda.device_count()
+    torch.cuda.set_device(rank % n_dev)
+    if n_dev >= world and os.environ.get("TP2_BACKEND", "auto") != "gloo":
+        dist.init_process_group("nccl", rank=rank, world_size=world)
+        return "nccl"
+    try:
+        dist.init_process_group("nccl", rank=rank, world_size=world)
+        dist.barrier()  # forces communicator init → duplicate-GPU raises
+        return "nccl"
+    except Exception as e:  # noqa: BLE001
+        print(f"[rank {rank}] RCCL same-device init refused ({type(e).__name__}); "
+              "falling back to gloo with CUDA tensors", flush=True)
+        try:
+            dist.destroy_process_group()
+        except Exception:  # noqa: BLE001
+            pass
+        dist.init_process_group("gloo", rank=rank, world_size=world)
+        return "gloo"
+
+
 def main():
     rank = int(os.environ["RANK"])
     world = int(os.environ["WORLD_SIZE"])
     assert world == 2
-    torch.cuda.set_device(0)  # both ranks share the one GPU
-    dist.init_process_group("nccl", rank=rank, world_size=world)
+    transport = _init_dist(rank, world)
 
     from dts_amd.llm.types import SamplingParams
     from dts_amd.models.config import get_model_spec
@@ -124,7 +149,8 @@ def main():
         decode_wall = time.time() - t1
         TPDriverMixin.shutdown()
         res = {
-            "probe": "tp2_one_gpu_rccl",
+            "probe": "tp2_one_gpu",
+            "transport": transport,
             "model": MODEL,
             "free_match": out[0] == ref[0],
             "guided_match": out[1] == ref[1],
