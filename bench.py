@@ -126,23 +126,21 @@ def main():
 
     if use_gpu:
         torch.cuda.set_device(device)
-        free, _ = torch.cuda.mem_get_info()
-        kv_bytes = int(free * args.kv_frac)
+        kv_bytes = None  # engines self-size from POST-weights free memory
     else:
         kv_bytes = 512 << 20
 
-    # two-engine runs (config 5): the actor engine takes a conservative
-    # slice and the judge engine sizes itself from memory REMAINING
-    # after the actor's weights+KV+graphs — a blind 50/50 split of
-    # pre-weights free memory over-committed (47B judge weights + pool
-    # exceeded the 288 GB card)
+    # KV sizing: each engine measures free memory AFTER its own weights
+    # load and takes kv_frac of it (a pre-weights split over-committed:
+    # 70B weights + 0.75-of-free pool exceeded the 288 GB card). For
+    # two-engine runs (config 5) the actor takes a conservative slice so
+    # the judge's 94 GB of weights still fit.
     engine = ServingEngine(
         model_name=args.model,
         device=device,
         dtype=dtype,
-        kv_memory_bytes=(
-            int(kv_bytes * 0.35) if args.judge_model else kv_bytes
-        ),
+        kv_memory_bytes=kv_bytes,
+        kv_frac=args.kv_frac * (0.45 if args.judge_model else 1.0),
         max_batch_tokens=16384,
         max_running=512,
         weight_seed=0,
@@ -153,9 +151,8 @@ def main():
             model_name=args.judge_model,
             device=device,
             dtype=dtype,
-            # None on GPU: the engine sizes its pool from memory free
-            # AFTER its own weights loaded (0.8 of the remainder)
             kv_memory_bytes=None if use_gpu else kv_bytes // 2,
+            kv_frac=args.kv_frac,
             max_batch_tokens=16384,
             max_running=512,
             weight_seed=1,

@@ -59,6 +59,7 @@ class ServingEngine:
         dtype: Optional[torch.dtype] = None,
         num_blocks: Optional[int] = None,
         kv_memory_bytes: Optional[int] = None,
+        kv_frac: float = 0.8,
         block_size: int = 16,
         max_batch_tokens: int = 8192,
         max_running: int = 256,
@@ -107,8 +108,11 @@ class ServingEngine:
         if num_blocks is None:
             if kv_memory_bytes is None:
                 if device.startswith("cuda"):
+                    # sized from free memory AFTER this engine's weights
+                    # loaded — callers wanting pressure control pass
+                    # kv_frac instead of guessing absolute bytes
                     free, _total = torch.cuda.mem_get_info()
-                    kv_memory_bytes = int(free * 0.8)
+                    kv_memory_bytes = int(free * kv_frac)
                 else:
                     kv_memory_bytes = 256 << 20  # CPU tests
             num_blocks = KVCachePool.blocks_for_memory(
