@@ -213,13 +213,15 @@ def linear_bf16(x: torch.Tensor, weight: torch.Tensor) -> torch.Tensor:
     through custom weight-streaming kernels; everything else is a
     hipBLASLt GEMM via F.linear.
 
-    Dispatch (measured A/B, profiles/microbench r2): hipBLASLt has a
-    ~20 us per-GEMM floor, which caps weights <= ~96 MB at ~2.5 TB/s —
-    there the MFMA skinny GEMM (gemm_skinny.hip) wins at every M<=16
-    (10.9 us on the 75 MB qkv). On the ~117 MB gate_up/down weights
-    hipBLASLt streams at 5.7-5.8 TB/s and wins from M>=5; at M<=2 the
-    VALU GEMV leads on shallow-K shapes but underfills the grid at
-    K=14336/N=4096 (32 us vs skinny's 20.6), hence the K split."""
+    Dispatch (measured A/B, profiles/microbench r2 + the r2 bench
+    kernel table): hipBLASLt has a ~20 us per-GEMM floor (caps <=96 MB
+    weights at ~2.5 TB/s), and INSIDE captured hipGraphs it picks
+    capture-unfriendly algorithms — the bench's graph-replayed gate_up
+    averaged 59.5 us (~2-4 TB/s) where isolation measured 20.7. The
+    skinny MFMA GEMM has no algo selection and measures 22-30 us on the
+    117 MB shapes at M=4..16, so it takes EVERY shape at M<=16; the
+    VALU GEMV keeps M<=2 on shallow-K big weights (19.7 us vs skinny's
+    21.2 on gate_up; it underfills the grid at K=14336)."""
     M = x.shape[0] if x.dim() == 2 else 0
     if (
         x.is_cuda
@@ -229,16 +231,15 @@ def linear_bf16(x: torch.Tensor, weight: torch.Tensor) -> torch.Tensor:
         and weight.stride(1) == 1
         and 1 <= M <= 16
     ):
-        big = weight.shape[0] * weight.shape[1] * 2 > (96 << 20)
-        if not (big and M >= 5):  # big weights at M>=5: hipBLASLt wins
-            ext = _require_hip()
-            if ext is not None:
-                out = torch.empty(
-                    (M, weight.shape[0]), dtype=x.dtype, device=x.device
-                )
-                if M <= 2 and big and x.shape[1] <= 8192:
-                    ext.gemv_bf16(out, x, weight)
-                else:
-                    ext.gemm_skinny_bf16(out, x, weight)
-                return out
+        ext = _require_hip()
+        if ext is not None:
+            out = torch.empty(
+                (M, weight.shape[0]), dtype=x.dtype, device=x.device
+            )
+            big = weight.shape[0] * weight.shape[1] * 2 > (96 << 20)
+            if M <= 2 and big and x.shape[1] <= 8192:
+                ext.gemv_bf16(out, x, weight)
+            else:
+                ext.gemm_skinny_bf16(out, x, weight)
+            return out
     return torch.nn.functional.linear(x, weight)

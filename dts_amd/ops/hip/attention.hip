@@ -821,7 +821,14 @@ void attn_decode_paged(torch::Tensor out, torch::Tensor q,
   // Fixed split count keeps the launch hipGraph-capturable while filling
   // the chip at small batch (B=1: 8 WGs -> 8*S*Hkv WGs) AND bounding the
   // serial page chain per wave (the decode critical path at long kv).
-  const int S = 32;
+  // DTS_DECODE_SPLITS tunes it: the r2 bench kernel table showed the
+  // combine kernel (whose work scales with S) at 6.7% of GPU busy while
+  // typical bench kv is ~3k (6 pages/split at S=32 — launch-bound).
+  static const int S = [] {
+    const char* e = getenv("DTS_DECODE_SPLITS");
+    int v = e ? atoi(e) : 32;
+    return (v >= 1 && v <= 64) ? v : 32;
+  }();
   // Workspace cached per shape: layers within a step run sequentially on
   // one stream, so one buffer serves all 32 layer calls (and, being
   // allocated at warm-up time, lives OUTSIDE the graph pool).
