@@ -165,15 +165,41 @@ def gelu(x):
     return torch_ref.gelu(x)
 
 
+_sampler_ws: dict = {}
+
+
+def _sampler_workspace(S: int, device):
+    """Cached v2 sampler workspace (stable pointers for graph capture)."""
+    key = (str(device), S)
+    ws = _sampler_ws.get(key)
+    if ws is None:
+        ws = (
+            torch.zeros(S, dtype=torch.int64, device=device),
+            torch.zeros(S, 1024, dtype=torch.float32, device=device),
+            torch.zeros(S, 16, dtype=torch.float32, device=device),
+            torch.zeros(S, dtype=torch.float32, device=device),
+        )
+        _sampler_ws[key] = ws
+    return ws
+
+
 def top_p_sample(logits, temperatures, top_ps, generators=None, seeds=None, out=None):
     if _on_gpu(logits):
         ext = _require_hip()
         if ext is not None and seeds is not None:
+            S, V = logits.shape
             if out is None:
-                out = torch.empty(
-                    logits.shape[0], dtype=torch.long, device=logits.device
+                out = torch.empty(S, dtype=torch.long, device=logits.device)
+            # v2 (gridded) at decode widths: the one-WG-per-row v1 left
+            # ~250 of 256 CUs idle (364 us avg on the r2 bench); v1 keeps
+            # tiny vocabs (guided sub-vocab picks) and very wide batches
+            if V >= 4096 and S <= 32 and os.environ.get("DTS_SAMPLER_V1") != "1":
+                ws = _sampler_workspace(S, logits.device)
+                ext.top_p_sample_v2(
+                    out, logits, temperatures, top_ps, seeds, *ws
                 )
-            ext.top_p_sample(out, logits, temperatures, top_ps, seeds)
+            else:
+                ext.top_p_sample(out, logits, temperatures, top_ps, seeds)
             return out
     return torch_ref.top_p_sample(logits, temperatures, top_ps, generators)
 

@@ -25,6 +25,11 @@ void top_p_sample(torch::Tensor out, torch::Tensor logits, torch::Tensor temps,
                   torch::Tensor top_ps, torch::Tensor seeds);
 void derive_seeds(torch::Tensor out, torch::Tensor bases,
                   torch::Tensor positions);
+void top_p_sample_v2(torch::Tensor out, torch::Tensor logits,
+                     torch::Tensor temps, torch::Tensor top_ps,
+                     torch::Tensor seeds, torch::Tensor ws_max,
+                     torch::Tensor ws_bins, torch::Tensor ws_slice,
+                     torch::Tensor ws_tau);
 void gemv_bf16(torch::Tensor out, torch::Tensor x, torch::Tensor w);
 void gemm_skinny_bf16(torch::Tensor out, torch::Tensor x, torch::Tensor w);
 void moe_grouped_linear(torch::Tensor out, torch::Tensor x, torch::Tensor w,
@@ -50,6 +55,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused temperature softmax + top-p sampling (sort-free)");
   m.def("derive_seeds", &derive_seeds,
         "stateless (seed, position) mix for chained decode");
+  m.def("top_p_sample_v2", &top_p_sample_v2,
+        "gridded top-p sampler (vocab sliced across workgroups)");
   m.def("gemv_bf16", &gemv_bf16,
         "skinny-batch (M<=8) bf16 weight-streaming GEMV");
   m.def("gemm_skinny_bf16", &gemm_skinny_bf16,

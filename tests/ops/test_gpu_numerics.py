@@ -353,6 +353,90 @@ class TestSampling:
             nucleus = set(si[: k + 32].tolist())
             assert int(out[i]) in nucleus
 
+    def _v2(self, out, logits, temps, tps, seeds):
+        S = logits.shape[0]
+        ws = (
+            torch.zeros(S, dtype=torch.int64, device=DEV),
+            torch.zeros(S, 1024, dtype=torch.float32, device=DEV),
+            torch.zeros(S, 16, dtype=torch.float32, device=DEV),
+            torch.zeros(S, dtype=torch.float32, device=DEV),
+        )
+        ext.top_p_sample_v2(out, logits, temps, tps, seeds, *ws)
+
+    def test_v2_greedy_matches_argmax(self):
+        S, V = 7, 128256
+        logits = torch.randn(S, V, device=DEV)
+        out = torch.empty(S, dtype=torch.long, device=DEV)
+        self._v2(
+            out, logits,
+            torch.zeros(S, device=DEV), torch.ones(S, device=DEV),
+            torch.arange(S, dtype=torch.long, device=DEV),
+        )
+        assert torch.equal(out.cpu(), logits.argmax(dim=-1).cpu())
+
+    def test_v2_greedy_tie_breaks_lowest(self):
+        V = 8192
+        logits = torch.full((1, V), -5.0, device=DEV)
+        logits[0, 137] = 3.0
+        logits[0, 4242] = 3.0  # exact tie: lowest index must win
+        out = torch.empty(1, dtype=torch.long, device=DEV)
+        self._v2(
+            out, logits,
+            torch.zeros(1, device=DEV), torch.ones(1, device=DEV),
+            torch.zeros(1, dtype=torch.long, device=DEV),
+        )
+        assert int(out[0]) == 137
+
+    def test_v2_top_p_mass_constraint(self):
+        S, V = 16, 128256
+        logits = torch.randn(S, V, device=DEV) * 3
+        temps = torch.full((S,), 0.7, device=DEV)
+        tps = torch.full((S,), 0.9, device=DEV)
+        out = torch.empty(S, dtype=torch.long, device=DEV)
+        self._v2(out, logits, temps, tps,
+                 torch.arange(S, dtype=torch.long, device=DEV))
+        probs = torch.softmax(logits.float() / 0.7, dim=-1)
+        for i in range(S):
+            sp, si = torch.sort(probs[i], descending=True)
+            cum = torch.cumsum(sp, 0)
+            k = int((cum - sp < 0.9).sum())
+            nucleus = set(si[: k + 32].tolist())
+            assert int(out[i]) in nucleus
+
+    def test_v2_deterministic(self):
+        S, V = 4, 128256
+        logits = torch.randn(S, V, device=DEV)
+        temps = torch.full((S,), 0.7, device=DEV)
+        tps = torch.full((S,), 0.95, device=DEV)
+        seeds = torch.tensor([11, 22, 33, 44], dtype=torch.long, device=DEV)
+        outs = []
+        for _ in range(2):
+            out = torch.empty(S, dtype=torch.long, device=DEV)
+            self._v2(out, logits, temps, tps, seeds)
+            outs.append(out.cpu())
+        assert torch.equal(outs[0], outs[1])
+
+    def test_v2_distribution(self):
+        """Over many draws, frequencies track the renormalized nucleus."""
+        V = 8192
+        logits = torch.randn(1, V, device=DEV) * 2
+        temps = torch.full((1,), 1.0, device=DEV)
+        tps = torch.full((1,), 0.95, device=DEV)
+        counts = torch.zeros(V)
+        N = 2000
+        out = torch.empty(1, dtype=torch.long, device=DEV)
+        for s in range(N):
+            self._v2(
+                out, logits, temps, tps,
+                torch.tensor([s * 7919 + 13], dtype=torch.long, device=DEV),
+            )
+            counts[int(out[0])] += 1
+        probs = torch.softmax(logits[0].float(), dim=-1).cpu()
+        top = torch.topk(probs, 5).indices
+        for t in top:
+            expected = float(probs[t]) / 0.95 * N
+            assert abs(counts[t] - expected) < max(6 * math.sqrt(expected), 25)
+
     def test_sampling_distribution(self):
         """Over many draws, frequencies track the renormalized nucleus."""
         V = 1024
