@@ -824,10 +824,15 @@ void attn_decode_paged(torch::Tensor out, torch::Tensor q,
   // DTS_DECODE_SPLITS tunes it: the r2 bench kernel table showed the
   // combine kernel (whose work scales with S) at 6.7% of GPU busy while
   // typical bench kv is ~3k (6 pages/split at S=32 — launch-bound).
+  // Measured on the flagship bench (r2): S=8 0.3549 traj/s, S=16
+  // 0.3491, S=32 0.3266 — at the search's typical kv (~3k) 32 splits
+  // left each workgroup ~6 pages of work (launch/latency-bound) and the
+  // combine kernel at 6.7% of GPU busy. S=8 still gives 8*Hkv*B
+  // workgroups (512 at B=8) and bounded the round-1 12k-kv probe fine.
   static const int S = [] {
     const char* e = getenv("DTS_DECODE_SPLITS");
-    int v = e ? atoi(e) : 32;
-    return (v >= 1 && v <= 64) ? v : 32;
+    int v = e ? atoi(e) : 8;
+    return (v >= 1 && v <= 64) ? v : 8;
   }();
   // Workspace cached per shape: layers within a step run sequentially on
   // one stream, so one buffer serves all 32 layer calls (and, being
