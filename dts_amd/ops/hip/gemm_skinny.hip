@@ -32,6 +32,7 @@ using f32x4 = __attribute__((ext_vector_type(4))) float;
 #define KSTEP 32  // one mfma k-chunk
 #define KUNROLL 4  // k-chunks in flight per iteration
 
+template <int KU>
 __global__ void __launch_bounds__(NWAVES* WAVE)
 gemm_skinny_kernel(short* __restrict__ out,      // [M, N] (row stride out_ts)
                    const short* __restrict__ x,  // [M, K] (row stride x_ts)
@@ -57,16 +58,16 @@ gemm_skinny_kernel(short* __restrict__ out,      // [M, N] (row stride out_ts)
   f32x4 acc = {0.f, 0.f, 0.f, 0.f};
   const bf16x8 zero8 = {0, 0, 0, 0, 0, 0, 0, 0};
 
-  for (int k0 = kbeg; k0 < kend; k0 += KSTEP * KUNROLL) {
-    bf16x8 a[KUNROLL], b[KUNROLL];
+  for (int k0 = kbeg; k0 < kend; k0 += KSTEP * KU) {
+    bf16x8 a[KU], b[KU];
 #pragma unroll
-    for (int u = 0; u < KUNROLL; ++u) {
+    for (int u = 0; u < KU; ++u) {
       const int kk = k0 + u * KSTEP + k_off;
       a[u] = a_live ? *(const bf16x8*)(xrow + kk) : zero8;
       b[u] = b_live ? *(const bf16x8*)(wrow + kk) : zero8;
     }
 #pragma unroll
-    for (int u = 0; u < KUNROLL; ++u)
+    for (int u = 0; u < KU; ++u)
       acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[u], b[u], acc, 0, 0, 0);
   }
 
@@ -195,9 +196,21 @@ void gemm_skinny_bf16(torch::Tensor out, torch::Tensor x, torch::Tensor w) {
   TORCH_CHECK(M >= 1 && M <= 16, "skinny path is for M<=16");
   TORCH_CHECK(K % (NWAVES * KSTEP * KUNROLL) == 0, "K must be /512");
   const int tiles = (N + 15) / 16;
-  hipLaunchKernelGGL(gemm_skinny_kernel, dim3(tiles), dim3(NWAVES * WAVE), 0,
-                     c10::hip::getCurrentHIPStream(), (short*)out.data_ptr(),
-                     (const short*)x.data_ptr(), (const short*)w.data_ptr(), M,
-                     N, K, x.stride(0), out.stride(0));
+  // deeper unroll keeps 128 B/operand/lane in flight (A/B via env)
+  static const int ku = [] {
+    const char* e = getenv("DTS_SKINNY_UNROLL");
+    return (e && e[0] == '8') ? 8 : 4;
+  }();
+  auto stream = c10::hip::getCurrentHIPStream();
+  if (ku == 8 && K % (NWAVES * KSTEP * 8) == 0)
+    hipLaunchKernelGGL((gemm_skinny_kernel<8>), dim3(tiles),
+                       dim3(NWAVES * WAVE), 0, stream, (short*)out.data_ptr(),
+                       (const short*)x.data_ptr(), (const short*)w.data_ptr(),
+                       M, N, K, x.stride(0), out.stride(0));
+  else
+    hipLaunchKernelGGL((gemm_skinny_kernel<4>), dim3(tiles),
+                       dim3(NWAVES * WAVE), 0, stream, (short*)out.data_ptr(),
+                       (const short*)x.data_ptr(), (const short*)w.data_ptr(),
+                       M, N, K, x.stride(0), out.stride(0));
   HIP_CHECK_LAST();
 }
