@@ -60,15 +60,16 @@ class GPT2Layer(nn.Module):
         out = torch.empty(q.shape, dtype=q.dtype, device=q.device)
         np_tok = batch.num_prefill_tokens
         if batch.num_prefill_seqs:
-            out[:np_tok] = ops.attn_prefill_paged(
+            ops.attn_prefill_paged(
                 q[:np_tok], batch.cu_q, batch.positions[:np_tok],
                 k_cache, v_cache, batch.prefill_block_tables,
-                batch.prefill_kv_lens, self.scale,
+                batch.prefill_kv_lens, self.scale, out=out[:np_tok],
             )
         if batch.num_decode_seqs:
-            out[np_tok:] = ops.attn_decode_paged(
+            ops.attn_decode_paged(
                 q[np_tok:], k_cache, v_cache,
                 batch.decode_block_tables, batch.decode_kv_lens, self.scale,
+                out=out[np_tok:],
             )
         hidden = hidden + F.linear(out.reshape(T, -1), self.o_w, self.o_b)
         x = ops.layernorm(hidden, self.ln2_w, self.ln2_b, self.eps)

@@ -114,36 +114,49 @@ def kv_append(k, v, k_cache, v_cache, slot_mapping):
     torch_ref.kv_append(k, v, k_cache, v_cache, slot_mapping)
 
 
-def attn_prefill_paged(q, cu_q, q_positions, k_cache, v_cache, block_tables, kv_lens, scale=None):
+def attn_prefill_paged(
+    q, cu_q, q_positions, k_cache, v_cache, block_tables, kv_lens, scale=None,
+    out=None,
+):
     if _on_gpu(q):
         ext = _require_hip()
         if ext is not None:
             import math
 
-            out = torch.empty(q.shape, dtype=q.dtype, device=q.device)
+            if out is None or not out.is_contiguous():
+                out = torch.empty(q.shape, dtype=q.dtype, device=q.device)
             ext.attn_prefill_paged(
                 out, q, cu_q, q_positions, k_cache, v_cache, block_tables, kv_lens,
                 scale if scale is not None else 1.0 / math.sqrt(q.shape[-1]),
             )
             return out
-    return torch_ref.attn_prefill_paged(
+    ref = torch_ref.attn_prefill_paged(
         q, cu_q, q_positions, k_cache, v_cache, block_tables, kv_lens, scale
     )
+    if out is not None:
+        out.copy_(ref)
+        return out
+    return ref
 
 
-def attn_decode_paged(q, k_cache, v_cache, block_tables, kv_lens, scale=None):
+def attn_decode_paged(q, k_cache, v_cache, block_tables, kv_lens, scale=None, out=None):
     if _on_gpu(q):
         ext = _require_hip()
         if ext is not None:
             import math
 
-            out = torch.empty(q.shape, dtype=q.dtype, device=q.device)
+            if out is None or not out.is_contiguous():
+                out = torch.empty(q.shape, dtype=q.dtype, device=q.device)
             ext.attn_decode_paged(
                 out, q, k_cache, v_cache, block_tables, kv_lens,
                 scale if scale is not None else 1.0 / math.sqrt(q.shape[-1]),
             )
             return out
-    return torch_ref.attn_decode_paged(q, k_cache, v_cache, block_tables, kv_lens, scale)
+    ref = torch_ref.attn_decode_paged(q, k_cache, v_cache, block_tables, kv_lens, scale)
+    if out is not None:
+        out.copy_(ref)
+        return out
+    return ref
 
 
 def silu_mul(gate_up):
