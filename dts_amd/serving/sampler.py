@@ -38,6 +38,9 @@ class Sampler:
     def release(self, seq) -> None:
         self._generators.pop(seq.seq_id, None)
 
+    # (the round-1 CPU-side gather of guided sub-logits was replaced by
+    # the on-device pick below)
+
     def _allowed_device_tensor(
         self, allowed: list, n_max: int, device
     ) -> torch.Tensor:
