@@ -543,6 +543,259 @@ attn_prefill_kernel_v2(short* __restrict__ out,      // [Tq, Hq, D]
 }
 
 // ---------------------------------------------------------------------------
+// Prefill (MFMA) v5 — swapped QK^T + in-register softmax (guide T12)
+//
+// PMC decomposition of v1 (profiles/pmc_prefill_r2.txt): 43% of wave
+// cycles PARKED (s_waitcnt/barrier), 24% issue-stalled, 33% active,
+// 40% LDS bank-conflict overhead, MFMA util ~4%. The parked+active time
+// is dominated by the softmax round trip: scores C-frag -> LDS ->
+// per-row scan -> P -> LDS -> P A-frag, with waitcnt(0)+wave_barrier
+// fences around each hop. v5 computes the TRANSPOSED score tile
+// C'[key, q] = mfma(A = K-frag, B = Q-frag) — the persistent Q
+// registers already have the B layout — so each lane holds its q
+// column's scores in registers; max/sum reduce with two shfl_xor ops,
+// P converts to the P·V A-fragment with 4 packs + 4 shfls, and the
+// scores/p LDS buffers AND their fences disappear (LDS drops 24 KB ->
+// more workgroups per CU).
+// ---------------------------------------------------------------------------
+
+template <int G, int D, int KSTEP>
+__global__ void __launch_bounds__(256)
+attn_prefill_kernel_v5(short* __restrict__ out,      // [Tq, Hq, D]
+                       const short* __restrict__ q,  // [Tq, Hq, D]
+                       const int* __restrict__ cu_q, // [P+1]
+                       const long* __restrict__ q_pos, // [Tq]
+                       const short* __restrict__ kcache,
+                       const short* __restrict__ vcache,
+                       const int* __restrict__ block_tables,
+                       const int* __restrict__ kv_lens, int max_blocks,
+                       int Hkv, float scale, long q_tstride) {
+  constexpr int BS = 16;
+  constexpr int KCHUNKS = D / 32;
+  constexpr int CTILES = D / 16;
+  constexpr int LDS_PAD = 8;
+  constexpr int LDK = D + LDS_PAD;
+  constexpr int ROWS = 64;
+  const int POS_PER_WG = ROWS / G;
+
+  const int seq = blockIdx.y;
+  const int kvh = blockIdx.z;
+  const int Hq = Hkv * G;
+  const int q_start = cu_q[seq];
+  const int q_len = cu_q[seq + 1] - q_start;
+  const int tile = blockIdx.x;
+  if (tile * POS_PER_WG >= q_len) return;
+
+  const int wave = threadIdx.x / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int kv_len = kv_lens[seq];
+  const int* bt = block_tables + (long)seq * max_blocks;
+
+  __shared__ short k_lds[KSTEP][LDK];
+  __shared__ short v_lds[KSTEP][LDK];
+
+  // Q B-fragments (persistent; layout identical to v1's A-frag)
+  short q_frag[KCHUNKS][8];
+  {
+    const int r = lane & 15;
+    const int row_global = wave * 16 + r;
+    const int pos_local = tile * POS_PER_WG + row_global / G;
+    const int head = row_global % G;
+    const bool valid_row = pos_local < q_len;
+    const int tok = q_start + (valid_row ? pos_local : 0);
+    const short* qp = q + (long)tok * q_tstride + (long)(kvh * G + head) * D;
+#pragma unroll
+    for (int kc = 0; kc < KCHUNKS; ++kc) {
+      const int kbase = kc * 32 + (lane >> 4) * 8;
+      bf16x8 v8 = *(const bf16x8*)(qp + kbase);
+#pragma unroll
+      for (int i = 0; i < 8; ++i)
+        q_frag[kc][i] = valid_row ? v8[i] : (short)0;
+    }
+  }
+  // this lane's softmax q (col of the swapped score tile)
+  const int my_q = lane & 15;
+  const int my_row_global = wave * 16 + my_q;
+  const long my_pos_local = tile * POS_PER_WG + my_row_global / G;
+  const bool my_valid = my_pos_local < q_len;
+  const long my_abs_pos = my_valid ? q_pos[q_start + my_pos_local] : -1;
+
+  f32x4 o_acc[CTILES];
+#pragma unroll
+  for (int ct = 0; ct < CTILES; ++ct) o_acc[ct] = {0.f, 0.f, 0.f, 0.f};
+  float run_m = -1e30f, run_l = 0.f;  // for q = my_q (replicated x4 lanes)
+
+  const int last_local_pos = min(q_len, tile * POS_PER_WG + POS_PER_WG) - 1;
+  const long last_abs_pos = q_pos[q_start + last_local_pos];
+  const int kv_hi = min((long)kv_len, last_abs_pos + 1);
+
+  for (int kv_base = 0; kv_base < kv_hi; kv_base += KSTEP) {
+    // ---- cooperative K/V tile load (shared by all 4 waves)
+    {
+      const int elems = KSTEP * D;
+      for (int base = threadIdx.x * 8; base < elems; base += 256 * 8) {
+        const int key = base / D;
+        const int d0 = base % D;
+        const int kglob = kv_base + key;
+        if (kglob < kv_len) {
+          const long blk = bt[kglob / BS];
+          const long off =
+              ((blk * Hkv + kvh) * BS + kglob % BS) * (long)D + d0;
+          *(bf16x8*)(&k_lds[key][d0]) = *(const bf16x8*)(kcache + off);
+          *(bf16x8*)(&v_lds[key][d0]) = *(const bf16x8*)(vcache + off);
+        } else {
+          const bf16x8 z = {0, 0, 0, 0, 0, 0, 0, 0};
+          *(bf16x8*)(&k_lds[key][d0]) = z;
+          *(bf16x8*)(&v_lds[key][d0]) = z;
+        }
+      }
+    }
+    __syncthreads();
+
+    // ---- swapped QK^T: C'[key, q] tiles, KSTEP/16 of them
+    constexpr int STILES = KSTEP / 16;
+    f32x4 sc[STILES];
+#pragma unroll
+    for (int st = 0; st < STILES; ++st) {
+      f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+      for (int kc = 0; kc < KCHUNKS; ++kc) {
+        // A-frag = K rows: row = st*16 + (lane&15), k-dim = d chunk
+        const int key = st * 16 + (lane & 15);
+        const int d0 = kc * 32 + (lane >> 4) * 8;
+        bf16x8 k8 = *(bf16x8*)(&k_lds[key][d0]);
+        acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            k8, *(bf16x8*)q_frag[kc], acc, 0, 0, 0);
+      }
+      sc[st] = acc;  // rows: key = st*16 + (lane>>4)*4 + i; col: my_q
+    }
+
+    // ---- in-register online softmax for column my_q
+    float p[STILES][4];
+    float tmax = -1e30f;
+#pragma unroll
+    for (int st = 0; st < STILES; ++st)
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        const long key_abs = kv_base + st * 16 + (lane >> 4) * 4 + i;
+        const bool ok =
+            my_valid && key_abs <= my_abs_pos && key_abs < (long)kv_len;
+        p[st][i] = ok ? sc[st][i] * scale : -1e30f;
+        tmax = fmaxf(tmax, p[st][i]);
+      }
+    tmax = fmaxf(tmax, __shfl_xor(tmax, 16, 64));
+    tmax = fmaxf(tmax, __shfl_xor(tmax, 32, 64));
+    const float m_new = fmaxf(run_m, tmax);
+    float alpha;
+    float rowsum = 0.f;
+    if (m_new > -1e30f) {
+      alpha = (run_m > -1e30f) ? __expf(run_m - m_new) : 0.f;
+#pragma unroll
+      for (int st = 0; st < STILES; ++st)
+#pragma unroll
+        for (int i = 0; i < 4; ++i) {
+          p[st][i] = (p[st][i] > -1e30f) ? __expf(p[st][i] - m_new) : 0.f;
+          rowsum += p[st][i];
+        }
+    } else {
+      alpha = 1.f;
+#pragma unroll
+      for (int st = 0; st < STILES; ++st)
+#pragma unroll
+        for (int i = 0; i < 4; ++i) p[st][i] = 0.f;
+    }
+    rowsum += __shfl_xor(rowsum, 16, 64);
+    rowsum += __shfl_xor(rowsum, 32, 64);
+    run_l = run_l * alpha + rowsum;
+    run_m = m_new;
+
+    // ---- P' -> P·V A-fragments in registers: pack adjacent-key bf16
+    // pairs, then gather the 8-consecutive-key groups across the 4
+    // lane-groups (keys 4*(l>>4)+i live here; target wants 8*(l>>4)+i)
+    unsigned int ppack[STILES][2];
+#pragma unroll
+    for (int st = 0; st < STILES; ++st) {
+#pragma unroll
+      for (int hpair = 0; hpair < 2; ++hpair) {
+        const unsigned int lo =
+            (unsigned short)f2bf(p[st][hpair * 2]);
+        const unsigned int hi =
+            (unsigned short)f2bf(p[st][hpair * 2 + 1]);
+        ppack[st][hpair] = lo | (hi << 16);
+      }
+    }
+    // per PV k-chunk h (32 keys): A-frag lane needs keys
+    // 8*(lane>>4)..+8 of chunk h — i.e. packs from source groups
+    // g0 = 2*(lane>>4), g1 = 2*(lane>>4)+1 of score-tile st = h*2 +
+    // (g/ (KSTEP/32))... for KSTEP=32 there are 2 score tiles (32 keys)
+    // and ONE PV chunk; for KSTEP=64, 4 tiles and 2 chunks.
+    constexpr int HCH = KSTEP / 32;
+    bf16x8 pa[HCH];
+#pragma unroll
+    for (int h = 0; h < HCH; ++h) {
+      // global key base for this lane's A-frag: h*32 + 8*(lane>>4)
+      // keys 8g..8g+3 come from tile st_a, group ga; keys 8g+4..8g+7
+      // from the NEXT group (same or next tile)
+      const int kb = 8 * (lane >> 4);     // within the 32-key chunk
+      const int st_a = h * 2 + kb / 16;   // score tile holding kb..kb+3
+      const int ga = (kb % 16) / 4;       // lane-group that owns them
+      const int st_b = h * 2 + (kb + 4) / 16;
+      const int gb = ((kb + 4) % 16) / 4;
+      unsigned int* pd = (unsigned int*)&pa[h];
+#pragma unroll
+      for (int pr = 0; pr < 2; ++pr) {
+        pd[pr] = __shfl((int)ppack[st_a][pr], (ga << 4) | my_q, 64);
+        pd[2 + pr] = __shfl((int)ppack[st_b][pr], (gb << 4) | my_q, 64);
+      }
+    }
+    // rescale O by alpha of row q' = (lane>>4)*4+i (held by lane q')
+#pragma unroll
+    for (int ct = 0; ct < CTILES; ++ct) {
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        const float al = __shfl(alpha, (lane >> 4) * 4 + i, 64);
+        o_acc[ct][i] *= al;
+      }
+    }
+#pragma unroll
+    for (int ct = 0; ct < CTILES; ++ct) {
+      const int dim = ct * 16 + (lane & 15);
+#pragma unroll
+      for (int h = 0; h < HCH; ++h) {
+        bf16x8 vb;
+        const int kk0 = h * 32 + (lane >> 4) * 8;
+#pragma unroll
+        for (int i = 0; i < 8; ++i) vb[i] = v_lds[kk0 + i][dim];
+        o_acc[ct] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            pa[h], vb, o_acc[ct], 0, 0, 0);
+      }
+    }
+    __syncthreads();
+  }
+
+  // ---- epilogue: normalize + store (invl of row q' via shfl)
+  {
+#pragma unroll
+    for (int ct = 0; ct < CTILES; ++ct) {
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        const int row = (lane >> 4) * 4 + i;
+        const float l = __shfl(run_l, row, 64);
+        const float invl = (l > 0.f) ? 1.f / l : 0.f;
+        const int row_global = wave * 16 + row;
+        const int pos_local = tile * POS_PER_WG + row_global / G;
+        if (pos_local >= q_len) continue;
+        const int head = row_global % G;
+        const int tok = q_start + pos_local;
+        const int dim = ct * 16 + (lane & 15);
+        out[((long)tok * Hq + kvh * G + head) * D + dim] =
+            f2bf(o_acc[ct][i] * invl);
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
 // Prefill (MFMA) v1 — KSTEP=32 reference implementation
 // ---------------------------------------------------------------------------
 
@@ -942,6 +1195,28 @@ void attn_prefill_paged(torch::Tensor out, torch::Tensor q, torch::Tensor cu_q,
       hipLaunchKernelGGL((attn_prefill_kernel_v2<g, d, 32>), grid, block, 0,  \
                          stream, (short*)out.data_ptr(),                      \
                          (const short*)q.data_ptr(),                          \
+                         (const int*)cu_q.data_ptr(),                         \
+                         (const long*)q_pos.data_ptr(),                       \
+                         (const short*)kcache.data_ptr(),                     \
+                         (const short*)vcache.data_ptr(),                     \
+                         (const int*)block_tables.data_ptr(),                 \
+                         (const int*)kv_lens.data_ptr(), max_blocks, Hkv,     \
+                         (float)scale, q_tstride);                            \
+    else if (swz == 4)                                                        \
+      hipLaunchKernelGGL((attn_prefill_kernel_v5<g, d, 32>), grid, block, 0,  \
+                         stream, (short*)out.data_ptr(),                      \
+                         (const short*)q.data_ptr(),                         \
+                         (const int*)cu_q.data_ptr(),                         \
+                         (const long*)q_pos.data_ptr(),                       \
+                         (const short*)kcache.data_ptr(),                     \
+                         (const short*)vcache.data_ptr(),                     \
+                         (const int*)block_tables.data_ptr(),                 \
+                         (const int*)kv_lens.data_ptr(), max_blocks, Hkv,     \
+                         (float)scale, q_tstride);                            \
+    else if (swz == 5)                                                        \
+      hipLaunchKernelGGL((attn_prefill_kernel_v5<g, d, 64>), grid, block, 0,  \
+                         stream, (short*)out.data_ptr(),                      \
+                         (const short*)q.data_ptr(),                         \
                          (const int*)cu_q.data_ptr(),                         \
                          (const long*)q_pos.data_ptr(),                       \
                          (const short*)kcache.data_ptr(),                     \

@@ -36,7 +36,7 @@ def probe(N=4096, Hq=32, Hkv=8, D=128, iters=20):
     # within-probe interleaved A/B (guide §5.4 rule 24): 6 rounds each
     # variants: 0 = v1 (KSTEP=32), 2 = v2 (KSTEP=64 + reg-staged
     # pipeline + transposed-V LDS)
-    for swz in (0, 2, 3, 0, 2, 3):
+    for swz in (0, 4, 5, 0, 4, 5):
         for _ in range(2):
             ext.attn_prefill_paged(out, q, cu_q, q_pos, kc, vc, bt, kvl,
                                    scale, swz)
@@ -53,16 +53,16 @@ def probe(N=4096, Hq=32, Hkv=8, D=128, iters=20):
     out0 = torch.empty_like(q)
     out1 = torch.empty_like(q)
     ext.attn_prefill_paged(out0, q, cu_q, q_pos, kc, vc, bt, kvl, scale, 0)
-    ext.attn_prefill_paged(out1, q, cu_q, q_pos, kc, vc, bt, kvl, scale, 3)
+    ext.attn_prefill_paged(out1, q, cu_q, q_pos, kc, vc, bt, kvl, scale, 4)
     max_err = float((out0.float() - out1.float()).abs().max())
     print(
         json.dumps(
             {
                 "probe": f"prefill_attn_N{N}",
                 "TF_v1": results[0],
-                "TF_v2k64": results[2],
-                "TF_v2k32": results.get(3),
-                "v1_v2_max_abs_err": max_err,
+                "TF_v5k32": results.get(4),
+                "TF_v5k64": results.get(5),
+                "v1_v5_max_abs_err": max_err,
             }
         )
     )
