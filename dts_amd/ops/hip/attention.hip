@@ -733,9 +733,12 @@ attn_prefill_kernel_v5(short* __restrict__ out,      // [Tq, Hq, D]
     bf16x8 pa[HCH];
 #pragma unroll
     for (int h = 0; h < HCH; ++h) {
-      // global key base for this lane's A-frag: h*32 + 8*(lane>>4)
-      // keys 8g..8g+3 come from tile st_a, group ga; keys 8g+4..8g+7
-      // from the NEXT group (same or next tile)
+      // global key base for this lane's A-frag: h*32 + 8*(lane>>4).
+      // Keys kb..kb+3 live in score tile st_a / lane-group ga, keys
+      // kb+4..kb+7 in st_b / gb. A shfl OPERAND must be indexed
+      // uniformly across lanes (each lane evaluates its own
+      // expression), so gather every tile with uniform indices and
+      // select by this lane's st_a/st_b.
       const int kb = 8 * (lane >> 4);     // within the 32-key chunk
       const int st_a = h * 2 + kb / 16;   // score tile holding kb..kb+3
       const int ga = (kb % 16) / 4;       // lane-group that owns them
@@ -744,8 +747,18 @@ attn_prefill_kernel_v5(short* __restrict__ out,      // [Tq, Hq, D]
       unsigned int* pd = (unsigned int*)&pa[h];
 #pragma unroll
       for (int pr = 0; pr < 2; ++pr) {
-        pd[pr] = __shfl((int)ppack[st_a][pr], (ga << 4) | my_q, 64);
-        pd[2 + pr] = __shfl((int)ppack[st_b][pr], (gb << 4) | my_q, 64);
+        unsigned int va = 0, vb2 = 0;
+#pragma unroll
+        for (int st = 0; st < STILES; ++st) {
+          const unsigned int ta =
+              (unsigned int)__shfl((int)ppack[st][pr], (ga << 4) | my_q, 64);
+          const unsigned int tb =
+              (unsigned int)__shfl((int)ppack[st][pr], (gb << 4) | my_q, 64);
+          if (st == st_a) va = ta;
+          if (st == st_b) vb2 = tb;
+        }
+        pd[pr] = va;
+        pd[2 + pr] = vb2;
       }
     }
     // rescale O by alpha of row q' = (lane>>4)*4+i (held by lane q')
