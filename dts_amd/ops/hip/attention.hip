@@ -744,7 +744,7 @@ attn_prefill_kernel_v5(short* __restrict__ out,      // [Tq, Hq, D]
       const int ga = (kb % 16) / 4;       // lane-group that owns them
       const int st_b = h * 2 + (kb + 4) / 16;
       const int gb = ((kb + 4) % 16) / 4;
-      unsigned int* pd = (unsigned int*)&pa[h];
+      unsigned int paw[4];
 #pragma unroll
       for (int pr = 0; pr < 2; ++pr) {
         unsigned int va = 0, vb2 = 0;
@@ -757,8 +757,13 @@ attn_prefill_kernel_v5(short* __restrict__ out,      // [Tq, Hq, D]
           if (st == st_a) va = ta;
           if (st == st_b) vb2 = tb;
         }
-        pd[pr] = va;
-        pd[2 + pr] = vb2;
+        paw[pr] = va;
+        paw[2 + pr] = vb2;
+      }
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        pa[h][2 * j] = (short)(paw[j] & 0xffff);
+        pa[h][2 * j + 1] = (short)(paw[j] >> 16);
       }
     }
     // rescale O by alpha of row q' = (lane>>4)*4+i (held by lane q')

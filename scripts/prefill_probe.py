@@ -33,9 +33,19 @@ def probe(N=4096, Hq=32, Hkv=8, D=128, iters=20):
     scale = D ** -0.5
     flops = 2 * 2 * Hq * (N * N / 2) * D
     results = {}
+    # numerics FIRST (and flushed) so a kernel fault localizes to its
+    # variant instead of discarding block-buffered timing output
+    out_ref = torch.empty_like(q)
+    ext.attn_prefill_paged(out_ref, q, cu_q, q_pos, kc, vc, bt, kvl, scale, 0)
+    torch.cuda.synchronize()
+    print(f"numerics N={N}: v1 ok", flush=True)
+    for var in (4, 5):
+        out_v = torch.empty_like(q)
+        ext.attn_prefill_paged(out_v, q, cu_q, q_pos, kc, vc, bt, kvl, scale, var)
+        torch.cuda.synchronize()
+        err = float((out_ref.float() - out_v.float()).abs().max())
+        print(f"numerics N={N}: variant {var} max_abs_err={err}", flush=True)
     # within-probe interleaved A/B (guide §5.4 rule 24): 6 rounds each
-    # variants: 0 = v1 (KSTEP=32), 2 = v2 (KSTEP=64 + reg-staged
-    # pipeline + transposed-V LDS)
     for swz in (0, 4, 5, 0, 4, 5):
         for _ in range(2):
             ext.attn_prefill_paged(out, q, cu_q, q_pos, kc, vc, bt, kvl,
@@ -55,7 +65,7 @@ def probe(N=4096, Hq=32, Hkv=8, D=128, iters=20):
     ext.attn_prefill_paged(out0, q, cu_q, q_pos, kc, vc, bt, kvl, scale, 0)
     ext.attn_prefill_paged(out1, q, cu_q, q_pos, kc, vc, bt, kvl, scale, 4)
     max_err = float((out0.float() - out1.float()).abs().max())
-    print(
+    print(  # noqa
         json.dumps(
             {
                 "probe": f"prefill_attn_N{N}",
@@ -64,7 +74,8 @@ def probe(N=4096, Hq=32, Hkv=8, D=128, iters=20):
                 "TF_v5k64": results.get(5),
                 "v1_v5_max_abs_err": max_err,
             }
-        )
+        ),
+        flush=True,
     )
 
 
