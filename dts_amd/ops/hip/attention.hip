@@ -1159,23 +1159,19 @@ void attn_prefill_paged(torch::Tensor out, torch::Tensor q, torch::Tensor cu_q,
                         torch::Tensor q_pos, torch::Tensor kcache,
                         torch::Tensor vcache, torch::Tensor block_tables,
                         torch::Tensor kv_lens, double scale, int64_t swz) {
-  // variant select: -1 (default) -> v2 unless DTS_PREFILL_V1=1;
-  // 0 -> v1 no-swizzle, 1 -> v1 swizzled (measured -24..-30%, kept for
-  // re-runs), 2 -> v2 (KSTEP=64, reg-staged pipeline, transposed-V LDS)
+  // variant select via DTS_PREFILL_V: 0 -> v1 (KSTEP=32 LDS softmax),
+  // 1 -> v1+XOR swizzle (-24..-30%), 2/3 -> v2 ablation (reg-staged +
+  // transposed-V; SLOWER, profiles/prefill_v2_ablation_r2.md),
+  // 4/5 -> v5 swapped-QK^T in-register softmax (KSTEP 32/64).
+  // Default: v5 KSTEP=64 — measured 149/195/242 TF at N=2k/4k/8k vs
+  // v1's 140/182/220 (PMC-guided: v1 parked 43% of wave cycles on the
+  // softmax LDS round-trip fences).
   if (swz < 0) {
     static int env_v = [] {
       const char* e = getenv("DTS_PREFILL_V");
       return e ? atoi(e) : -1;
     }();
-    // default: v1. The v2 ladder steps measured SLOWER on hardware —
-    // v1 147/179/220 TF at N=2k/4k/8k vs v2(KSTEP=64) 127/167/194 and
-    // v2(KSTEP=32, same LDS budget as v1) 119/156/184
-    // (profiles/prefill_v2_ablation_r2.md): at 5 WGs/CU the cooperative
-    // loads are already TLP-hidden, so the reg-staged pipeline only adds
-    // VGPR pressure, and the transposed-V image trades scalar reads for
-    // 4-way-conflicted scalar stores. Kept behind DTS_PREFILL_V=2/3 as
-    // the measured ablation.
-    swz = (env_v >= 0) ? env_v : 0;
+    swz = (env_v >= 0) ? env_v : 5;
   }
   const int Hq = q.size(1), D = q.size(2);
   const int Hkv = kcache.size(1);
