@@ -267,3 +267,54 @@ class TestSchedulerDifferential:
                 )
             )
         assert results[0] == results[1]
+
+
+class TestSpecUnderPressure:
+    """Drafts must not break KV accounting under pool pressure: draft
+    rows allocate blocks ahead of acceptance, preemption recomputes, and
+    the stream must still equal the spec-off run exactly."""
+
+    @pytest.mark.parametrize("use_native", [False, True])
+    def test_spec_with_preemption_matches(self, use_native, monkeypatch):
+        if use_native:
+            from dts_amd.core import load_core
+
+            if load_core() is None:
+                pytest.skip("native core not built")
+        monkeypatch.setenv("DTS_NATIVE_CORE", "1" if use_native else "0")
+        from dts_amd.serving.engine import ServingEngine
+
+        def run(spec_k):
+            eng = ServingEngine(
+                model_name="llama-tiny",
+                device="cpu",
+                dtype=torch.float32,
+                num_blocks=48,  # tight pool: forces preemption
+                block_size=4,
+                weight_seed=11,
+                spec_k=spec_k,
+            )
+            # repetitive prompts → drafts propose; several seqs compete
+            prompts = [
+                [300 + (i % 6) for i in range(24)],
+                [400 + (i % 5) for i in range(20)],
+                [500 + (i % 4) for i in range(16)],
+            ]
+            futs = [
+                eng.submit_tokens(
+                    list(p),
+                    SamplingParams(max_tokens=24, temperature=0.0, seed=None),
+                )
+                for p in prompts
+            ]
+            eng.run_until_idle()
+            outs = [f.result(timeout=60).token_ids for f in futs]
+            free = eng.scheduler.num_free() if hasattr(
+                eng.scheduler, "num_free") else eng.block_manager.num_free()
+            return outs, eng.spec_draft_tokens, free
+
+        ref, _, _ = run(0)
+        out, drafted, free = run(4)
+        assert out == ref
+        assert drafted > 0
+        assert free == 47  # whole pool drained (1 block is scratch-reserved)
