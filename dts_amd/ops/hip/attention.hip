@@ -104,60 +104,31 @@ attn_decode_kernel(float* __restrict__ ws_m,  // [B, Hkv, S, G]
 
   const int* bt = block_tables + (long)seq * max_blocks;
 
-  // software pipeline over this wave's owned pages: the grid is
-  // GRID-limited at short kv (B*Hkv*S workgroups ~ 1.5 waves/SIMD), so
-  // the serial K-load -> scores -> V-load chain exposed a full HBM
-  // latency per page; prefetching the NEXT owned page's K/V words into
-  // spare registers hides it (VGPRs are free at this occupancy)
-  int kw_cur[DPL / 2], vw_cur[BS], kw_nxt[DPL / 2], vw_nxt[BS];
-  auto load_page = [&](int page, int kw[], int vw[]) {
-    const long blk = bt[page];
-    const short* kbase = kcache + ((blk * Hkv + kvh) * BS) * (long)D;
-    const short* vbase = vcache + ((blk * Hkv + kvh) * BS) * (long)D;
-    const int valid = min(BS, kv_len - page * BS);
-    if (key_of_lane < valid) {
-      const short* kp = kbase + key_of_lane * D + c * DPL;
-#pragma unroll
-      for (int i = 0; i < DPL / 2; ++i) kw[i] = ((const int*)kp)[i];
-    } else {
-#pragma unroll
-      for (int i = 0; i < DPL / 2; ++i) kw[i] = 0;
-    }
-#pragma unroll
-    for (int j = 0; j < BS; ++j) {
-      if (j < valid) {
-        if constexpr (DPV == 2)
-          vw[j] = *(const int*)(vbase + j * D + 2 * lane);
-        else
-          vw[j] = (int)*(const unsigned short*)(vbase + j * D + lane);
-      } else {
-        vw[j] = 0;
-      }
-    }
-  };
-
-  if (page_lo + wave < page_hi) load_page(page_lo + wave, kw_cur, vw_cur);
-
   for (int page = page_lo + wave; page < page_hi; page += NWAVE) {
-    const int nxt = page + NWAVE;
-    if (nxt < page_hi) load_page(nxt, kw_nxt, vw_nxt);
+    const long blk = bt[page];
+    const short* kbase =
+        kcache + ((blk * Hkv + kvh) * BS) * (long)D;
+    const short* vbase =
+        vcache + ((blk * Hkv + kvh) * BS) * (long)D;
     const int valid = min(BS, kv_len - page * BS);
 
-    // ---- K phase: scores for the 16 keys of this page (zero-filled
-    // K words of invalid keys contribute 0 and are masked at the write)
+    // ---- K phase: scores for the 16 keys of this page
     float partial[G];
 #pragma unroll
     for (int g = 0; g < G; ++g) partial[g] = 0.f;
+    if (key_of_lane < valid) {
+      const short* kp = kbase + key_of_lane * D + c * DPL;
 #pragma unroll
-    for (int i = 0; i < DPL / 2; ++i) {
-      int kw = kw_cur[i];
-      float k0 = bf2f((short)(kw & 0xffff));
-      float k1 = bf2f((short)((kw >> 16) & 0xffff));
+      for (int i = 0; i < DPL / 2; ++i) {
+        int kw = ((const int*)kp)[i];
+        float k0 = bf2f((short)(kw & 0xffff));
+        float k1 = bf2f((short)((kw >> 16) & 0xffff));
 #pragma unroll
-      for (int g = 0; g < G; ++g) {
-        float q0 = bf2f((short)(qp_[g][i] & 0xffff));
-        float q1 = bf2f((short)((qp_[g][i] >> 16) & 0xffff));
-        partial[g] += q0 * k0 + q1 * k1;
+        for (int g = 0; g < G; ++g) {
+          float q0 = bf2f((short)(qp_[g][i] & 0xffff));
+          float q1 = bf2f((short)((qp_[g][i] >> 16) & 0xffff));
+          partial[g] += q0 * k0 + q1 * k1;
+        }
       }
     }
     // reduce over the 4 lanes of the key group
@@ -172,7 +143,21 @@ attn_decode_kernel(float* __restrict__ ws_m,  // [B, Hkv, S, G]
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
     __builtin_amdgcn_wave_barrier();
 
-    const int* vw = vw_cur;
+    // ---- batch the page's V words up front: 16 independent loads in
+    // flight at once (the per-(g,j) load placement serialized an L2/HBM
+    // round trip into the online-softmax chain — ~5 us/page at long kv)
+    int vw[BS];
+#pragma unroll
+    for (int j = 0; j < BS; ++j) {
+      if (j < valid) {
+        if constexpr (DPV == 2)
+          vw[j] = *(const int*)(vbase + j * D + 2 * lane);
+        else
+          vw[j] = (int)*(const unsigned short*)(vbase + j * D + lane);
+      } else {
+        vw[j] = 0;
+      }
+    }
 
     // ---- softmax update + V accumulate; lane owns output dims {2l, 2l+1}
 #pragma unroll
@@ -204,12 +189,6 @@ attn_decode_kernel(float* __restrict__ ws_m,  // [B, Hkv, S, G]
       }
     }
     __builtin_amdgcn_wave_barrier();
-    if (nxt < page_hi) {
-#pragma unroll
-      for (int i = 0; i < DPL / 2; ++i) kw_cur[i] = kw_nxt[i];
-#pragma unroll
-      for (int j = 0; j < BS; ++j) vw_cur[j] = vw_nxt[j];
-    }
   }
 
   // ---- cross-wave combine via LDS, then publish split partials
