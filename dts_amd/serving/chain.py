@@ -29,8 +29,6 @@ Correctness notes:
 
 from __future__ import annotations
 
-from typing import Optional
-
 import torch
 
 from dts_amd import ops
