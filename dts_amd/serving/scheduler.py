@@ -92,7 +92,13 @@ class Scheduler:
         return True
 
     def _prompt_key(self, seq: Sequence) -> int:
-        return hash(tuple(seq.tokens[: seq.num_prompt_tokens]))
+        # prompts are immutable — cache (this runs per waiting seq per
+        # schedule call)
+        k = getattr(seq, "_prompt_key_c", None)
+        if k is None:
+            k = hash(tuple(seq.tokens[: seq.num_prompt_tokens]))
+            seq._prompt_key_c = k  # type: ignore[attr-defined]
+        return k
 
     def _prefix_key(self, seq: Sequence) -> int:
         """First-512-token key: requests sharing a long prompt prefix
@@ -101,7 +107,11 @@ class Scheduler:
         hit instead of all prefilling the shared 10k tokens in parallel.
         Branch rollout prompts diverge well before 512 tokens (strategy
         text), so this only serializes genuinely shared prefixes."""
-        return hash(tuple(seq.tokens[: min(512, seq.num_prompt_tokens)]))
+        k = getattr(seq, "_prefix_key_c", None)
+        if k is None:
+            k = hash(tuple(seq.tokens[: min(512, seq.num_prompt_tokens)]))
+            seq._prefix_key_c = k  # type: ignore[attr-defined]
+        return k
 
     def has_work(self) -> bool:
         return bool(self.waiting or self.running)
