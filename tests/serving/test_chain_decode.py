@@ -171,3 +171,61 @@ class TestChainHooks:
         assert not sched.reserve_tokens(seq, 10_000)
         # and a normal small reserve still succeeds afterwards
         assert sched.reserve_tokens(seq, len(seq.tokens) + 4)
+
+
+class TestChainEligibility:
+    """ChainRunner.eligible() logic with a stub graph runner (CPU)."""
+
+    def _mk(self):
+        from dts_amd.serving.chain import ChainRunner
+
+        class _GR:
+            def __init__(self):
+                self.kv_pool = type("P", (), {"block_size": 16})()
+
+            def can_run(self, batch):
+                return batch.num_decode_seqs <= 16
+
+        r = ChainRunner.__new__(ChainRunner)  # skip cuda Event alloc
+        r.gr = _GR()
+        r.device = "cpu"
+        return r
+
+    def _batch(self, n_seqs, n_rows=None, guides=None, drafts=None):
+        from dts_amd.serving.batch import ForwardBatch
+        from dts_amd.llm.types import SamplingParams
+
+        seqs = []
+        for i in range(n_seqs):
+            s = Sequence(tokens=[1, 2, 3], params=SamplingParams())
+            s.guide = guides[i] if guides else None
+            seqs.append(s)
+        b = ForwardBatch(
+            token_ids=torch.zeros(n_rows or n_seqs, dtype=torch.long),
+            positions=torch.zeros(n_rows or n_seqs, dtype=torch.long),
+            slot_mapping=torch.zeros(n_rows or n_seqs, dtype=torch.long),
+            num_decode_seqs=n_rows or n_seqs,
+        )
+        b._sampled_seqs = seqs
+        b._spec_drafts = drafts or {}
+        return b
+
+    def test_plain_decode_eligible(self):
+        r = self._mk()
+        assert r.eligible(self._batch(4))
+
+    def test_guided_blocks_chain(self):
+        r = self._mk()
+        g = object()
+        assert not r.eligible(self._batch(2, guides=[None, g]))
+
+    def test_draft_rows_block_chain(self):
+        r = self._mk()
+        b = self._batch(2, n_rows=5, drafts={1: [7, 8, 9]})
+        assert not r.eligible(b)
+
+    def test_prefill_blocks_chain(self):
+        r = self._mk()
+        b = self._batch(2)
+        b.num_prefill_seqs = 1
+        assert not r.eligible(b)
