@@ -97,9 +97,15 @@ class MoEMLP(nn.Module):
         inv.scatter_(
             0, order, torch.arange(P, device=x.device, dtype=order.dtype)
         )
-        counts = torch.bincount(
-            flat_expert, minlength=self.num_experts
-        ).to(torch.int32)
+        # capture-safe histogram: torch.bincount syncs the stream
+        # internally (aborted hipGraph capture mid-bench; caught by
+        # test_grouped_path_is_capture_safe) — index_add_ does not
+        counts = torch.zeros(
+            self.num_experts, dtype=torch.int32, device=x.device
+        )
+        counts.index_add_(
+            0, flat_expert, torch.ones_like(flat_expert, dtype=torch.int32)
+        )
         offsets = (torch.cumsum(counts, 0) - counts).to(torch.int32)
         x_sorted = x.index_select(0, flat_tok.index_select(0, order))
         gu = ops.moe_grouped_linear(x_sorted, self.gate_up_w, counts, offsets)
