@@ -244,3 +244,62 @@ class TestOversubscribedPool:
         # (the engine reserves one scratch block for graph padding)
         assert eng.cache_stats["free_blocks"] >= 63
         eng.stop()
+
+
+class TestGraphCaptureFallback:
+    """A capture failure must disable graphs and degrade to eager, not
+    fail pending request futures (the round-2 Mixtral bincount abort
+    killed every judge call before this guard existed)."""
+
+    class _FailingGR:
+        def __init__(self):
+            self.calls = 0
+
+        def can_run(self, batch):
+            return True
+
+        def run(self, batch):
+            self.calls += 1
+            raise RuntimeError(
+                "HIP error: operation not permitted when stream is capturing"
+            )
+
+    def _engine(self):
+        return ServingEngine(
+            model_name="llama-tiny",
+            device="cpu",
+            dtype=torch.float32,
+            num_blocks=256,
+            block_size=4,
+            max_batch_tokens=128,
+            weight_seed=1,
+        )
+
+    def test_capture_error_degrades_to_eager(self):
+        eng = self._engine()
+        gr = self._FailingGR()
+        eng._graph_runner = gr
+        eng._chain = None
+        fut = eng.submit_tokens([1, 2, 3, 4], SamplingParams(max_tokens=4, seed=0))
+        eng.run_until_idle()
+        res = fut.result(timeout=5)  # request survives the failed capture
+        assert res.completion_tokens >= 1
+        assert gr.calls == 1
+        assert eng._graph_runner is None  # graphs disabled after failure
+        assert eng.cache_stats["eager_decode_steps"] >= 1
+        eng.stop()
+
+    def test_non_capture_runtime_error_propagates(self):
+        class _OtherGR(self._FailingGR):
+            def run(self, batch):
+                raise RuntimeError("out of memory")
+
+        eng = self._engine()
+        eng._graph_runner = _OtherGR()
+        eng._chain = None
+        eng.submit_tokens([1, 2, 3], SamplingParams(max_tokens=2, seed=0))
+        # only capture errors are swallowed; anything else surfaces
+        # (here synchronously; in the background loop via failed futures)
+        with pytest.raises(RuntimeError, match="out of memory"):
+            eng.run_until_idle()
+        eng.stop()
