@@ -195,3 +195,76 @@ class TestReasoningEnabled:
         # the instruction reached the model
         sys_prompts = [c["system"] for c in backend.calls if "[dts:assistant]" in c["system"]]
         assert sys_prompts and "<think>" in sys_prompts[-1]
+
+
+class TestTaskBudget:
+    """Per-task wall budget (ref simulator.py:199-214): a hung expansion
+    times out without sinking the batch."""
+
+    def test_hung_expansion_times_out(self, run_async, monkeypatch):
+        import asyncio
+
+        import dts_amd.search.simulator as sim_mod
+
+        monkeypatch.setattr(sim_mod, "TASK_TIMEOUT_S", 0.2)
+
+        class _HangBackend(FakeBackend):
+            def __init__(self):
+                super().__init__()
+                self.n = 0
+
+            async def chat(self, messages, params, model=None):
+                self.n += 1
+                if self.n == 1:  # first fork's rephrase call hangs
+                    await asyncio.sleep(30)
+                return await super().chat(messages, params, model)
+
+        sim = make_sim(_HangBackend())
+        root = DialogueNode(id=generate_node_id(), messages=[Message.user("s")])
+        tree = DialogueTree.create(root)
+        a, b = make_node("first branch"), make_node("second branch")
+        tree.add_child(root.id, a)
+        tree.add_child(root.id, b)
+
+        intent2 = UserIntent(
+            id="i2",
+            label="L2",
+            description="D2",
+            emotional_tone="curious",
+            cognitive_stance="probing",
+        )
+
+        async def gen_intents(history, count):
+            return [INTENT, intent2][:count]
+
+        async def run():
+            return await asyncio.wait_for(
+                sim.expand_nodes(
+                    [a, b],
+                    turns=1,
+                    intents_per_node=2,
+                    tree=tree,
+                    generate_intents=gen_intents,
+                ),
+                timeout=20,
+            )
+
+        expanded = run_async(run())
+        # 4 fork tasks; the hung one dropped, the healthy ones completed
+        assert len(expanded) == 3
+
+
+class TestTerminalExclusion:
+    """Terminal/pruned nodes never re-expand (ref tree.py:85-89
+    active_leaves filters by NodeStatus.ACTIVE)."""
+
+    def test_active_leaves_excludes_terminal_and_pruned(self):
+        root = DialogueNode(id=generate_node_id(), messages=[Message.user("s")])
+        tree = DialogueTree.create(root)
+        kids = [make_node(f"k{i}") for i in range(3)]
+        for k in kids:
+            tree.add_child(root.id, k)
+        kids[0].status = NodeStatus.TERMINAL
+        kids[1].status = NodeStatus.PRUNED
+        leaves = tree.active_leaves()
+        assert leaves == [kids[2]]
