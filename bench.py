@@ -130,6 +130,9 @@ def main():
     else:
         kv_bytes = 512 << 20
 
+    if args.judge_model == args.model:
+        args.judge_model = None  # same model: one engine serves both roles
+
     # KV sizing: each engine measures free memory AFTER its own weights
     # load and takes kv_frac of it (a pre-weights split over-committed:
     # 70B weights + 0.75-of-free pool exceeded the 288 GB card). For
