@@ -213,3 +213,130 @@ def test_core_api_tolerates_bad_ids():
     cs.abort(99)  # unknown ids: no-op
     cs.finish(42)
     assert cs.num_free() == 8
+
+
+@pytest.mark.parametrize("seed", range(4))
+def test_lockstep_full_hook_surface(seed):
+    """Randomized differential over the ENTIRE round-2 scheduler API:
+    chunked prefill, speculative draft rows with partial acceptance,
+    chained-decode bursts (reserve/chain_advance/set_accepted(0)),
+    512-token shared-prefix holdback, forced extensions and finishes —
+    every call mirrored into both schedulers, every observable compared."""
+    rng = random.Random(1000 + seed)
+    py = Scheduler(
+        BlockManager(256, 8), max_batch_tokens=96, spec_k=3, max_spec_rows=8
+    )
+    nat = NativeScheduler(
+        256, 8, max_batch_tokens=96, spec_k=3, max_spec_rows=8
+    )
+    shared_prefix = [(i * 11) % 97 for i in range(516)]
+    live = []
+
+    def new_seq(tokens):
+        sp = Sequence(tokens=list(tokens), params=SamplingParams(max_tokens=64))
+        sn = Sequence(tokens=list(tokens), params=SamplingParams(max_tokens=64))
+        sn.seq_id = sp.seq_id
+        py.add(sp)
+        nat.add(sn)
+        live.append((sp, sn))
+
+    def spec_view(b):
+        if b is None:
+            return None
+        groups = getattr(b, "_row_groups", None)
+        return (
+            [(s.seq_id, n) for s, n in groups] if groups else None,
+            dict(getattr(b, "_spec_drafts", None) or {}),
+            [int(x) for x in getattr(b, "_sample_pos", None) or []],
+        )
+
+    for step in range(90):
+        r = rng.random()
+        if r < 0.22 and len(live) < 5:
+            if r < 0.06:
+                # shared long prefix: triggers the 512-token holdback
+                tail = [rng.randrange(100, 200) for _ in range(rng.randrange(2, 9))]
+                new_seq(shared_prefix + tail)
+            else:
+                n = rng.randrange(3, 30)
+                base = rng.randrange(0, 60)
+                # repetitive tail so the bigram proposer fires
+                new_seq(([base, base + 1, base + 2] * 12)[:n])
+        ba, bb = py.schedule(), nat.schedule()
+        assert batches_equal(ba, bb), f"seed {seed} diverged at step {step}"
+        assert spec_view(ba) == spec_view(bb), f"seed {seed} spec diverged at {step}"
+        assert py.waiting_count() == nat.waiting_count()
+        if ba is None:
+            if not live:
+                break
+            continue
+        # mirror the engine: appends per row group FIRST, then advance
+        groups = getattr(ba, "_row_groups", None)
+        sampled = list(zip(ba._sampled_seqs, bb._sampled_seqs))
+        assert [a.seq_id for a, _ in sampled] == [b.seq_id for _, b in sampled]
+        if groups is None:
+            for sp, sn in sampled:
+                tok = rng.randrange(100, 200)
+                py.append_token(sp, tok)
+                nat.append_token(sn, tok)
+        else:
+            drafts = dict(getattr(ba, "_spec_drafts", None) or {})
+            nat_by_id = {s.seq_id: s for _, s in sampled}
+            for sp, n_rows in groups:
+                sn = nat_by_id[sp.seq_id]
+                if n_rows == 1:
+                    tok = rng.randrange(100, 200)
+                    py.append_token(sp, tok)
+                    nat.append_token(sn, tok)
+                else:
+                    draft = drafts[sp.seq_id]
+                    a = rng.randrange(0, len(draft) + 1)  # accepted prefix
+                    toks = list(draft[:a])
+                    if a < len(draft):
+                        toks.append((draft[a] + 1) % 500 + 1)
+                    else:
+                        toks.append(rng.randrange(100, 200))
+                    emitted = len(toks)  # a+1, per the engine's walk
+                    for t in toks:
+                        py.append_token(sp, t)
+                        nat.append_token(sn, t)
+                    py.set_accepted(sp, emitted)
+                    nat.set_accepted(sn, emitted)
+        py.advance_computed(ba)
+        nat.advance_computed(bb)
+        # chained-decode burst on one running sampled seq
+        if sampled and rng.random() < 0.25:
+            sp, sn = rng.choice(sampled)
+            if sp.status.value == "running" and sn.status.value == "running":
+                burst = rng.randrange(2, 10)
+                okp = py.reserve_tokens(sp, len(sp.tokens) + burst)
+                okn = nat.reserve_tokens(sn, len(sn.tokens) + burst)
+                assert okp == okn, f"seed {seed} reserve diverged at {step}"
+                if okp:
+                    for _ in range(burst):
+                        t = rng.randrange(100, 200)
+                        py.chain_advance(sp, t)
+                        nat.chain_advance(sn, t)
+                    py.set_accepted(sp, 0)
+                    nat.set_accepted(sn, 0)
+        # occasional forced extension (guided-form skeleton tokens)
+        if sampled and rng.random() < 0.15:
+            sp, sn = rng.choice(sampled)
+            if sp.status.value == "running" and sn.status.value == "running":
+                forced = [rng.randrange(200, 300) for _ in range(rng.randrange(1, 5))]
+                py.extend_tokens(sp, forced)
+                nat.extend_tokens(sn, forced)
+        # per-seq observable state must agree every step
+        for sp, sn in live:
+            assert len(sp.tokens) == len(sn.tokens)
+            assert py.num_computed_of(sp) == nat.num_computed_of(sn), (
+                f"seed {seed} num_computed diverged at {step} for {sp.seq_id}"
+            )
+        # finishes
+        for sp, sn in list(live):
+            if len(sp.output_tokens) > rng.randrange(10, 60):
+                py.finish(sp, "stop")
+                nat.finish(sn, "stop")
+                live.remove((sp, sn))
+    assert py.bm.cache_hit_tokens == nat.cache_hit_tokens
+    assert py.bm.cache_miss_tokens == nat.cache_miss_tokens
