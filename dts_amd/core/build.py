@@ -17,13 +17,28 @@ def build(verbose: bool = True) -> Path:
     profiles/asan_core.md records the clean pass."""
     os.environ.setdefault("MAX_JOBS", str(os.cpu_count() or 4))
     asan = os.environ.get("DTS_CORE_ASAN") == "1"
-    build_dir = BUILD_DIR.parent / "build_asan" if asan else BUILD_DIR
+    ubsan = os.environ.get("DTS_CORE_UBSAN") == "1"
+    if asan:
+        build_dir = BUILD_DIR.parent / "build_asan"
+    elif ubsan:
+        build_dir = BUILD_DIR.parent / "build_ubsan"
+    else:
+        build_dir = BUILD_DIR
     build_dir.mkdir(parents=True, exist_ok=True)
     cflags = ["-O3", "-std=c++17"]
     ldflags = []
     if asan:
         cflags += ["-fsanitize=address", "-fno-omit-frame-pointer", "-g"]
         ldflags += ["-fsanitize=address"]
+    if ubsan:
+        # halt on the first UB report so the test run fails loudly
+        cflags += [
+            "-fsanitize=undefined",
+            "-fno-sanitize-recover=all",
+            "-fno-omit-frame-pointer",
+            "-g",
+        ]
+        ldflags += ["-fsanitize=undefined"]
     from torch.utils.cpp_extension import load
 
     load(
