@@ -153,8 +153,22 @@ class FakeBackend:
     def _comparative_judgment(self, user: str) -> str:
         import re
 
-        ids = re.findall(r"--- Trajectory ([0-9a-f-]+)", user)
-        order = sorted(ids, key=lambda i: _stable_hash(self.score_salt + i), reverse=True)
+        # rank by trajectory CONTENT (uuid only as tiebreak): node ids are
+        # random per run, so an id-hash ranking made seeded searches
+        # non-reproducible run to run (caught by the latency-invariance
+        # test — a content-blind judge is also just a worse fake)
+        parts = re.split(r"--- Trajectory ([0-9a-f-]+)", user)
+        ids = parts[1::2]
+        bodies = parts[2::2]
+        content = dict(zip(ids, bodies))
+        order = sorted(
+            ids,
+            key=lambda i: (
+                _stable_hash(self.score_salt + content.get(i, "")),
+                i,
+            ),
+            reverse=True,
+        )
         ranking = []
         critiques = {}
         for rank, tid in enumerate(order, start=1):

@@ -350,8 +350,10 @@ class DTSEngine:
     # ------------------------------------------------------------------
     async def _initialize_tree(self) -> DialogueTree:
         cfg = self.config
+        from dts_amd.search.tree import derive_node_id
+
         root = DialogueNode(
-            id=generate_node_id(),
+            id=derive_node_id("dts-root", f"{cfg.goal}::{cfg.first_message}"),
             depth=0,
             messages=[Message.user(cfg.first_message)],
         )
@@ -397,9 +399,11 @@ class DTSEngine:
                 },
             )
 
-        for strategy in strategies:
+        from dts_amd.search.tree import derive_node_id
+
+        for s_idx, strategy in enumerate(strategies):
             child = DialogueNode(
-                id=generate_node_id(),
+                id=derive_node_id(root.id, f"strategy:{s_idx}"),
                 strategy=strategy,
                 messages=[Message.user(cfg.first_message)],
             )

@@ -24,7 +24,7 @@ from dts_amd.llm.types import Message
 from dts_amd.search import prompts
 from dts_amd.search.config import GenerationBudget
 from dts_amd.search.events import create_event_emitter
-from dts_amd.search.tree import DialogueTree, generate_node_id
+from dts_amd.search.tree import DialogueTree, derive_node_id, generate_node_id
 from dts_amd.search.types import DialogueNode, NodeStatus, Strategy, UserIntent
 from dts_amd.utils.logging import log_phase, logger
 
@@ -101,6 +101,7 @@ class ConversationSimulator:
 
         expansion_tasks = []
         fallback_nodes = []
+        creation_order: dict = {}
         for node, intents in zip(nodes, intent_results):
             if isinstance(intents, Exception) or not intents:
                 logger.warning("Intent generation failed for %s; linear expansion", node.id)
@@ -120,7 +121,7 @@ class ConversationSimulator:
                     },
                 )
                 child = DialogueNode(
-                    id=generate_node_id(),
+                    id=derive_node_id(node.id, f"intent:{idx}"),
                     parent_id=node.id,
                     depth=node.depth + 1,
                     strategy=node.strategy,
@@ -130,9 +131,11 @@ class ConversationSimulator:
                 if tree is not None:
                     tree.add_child(node.id, child)
                 expansion_tasks.append(self._expand_with_intent(child, turns, intent))
+                creation_order[child.id] = len(creation_order)
 
         for node in fallback_nodes:
             expansion_tasks.append(self._expand_linear(node, turns))
+            creation_order[node.id] = len(creation_order)
 
         log_phase("FORK", f"Expanding {len(expansion_tasks)} branches...", indent=1)
         total_timeout = TASK_TIMEOUT_S * max(1, len(expansion_tasks))
@@ -154,6 +157,13 @@ class ConversationSimulator:
         except (TimeoutError, asyncio.TimeoutError):
             logger.warning("Expansion batch timed out")
         log_phase("FORK", f"Completed: {len(expanded)} | Failed: {failed}", indent=1)
+        # as_completed yields in COMPLETION order; downstream the sibling
+        # order reaches the comparative-judge prompt, so a latency wiggle
+        # would change judge inputs run to run. Restore creation order —
+        # the reference has the same incidental nondeterminism
+        # (ref simulator.py:199-214); determinism is a deliberate
+        # improvement here (seeded runs must reproduce).
+        expanded.sort(key=lambda n: creation_order.get(n.id, 1 << 30))
         return expanded
 
     async def _expand_linear_batch(self, nodes: list, turns: int) -> list:

@@ -23,6 +23,19 @@ def generate_node_id() -> str:
     return str(uuid.uuid4())
 
 
+def derive_node_id(parent_id: str, tag: str) -> str:
+    """Deterministic child id (uuid5 of parent:tag).
+
+    Node ids must be stable across reruns and across DP ranks: sibling
+    order in comparative judging and chunk sharding both sort by id
+    (evaluator.comparative_chunks, dist_engine), so random uuid4 ids made
+    seeded runs irreproducible — which sibling ranked first depended on
+    the draw. The reference uses uuid4 (ref tree.py:20-22); determinism
+    is a deliberate improvement here.
+    """
+    return str(uuid.uuid5(uuid.NAMESPACE_URL, f"{parent_id}:{tag}"))
+
+
 class DialogueTree:
     def __init__(self, root_id: str) -> None:
         self.root_id = root_id

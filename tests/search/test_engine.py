@@ -228,3 +228,59 @@ class TestStrategySplit:
         out = run_async(gen.generate_strategies("hello", 2))
         assert len(out) == 2
         assert len(backend.calls) == 1
+
+
+class TestLatencyInvariance:
+    """The search result must not depend on backend completion ORDER —
+    phase barriers gather in submission order, so arbitrary per-call
+    latency jitter must produce an identical tree (same shape, same
+    scores, same best). Guards against hidden as_completed/ordering
+    dependencies in the asyncio orchestration."""
+
+    def _run_with_jitter(self, run_async, jitter_seed):
+        import asyncio as _aio
+        import random as _rnd
+
+        class JitterBackend(FakeBackend):
+            def __init__(self, seed):
+                super().__init__()
+                self._rng = _rnd.Random(seed)
+
+            async def chat(self, messages, params, model=None):
+                await _aio.sleep(self._rng.random() * 0.02)
+                return await super().chat(messages, params, model)
+
+        cfg = DTSConfig(
+            goal="Plan a product launch",
+            first_message="Where do we start?",
+            init_branches=3,
+            turns_per_branch=2,
+            user_intents_per_branch=2,
+            user_variability=True,
+            scoring_mode="comparative",
+            prune_threshold=0.0,
+            seed=11,
+        )
+        llm = LLM(JitterBackend(jitter_seed), default_model="fake-model")
+        result = run_async(DTSEngine(llm, cfg).run(rounds=1))
+
+        def shape(res):
+            nodes = sorted(
+                (
+                    n.depth,
+                    n.strategy.tagline if n.strategy else None,
+                    n.user_intent.label if n.user_intent else None,
+                    tuple(round(s, 4) for s in n.stats.judge_scores),
+                    len(n.messages),
+                )
+                for n in res.all_nodes
+            )
+            return nodes, round(res.best_score, 4)
+
+        return shape(result)
+
+    def test_identical_tree_under_jitter(self, run_async):
+        a = self._run_with_jitter(run_async, jitter_seed=1)
+        b = self._run_with_jitter(run_async, jitter_seed=2)
+        c = self._run_with_jitter(run_async, jitter_seed=3)
+        assert a == b == c
