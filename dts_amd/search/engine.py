@@ -420,13 +420,23 @@ class DTSEngine:
         if isinstance(checkpoint, str):
             with open(checkpoint) as f:
                 checkpoint = _json.load(f)
-        # deterministic root id: every DP rank loads the same checkpoint
-        # independently and their trees must match node-for-node
-        import uuid as _uuid
+        # recover the ORIGINAL root id: depth-1 branches carry it as their
+        # parent_id, and ids derived from it must keep matching after
+        # resume (resume == uninterrupted run; also every DP rank loads
+        # the checkpoint independently and trees must match node-for-node)
+        from dts_amd.search.tree import derive_node_id
 
-        root_seed = f"{cfg.goal}::{cfg.first_message}::{len(checkpoint.get('branches', []))}"
+        branch_ids = {b["id"] for b in checkpoint.get("branches", [])}
+        root_id = next(
+            (
+                b["parent_id"]
+                for b in checkpoint.get("branches", [])
+                if b.get("parent_id") and b["parent_id"] not in branch_ids
+            ),
+            derive_node_id("dts-root", f"{cfg.goal}::{cfg.first_message}"),
+        )
         root = DialogueNode(
-            id=str(_uuid.uuid5(_uuid.NAMESPACE_URL, root_seed)),
+            id=root_id,
             depth=0,
             messages=[Message.user(cfg.first_message)],
         )

@@ -126,3 +126,37 @@ class TestPeriodicCheckpoint:
         engine, _ = make_engine()
         asyncio.run(engine.run(rounds=1))
         assert not list(tmp_path.glob("*.json"))
+
+
+class TestResumeEquivalence:
+    """With deterministic node ids, interrupt + resume must reproduce an
+    uninterrupted run EXACTLY: same node ids, same trajectories, same
+    scores. (Linear rounds mutate leaves in place — quirk 6 — so 2
+    straight rounds == 1 round + checkpoint + 1 resumed round.)"""
+
+    def _shape(self, result):
+        return sorted(
+            (
+                n.id,
+                n.depth,
+                n.status.value,
+                tuple(m.content for m in n.messages),
+                tuple(n.stats.judge_scores),
+                round(n.stats.aggregated_score, 4),
+            )
+            for n in result.all_nodes
+        )
+
+    @pytest.mark.parametrize("mode", ["absolute", "comparative"])
+    def test_resume_equals_uninterrupted(self, tmp_path, mode):
+        straight_engine, _ = make_engine(scoring_mode=mode)
+        straight = asyncio.run(straight_engine.run(rounds=2))
+
+        first_engine, _ = make_engine(scoring_mode=mode)
+        first = asyncio.run(first_engine.run(rounds=1))
+        path = tmp_path / "ckpt.json"
+        first.save_json(str(path))
+        second_engine, _ = make_engine(scoring_mode=mode)
+        resumed = asyncio.run(second_engine.run(rounds=1, resume_from=str(path)))
+
+        assert self._shape(resumed) == self._shape(straight)
