@@ -121,6 +121,8 @@ class CoreScheduler {
   // ---- sequence lifecycle ------------------------------------------------
   void add(int64_t seq_id, const std::vector<int32_t>& tokens,
            bool allow_spec = true) {
+    if (tokens.empty())  // empty prompts crashed the step path (UB)
+      throw py::value_error("empty prompt: at least one token required");
     Seq s;
     s.id = seq_id;
     s.tokens = tokens;

@@ -230,6 +230,10 @@ class ServingEngine:
         guide: Optional[object] = None,
         stream_cb: Optional[object] = None,
     ) -> concurrent.futures.Future:
+        if not prompt_ids:
+            # an empty prompt segfaulted the native core and raised a raw
+            # IndexError in the Python scheduler — reject it cleanly
+            raise ValueError("empty prompt: at least one token required")
         if guide is not None and hasattr(guide, "token_budget"):
             # a finite form defines its own output size; never let a
             # free-text phase budget truncate it mid-form

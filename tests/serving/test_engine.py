@@ -303,3 +303,37 @@ class TestGraphCaptureFallback:
         with pytest.raises(RuntimeError, match="out of memory"):
             eng.run_until_idle()
         eng.stop()
+
+
+class TestEmptyPrompt:
+    """Empty prompts must be rejected cleanly (an empty prompt
+    segfaulted the native core before the guard)."""
+
+    @pytest.mark.parametrize("native", ["0", "1"])
+    def test_rejected_at_submit(self, native, monkeypatch):
+        monkeypatch.setenv("DTS_NATIVE_CORE", native)
+        eng = ServingEngine(
+            model_name="llama-tiny",
+            device="cpu",
+            dtype=torch.float32,
+            num_blocks=64,
+            block_size=4,
+            weight_seed=1,
+        )
+        with pytest.raises(ValueError, match="empty prompt"):
+            eng.submit_tokens([], SamplingParams(max_tokens=3, seed=0))
+        # the engine stays fully serviceable afterwards
+        f = eng.submit_tokens([5], SamplingParams(max_tokens=3, seed=0))
+        eng.run_until_idle()
+        assert f.result(timeout=5).completion_tokens == 3
+        eng.stop()
+
+    def test_core_add_guard(self):
+        from dts_amd.core import load_core
+
+        core = load_core()
+        if core is None:
+            pytest.skip("native core not built")
+        sched = core.CoreScheduler(16, 4, 64, 64)
+        with pytest.raises(ValueError, match="empty prompt"):
+            sched.add(1, [], True)
