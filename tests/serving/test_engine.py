@@ -297,11 +297,18 @@ class TestGraphCaptureFallback:
         eng = self._engine()
         eng._graph_runner = _OtherGR()
         eng._chain = None
-        eng.submit_tokens([1, 2, 3], SamplingParams(max_tokens=2, seed=0))
-        # only capture errors are swallowed; anything else surfaces
-        # (here synchronously; in the background loop via failed futures)
-        with pytest.raises(RuntimeError, match="out of memory"):
-            eng.run_until_idle()
+        fut = eng.submit_tokens([1, 2, 3], SamplingParams(max_tokens=2, seed=0))
+        eng.run_until_idle()
+        # only capture errors degrade to eager; anything else fails the
+        # batch's futures loudly — and ONLY that batch: the engine stays
+        # serviceable (before the step-level guard, un-cleared in_flight
+        # flags live-locked every later request)
+        with pytest.raises(Exception, match="out of memory"):
+            fut.result(timeout=5)
+        eng._graph_runner = None
+        f2 = eng.submit_tokens([4, 5, 6], SamplingParams(max_tokens=2, seed=0))
+        eng.run_until_idle()
+        assert f2.result(timeout=5).completion_tokens >= 1
         eng.stop()
 
 
@@ -337,3 +344,59 @@ class TestEmptyPrompt:
         sched = core.CoreScheduler(16, 4, 64, 64)
         with pytest.raises(ValueError, match="empty prompt"):
             sched.add(1, [], True)
+
+
+class TestHostileInputs:
+    """Degenerate requests must fail their own future (or raise at
+    submit) without poisoning the engine for everyone else."""
+
+    @pytest.mark.parametrize("native", ["0", "1"])
+    def test_oob_token_id_rejected(self, native, monkeypatch):
+        monkeypatch.setenv("DTS_NATIVE_CORE", native)
+        eng = ServingEngine(
+            model_name="llama-tiny",
+            device="cpu",
+            dtype=torch.float32,
+            num_blocks=64,
+            block_size=4,
+            weight_seed=1,
+        )
+        with pytest.raises(ValueError, match="out of range"):
+            eng.submit_tokens([1, 2, 10_000], SamplingParams(max_tokens=2))
+        with pytest.raises(ValueError, match="out of range"):
+            eng.submit_tokens([1, -3], SamplingParams(max_tokens=2))
+        f = eng.submit_tokens([4, 5, 6], SamplingParams(max_tokens=2, seed=0))
+        eng.run_until_idle()
+        assert f.result(timeout=5).completion_tokens >= 1
+        eng.stop()
+
+    def test_failed_batch_scoped_not_global(self, monkeypatch):
+        """A mid-step model error aborts that batch's requests; a fresh
+        request afterwards is served normally."""
+        monkeypatch.setenv("DTS_NATIVE_CORE", "1")
+        eng = ServingEngine(
+            model_name="llama-tiny",
+            device="cpu",
+            dtype=torch.float32,
+            num_blocks=128,
+            block_size=4,
+            weight_seed=1,
+        )
+        orig_forward = eng.model.forward
+        calls = {"n": 0}
+
+        def boom(batch, kv_pool):
+            calls["n"] += 1
+            if calls["n"] == 1:
+                raise RuntimeError("injected model failure")
+            return orig_forward(batch, kv_pool)
+
+        monkeypatch.setattr(eng.model, "forward", boom)
+        bad = eng.submit_tokens([1, 2, 3], SamplingParams(max_tokens=2, seed=0))
+        eng.run_until_idle()
+        with pytest.raises(Exception, match="injected model failure"):
+            bad.result(timeout=5)
+        good = eng.submit_tokens([7, 8, 9], SamplingParams(max_tokens=2, seed=0))
+        eng.run_until_idle()
+        assert good.result(timeout=5).completion_tokens >= 1
+        eng.stop()
