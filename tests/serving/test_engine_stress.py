@@ -298,3 +298,73 @@ class TestThreadedLoopConcurrency:
         assert len(results) == 48
         eng.stop()
         assert not eng._futures
+
+
+class TestThreadedMixedPaths:
+    """The background step thread serving concurrent submitters across
+    ALL request kinds at once — guided forms, speculative-eligible
+    repetitive prompts, and streaming callbacks. Exercises the spec
+    verification walk and guide advancement under the engine lock."""
+
+    def test_mixed_kinds_under_running_loop(self, monkeypatch):
+        import threading
+
+        monkeypatch.setenv("DTS_NATIVE_CORE", "1")
+        eng = ServingEngine(
+            model_name="llama-tiny",
+            device="cpu",
+            dtype=torch.float32,
+            num_blocks=1024,
+            block_size=4,
+            weight_seed=1,
+            spec_k=4,
+        )
+        eng.start()
+        errors, results = [], []
+
+        def submitter(tid):
+            rng = random.Random(tid)
+            try:
+                for i in range(6):
+                    kind = rng.random()
+                    if kind < 0.3:
+                        g = strategy_form(eng.tokenizer, 2)
+                        f = eng.submit_tokens(
+                            [tid * 7 + j for j in range(5)],
+                            SamplingParams(max_tokens=64, seed=i, temperature=0.7),
+                            guide=g,
+                        )
+                        json.loads(f.result(timeout=60).text)
+                        results.append(1)
+                    elif kind < 0.6:
+                        motif = [rng.randrange(250, 400) for _ in range(4)]
+                        f = eng.submit_tokens(
+                            motif * 4,
+                            SamplingParams(max_tokens=24, seed=i, temperature=0.0),
+                        )
+                        results.append(f.result(timeout=60).completion_tokens)
+                    else:
+                        chunks = []
+                        f = eng.submit_tokens(
+                            [rng.randrange(1, 500) for _ in range(rng.randrange(5, 50))],
+                            SamplingParams(max_tokens=12, seed=i, temperature=0.8),
+                            stream_cb=lambda t, c=chunks: c.extend(t),
+                        )
+                        r = f.result(timeout=60)
+                        assert len(chunks) >= r.completion_tokens - 1
+                        results.append(r.completion_tokens)
+            except Exception:  # noqa: BLE001
+                import traceback
+
+                errors.append(traceback.format_exc())
+
+        threads = [threading.Thread(target=submitter, args=(t,)) for t in range(6)]
+        for t in threads:
+            t.start()
+        for t in threads:
+            t.join(timeout=120)
+            assert not t.is_alive(), "submitter hung"
+        assert not errors, errors[:1]
+        assert len(results) == 36
+        eng.stop()
+        assert not eng._futures
