@@ -414,6 +414,15 @@ class DTSEngine:
     def _load_tree(self, checkpoint) -> DialogueTree:
         """Rebuild a DialogueTree from a saved exploration dict
         (to_exploration_dict output, possibly loaded from disk)."""
+        try:
+            return self._load_tree_inner(checkpoint)
+        except Exception as e:  # noqa: BLE001 — malformed file, not a bug here
+            raise ValueError(
+                f"invalid checkpoint ({type(e).__name__}: {e}); expected "
+                "a to_exploration_dict JSON with a 'branches' list"
+            ) from e
+
+    def _load_tree_inner(self, checkpoint) -> DialogueTree:
         import json as _json
 
         cfg = self.config

@@ -160,3 +160,24 @@ class TestResumeEquivalence:
         resumed = asyncio.run(second_engine.run(rounds=1, resume_from=str(path)))
 
         assert self._shape(resumed) == self._shape(straight)
+
+
+class TestCorruptCheckpoint:
+    """Malformed resume files must raise an actionable ValueError (the
+    WS service turns it into an error event; the CLI prints it)."""
+
+    @pytest.mark.parametrize(
+        "content",
+        [
+            "garbage {{{",
+            json.dumps({"branches": 42}),
+            json.dumps({"branches": [{"depth": 1}]}),
+            json.dumps({"branches": [{"id": "x", "status": "banana"}]}),
+        ],
+    )
+    def test_invalid_checkpoint_raises_value_error(self, tmp_path, content):
+        p = tmp_path / "bad.json"
+        p.write_text(content)
+        engine, _ = make_engine()
+        with pytest.raises(ValueError, match="invalid checkpoint"):
+            asyncio.run(engine.run(rounds=1, resume_from=str(p)))
